@@ -1,0 +1,449 @@
+"""The embedding engine: dedup → route → lookup → postprocess → backward.
+
+This is the MI355X-native collapse of three reference tiers into one
+per-GPU object:
+
+* nn-worker forward/backward engines (rust/persia-core/src/forward.rs,
+  backward.rs) — prefetch pipeline, staleness, H2D/D2H;
+* embedding worker (embedding_worker_service/mod.rs) — dedup, sharding,
+  summation/raw postprocess, gradient merge;
+* embedding parameter server (embedding_parameter_service/mod.rs) — the
+  sharded table + sparse optimizers.
+
+Per batch (one dim-group):
+  1. sign prep: hashstack folding + feature-group prefix + splitmix64 mix
+     (the mixed key is both the dedup sort key and the shard router);
+  2. dedup: radix sort + unique over the *whole batch* (all slots at once —
+     the reference dedups per-slot per-worker with CPU hashmaps);
+  3. route: contiguous range-partition of the sorted unique keys by owner
+     rank, RCCL all_to_all_single over xGMI for keys, then rows (f16 wire,
+     like the reference's f16 EmbeddingBatch wire format);
+  4. owner-side: set-associative HBM hash-table probe/insert (HIP kernel);
+  5. postprocess: fused gather + segment-sum into per-slot (B, dim) f16
+     tensors, or raw distinct+index tensors (exact reference contract);
+  6. backward: reverse — segment grad scatter, all_to_all, fused
+     optimizer update on the owner shard.
+"""
+import queue
+import threading
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from persia_amd.core import hashing
+from persia_amd.core.comm import DistContext
+from persia_amd.core.schema import EmbeddingSchema, GlobalConfig, SlotConfig
+from persia_amd.core.store import make_store
+from persia_amd.embedding import EmbeddingConfig
+from persia_amd.embedding.data import PersiaBatch
+from persia_amd.embedding.optim import Optimizer
+from persia_amd.logger import get_default_logger
+from persia_amd.ops import reference as R
+
+_logger = get_default_logger("persia_amd.engine")
+
+_FLIP = -(2 ** 63)  # XOR with sign bit: signed sort order == unsigned order
+
+
+def _owner_of_keys(keys: torch.Tensor, world_size: int) -> torch.Tensor:
+    """Monotone range-partition of u64 bit-pattern keys (int64 tensors) to
+    ranks — same math as hashing.owner_of."""
+    hi = (keys >> 32) & 0xFFFFFFFF
+    return (hi * world_size) >> 32
+
+
+@dataclass
+class SlotPayload:
+    """What the dense side consumes for one slot (reference
+    FeatureEmbeddingBatch, persia-common/src/lib.rs:85-113)."""
+
+    name: str
+    cfg: SlotConfig
+    sum_tensor: Optional[torch.Tensor] = None  # [B, dim] f16
+    raw_distinct: Optional[torch.Tensor] = None  # [U_slot+1, dim] f16, row0 pad
+    raw_index: Optional[torch.Tensor] = None  # [B*sfs] i64
+    raw_non_empty_index: Optional[torch.Tensor] = None
+    raw_sample_id_num: Optional[torch.Tensor] = None
+
+    @property
+    def is_raw(self) -> bool:
+        return self.raw_distinct is not None
+
+
+@dataclass
+class _SlotCtx:
+    name: str
+    cfg: SlotConfig
+    pos_slice: Tuple[int, int]  # slice of group position space
+    seg_offsets: torch.Tensor  # [B+1] per-slot CSR (into its own positions)
+    slot_uniq_global: Optional[torch.Tensor] = None  # raw slots: local->global uniq
+    slot_inverse: Optional[torch.Tensor] = None  # raw slots: pos->local distinct
+
+
+@dataclass
+class _GroupCtx:
+    dim: int
+    uniq_keys: torch.Tensor  # [U] int64 (u64 bit pattern)
+    inverse: torch.Tensor  # [nnz] i64
+    slots: List[_SlotCtx] = field(default_factory=list)
+    send_counts: Optional[List[int]] = None
+    recv_counts: Optional[List[int]] = None
+
+
+class PersiaTrainingBatch:
+    """Device-side batch ready for the dense model (reference
+    PersiaTrainingBatch, persia-core/src/forward.rs:256-331)."""
+
+    def __init__(self):
+        self.payloads: List[SlotPayload] = []
+        self.non_id_type_tensors: List[torch.Tensor] = []
+        self.label_tensors: List[torch.Tensor] = []
+        self.batch_size: int = 0
+        self.requires_grad: bool = True
+        self.meta: Optional[bytes] = None
+        self.batch_id: Optional[int] = None
+        self._groups: List[_GroupCtx] = []
+        self._engine: Optional["EmbeddingEngine"] = None
+
+
+class EmbeddingEngine:
+    def __init__(
+        self,
+        schema: EmbeddingSchema,
+        hyper: EmbeddingConfig,
+        optimizer: Optional[Optimizer],
+        gconf: GlobalConfig,
+        device: torch.device,
+        dist_ctx: Optional[DistContext] = None,
+        wire_dtype: torch.dtype = torch.float16,
+    ):
+        self.schema = schema
+        self.hyper = hyper
+        self.optimizer = optimizer
+        self.gconf = gconf
+        self.device = device
+        self.dist = dist_ctx or DistContext.from_default_group()
+        self.wire_dtype = wire_dtype
+        # group slots by dim (slots of one dim share one store + one a2a)
+        self.groups: Dict[int, List[SlotConfig]] = {}
+        for name in schema.slot_names():
+            cfg = schema.get_slot(name)
+            self.groups.setdefault(cfg.dim, []).append(cfg)
+        if optimizer is None:
+            from persia_amd.embedding.optim import SGD
+
+            optimizer = SGD(lr=0.0)
+            self.optimizer = optimizer
+        self.stores = {
+            dim: make_store(dim, gconf.capacity, optimizer, hyper, device)
+            for dim in self.groups
+        }
+        self.skipped_grad_signs = 0
+        self.nan_grad_batches = 0
+
+    # ------------------------------------------------------------- sign prep
+
+    def _prepare_slot_keys(self, feat) -> Tuple[np.ndarray, np.ndarray]:
+        """raw uint64 ids -> prepared (mixed keys u64[nnz'], offsets i64[B+1]).
+        Applies hashstack folding then feature-group prefix then splitmix64
+        (reference: mod.rs:348-429 + sign_to_shard hashing)."""
+        cfg = self.schema.get_slot(feat.name)
+        values, offsets = feat.values, feat.offsets
+        rounds = cfg.hash_stack_rounds
+        if rounds > 0:
+            stacked = hashing.hash_stack(
+                values, rounds, cfg.hash_stack_config.embedding_size
+            )  # (rounds, nnz)
+            # per-sample segments repeat each position `rounds` times;
+            # interleave position-major so each sample's span stays contiguous:
+            # new position layout = for each sample, [r0 ids..., r1 ids..., ...]
+            B = len(offsets) - 1
+            lens = offsets[1:] - offsets[:-1]
+            new_vals = np.empty(len(values) * rounds, dtype=np.uint64)
+            new_offsets = np.zeros(B + 1, dtype=np.int64)
+            np.cumsum(lens * rounds, out=new_offsets[1:])
+            for r in range(rounds):
+                for b in range(B):
+                    s, e = offsets[b], offsets[b + 1]
+                    dst = new_offsets[b] + r * (e - s)
+                    new_vals[dst : dst + (e - s)] = stacked[r, s:e]
+            values, offsets = new_vals, new_offsets
+        signs = hashing.apply_prefix(values, cfg.index_prefix, self.schema.feature_spacing)
+        keys = hashing.splitmix64(signs)
+        keys[keys == np.uint64(0)] = np.uint64(0xD1B54A32D192ED03)
+        return keys, offsets
+
+    # ---------------------------------------------------------------- lookup
+
+    def _exchange_rows(self, group: _GroupCtx, train: bool) -> torch.Tensor:
+        """uniq keys -> rows [U, dim] (wire dtype on GPU path, f32 on CPU)."""
+        store = self.stores[group.dim]
+        if not self.dist.distributed:
+            rows = store.lookup(group.uniq_keys, train)
+            return rows
+        owner = _owner_of_keys(group.uniq_keys, self.dist.world_size)
+        send_counts = torch.bincount(owner, minlength=self.dist.world_size).tolist()
+        recv_counts = self.dist.all_to_all_lengths(send_counts)
+        group.send_counts, group.recv_counts = send_counts, recv_counts
+        recv_keys = self.dist.all_to_all(group.uniq_keys, send_counts, recv_counts)
+        rows_local = store.lookup(recv_keys, train).to(self.wire_dtype)
+        rows = self.dist.all_to_all(rows_local, recv_counts, send_counts)
+        return rows
+
+    # ---------------------------------------------------------- forward path
+
+    def process_batch(self, batch: PersiaBatch, train: Optional[bool] = None) -> PersiaTrainingBatch:
+        train = batch.requires_grad if train is None else train
+        if self.gconf.job_type == "infer":
+            train = False
+        out = PersiaTrainingBatch()
+        out.batch_size = batch.batch_size
+        out.requires_grad = batch.requires_grad
+        out.meta = batch.meta
+        out.batch_id = batch.batch_id
+        out._engine = self
+
+        dev = self.device
+        for x in batch.non_id_type_features:
+            out.non_id_type_tensors.append(torch.from_numpy(np.ascontiguousarray(x.data)).to(dev))
+        for x in batch.labels:
+            out.label_tensors.append(torch.from_numpy(np.ascontiguousarray(x.data)).to(dev))
+
+        feats_by_dim: Dict[int, List] = {}
+        for feat in batch.id_type_features:
+            cfg = self.schema.get_slot(feat.name)
+            feats_by_dim.setdefault(cfg.dim, []).append(feat)
+
+        for dim, feats in feats_by_dim.items():
+            group = self._process_group(dim, feats, out, train)
+            out._groups.append(group)
+        # keep payloads in the original id_type_features order
+        order = {f.name: i for i, f in enumerate(batch.id_type_features)}
+        out.payloads.sort(key=lambda p: order[p.name])
+        return out
+
+    def _process_group(self, dim: int, feats, out: PersiaTrainingBatch, train: bool) -> _GroupCtx:
+        dev = self.device
+        key_arrays = []
+        slot_ctxs: List[_SlotCtx] = []
+        pos = 0
+        for feat in feats:
+            cfg = self.schema.get_slot(feat.name)
+            keys, offsets = self._prepare_slot_keys(feat)
+            key_arrays.append(keys)
+            sc = _SlotCtx(
+                name=feat.name,
+                cfg=cfg,
+                pos_slice=(pos, pos + len(keys)),
+                seg_offsets=torch.from_numpy(np.ascontiguousarray(offsets)).to(dev),
+            )
+            slot_ctxs.append(sc)
+            pos += len(keys)
+
+        all_keys = np.concatenate(key_arrays) if key_arrays else np.empty(0, np.uint64)
+        keys_t = torch.from_numpy(all_keys.view(np.int64)).to(dev)
+        flipped = keys_t ^ _FLIP
+        uniq_flipped, inverse = torch.unique(flipped, sorted=True, return_inverse=True)
+        uniq_keys = uniq_flipped ^ _FLIP
+
+        group = _GroupCtx(dim=dim, uniq_keys=uniq_keys, inverse=inverse, slots=slot_ctxs)
+        rows = self._exchange_rows(group, train)  # [U, dim]
+
+        # --- postprocess into per-slot payloads
+        for sc in slot_ctxs:
+            s, e = sc.pos_slice
+            inv_slot = inverse[s:e]
+            payload = SlotPayload(name=sc.name, cfg=sc.cfg)
+            if sc.cfg.embedding_summation:
+                payload.sum_tensor = R.segment_sum_rows(
+                    rows, inv_slot, sc.seg_offsets, sc.cfg.sqrt_scaling, torch.float16
+                )
+            else:
+                slot_uniq, slot_inv = torch.unique(inv_slot, sorted=True, return_inverse=True)
+                sc.slot_uniq_global = slot_uniq
+                sc.slot_inverse = slot_inv
+                scale = 1.0
+                rounds = sc.cfg.hash_stack_rounds
+                if sc.cfg.sqrt_scaling and rounds > 1:
+                    scale = 1.0 / float(np.sqrt(rounds))
+                distinct, index, non_empty, num = R.raw_embedding_tensors(
+                    rows[slot_uniq],
+                    slot_inv,
+                    sc.seg_offsets,
+                    sc.cfg.sample_fixed_size,
+                    scale,
+                    torch.float16,
+                )
+                payload.raw_distinct = distinct
+                payload.raw_index = index
+                payload.raw_non_empty_index = non_empty
+                payload.raw_sample_id_num = num
+            out.payloads.append(payload)
+        return group
+
+    # --------------------------------------------------------- backward path
+
+    def apply_gradients(
+        self,
+        training_batch: PersiaTrainingBatch,
+        grads: Dict[str, Optional[torch.Tensor]],
+        loss_scale: float = 1.0,
+    ) -> None:
+        """grads: slot name -> grad tensor ((B,dim) f16 for sum slots,
+        (U_slot, dim) f32 for raw slots — the contract of ctx._on_backward,
+        reference persia/ctx.py:926-1005) or None (skipped)."""
+        for group in training_batch._groups:
+            U = group.uniq_keys.numel()
+            buf = torch.zeros(U, group.dim, dtype=torch.float32, device=self.device)
+            any_grad = False
+            for sc in group.slots:
+                g = grads.get(sc.name)
+                if g is None:
+                    continue
+                # NaN filter (reference mod.rs:731-746: skip the whole slot)
+                if bool(torch.isnan(g).any()):
+                    self.nan_grad_batches += 1
+                    _logger.warning(f"nan found in gradient update of {sc.name}, skipping")
+                    continue
+                any_grad = True
+                s, e = sc.pos_slice
+                inv_slot = group.inverse[s:e]
+                if sc.cfg.embedding_summation:
+                    contrib = R.segment_grad_scatter(
+                        g,
+                        inv_slot,
+                        sc.seg_offsets,
+                        U,
+                        scale_factor=loss_scale,
+                        sqrt_scaling=sc.cfg.sqrt_scaling,
+                    )
+                    buf += contrib
+                else:
+                    gf = g.float()
+                    if loss_scale != 1.0:
+                        gf = gf / loss_scale
+                    rounds = sc.cfg.hash_stack_rounds
+                    if sc.cfg.sqrt_scaling and rounds > 0:
+                        gf = gf / float(np.sqrt(rounds))
+                    buf.index_add_(0, sc.slot_uniq_global, gf)
+            if not any_grad:
+                continue
+            store = self.stores[group.dim]
+            if not self.dist.distributed:
+                self.skipped_grad_signs += store.update_gradients(group.uniq_keys, buf)
+            else:
+                send_counts, recv_counts = group.send_counts, group.recv_counts
+                if send_counts is None:
+                    owner = _owner_of_keys(group.uniq_keys, self.dist.world_size)
+                    send_counts = torch.bincount(owner, minlength=self.dist.world_size).tolist()
+                    recv_counts = self.dist.all_to_all_lengths(send_counts)
+                keys_recv = self.dist.all_to_all(group.uniq_keys, send_counts, recv_counts)
+                grads_recv = self.dist.all_to_all(
+                    buf.to(self.wire_dtype), send_counts, recv_counts
+                )
+                # merge duplicate keys across source ranks: pre-aggregate into
+                # one optimizer application per sign (the reference applies
+                # each rank's RPC sequentially under a per-sign lock — summing
+                # first is the synchronous-equivalent, deterministic, and
+                # race-free on GPU)
+                uniq_f, inv = torch.unique(
+                    keys_recv ^ _FLIP, sorted=True, return_inverse=True
+                )
+                merged = torch.zeros(
+                    uniq_f.numel(), group.dim, dtype=torch.float32, device=self.device
+                )
+                merged.index_add_(0, inv, grads_recv.float())
+                self.skipped_grad_signs += store.update_gradients(uniq_f ^ _FLIP, merged)
+
+    # ------------------------------------------------------------ checkpoint
+
+    def dump(self, dst_dir: str, blocking: bool = True) -> None:
+        from persia_amd.core.checkpoint import dump_embedding
+
+        dump_embedding(self, dst_dir)
+
+    def load(self, src_dir: str, blocking: bool = True) -> None:
+        from persia_amd.core.checkpoint import load_embedding
+
+        load_embedding(self, src_dir)
+
+    def num_resident_rows(self) -> int:
+        return sum(len(s) for s in self.stores.values())
+
+
+class ForwardPipeline:
+    """Async prefetch with bounded staleness (reference ForwardImpl,
+    forward.rs:470-528: input channel → lookup workers gated by a staleness
+    Semaphore → output channel; the permit is released when the batch's
+    gradients have been pushed — backward.rs:341-343)."""
+
+    def __init__(self, engine: EmbeddingEngine, staleness: int = 8, out_buffer: int = 8):
+        self.engine = engine
+        self.staleness = max(1, staleness)
+        self._sem = threading.Semaphore(self.staleness)
+        self._in: "queue.Queue" = queue.Queue(maxsize=max(2, out_buffer))
+        self._out: "queue.Queue" = queue.Queue(maxsize=max(2, out_buffer))
+        self._thread: Optional[threading.Thread] = None
+        self._stop = threading.Event()
+        self._exc: Optional[BaseException] = None
+
+    def start(self):
+        if self._thread is None:
+            self._thread = threading.Thread(target=self._run, daemon=True, name="persia-forward")
+            self._thread.start()
+
+    def _run(self):
+        # On GPU, lookups run on a dedicated HIP stream so the sparse pipeline
+        # overlaps the dense fwd/bwd on the default stream.
+        stream = None
+        if self.engine.device.type == "cuda":
+            stream = torch.cuda.Stream(device=self.engine.device)
+        while not self._stop.is_set():
+            try:
+                item = self._in.get(timeout=0.1)
+            except queue.Empty:
+                continue
+            if item is None:
+                self._out.put(None)
+                break
+            self._sem.acquire()
+            try:
+                if stream is not None:
+                    with torch.cuda.stream(stream):
+                        tb = self.engine.process_batch(item)
+                    ev = torch.cuda.Event()
+                    ev.record(stream)
+                    tb._ready_event = ev
+                else:
+                    tb = self.engine.process_batch(item)
+                self._out.put(tb)
+            except BaseException as e:  # propagate to consumer
+                self._exc = e
+                self._out.put(None)
+                break
+
+    def put(self, batch: PersiaBatch):
+        self.start()
+        self._in.put(batch)
+
+    def finish(self):
+        self.start()
+        self._in.put(None)
+
+    def get(self, timeout: Optional[float] = None) -> Optional[PersiaTrainingBatch]:
+        tb = self._out.get(timeout=timeout)
+        if tb is None and self._exc is not None:
+            raise self._exc
+        if tb is not None and getattr(tb, "_ready_event", None) is not None:
+            torch.cuda.current_stream().wait_event(tb._ready_event)
+        return tb
+
+    def release_permit(self):
+        """Called after the batch's gradients were applied (or the batch was
+        dropped) — the staleness window slides."""
+        self._sem.release()
+
+    def stop(self):
+        self._stop.set()

@@ -1,0 +1,386 @@
+"""The embedding store: one shard of the (conceptually trillion-row) table.
+
+Re-architecture of the reference parameter-server storage stack
+(`persia-embedding-holder`: Sharded<EvictionMap<u64, HashMapEmbeddingEntry>>,
+LRU via ArrayLinkedList — rust/persia-embedding-holder/src/lib.rs:28-64) as a
+GPU-friendly **set-associative hash table**:
+
+* keys:  ``uint64[n_slots]``   — splitmix64-mixed signs (bijective, so the
+  mixed key IS the hash; 0 = empty sentinel)
+* ticks: ``uint32[n_slots]``   — last-access batch counter (approximate LRU:
+  evict the min-tick slot of the probe window when full)
+* arena: ``float32[n_slots, row_width]`` — row = [emb(dim) | opt_state]
+  (same logical layout as the reference's HashMapEmbeddingEntry.inner)
+
+Buckets of ``BUCKET_SIZE`` slots; probing scans ``PROBE_BUCKETS`` consecutive
+buckets.  Eviction is bounded-window LRU instead of the reference's global
+linked-list LRU — O(1), no global state, GPU-parallel.
+
+Two implementations with identical semantics:
+* :class:`CpuEmbeddingStore` — numpy/torch sequential model (the oracle;
+  also the CPU execution backend).
+* :class:`HipEmbeddingStore` — HBM-resident, HIP kernels via persia_amd._C.
+"""
+from typing import Optional, Tuple
+
+import numpy as np
+import torch
+
+from persia_amd.core import hashing
+from persia_amd.embedding import EmbeddingConfig
+from persia_amd.embedding.optim import Adagrad, Adam, Optimizer, SGD
+
+BUCKET_SIZE = 8
+PROBE_BUCKETS = 4
+EMPTY_KEY = np.uint64(0)
+_ZERO_REMAP = np.uint64(0xD1B54A32D192ED03)  # mixed-key 0 is remapped here
+
+
+def _next_pow2(x: int) -> int:
+    p = 1
+    while p < x:
+        p <<= 1
+    return p
+
+
+def _u01_from_u64(u: np.ndarray) -> np.ndarray:
+    """Uniform [0,1) from the top 24 bits — must match csrc/common.h."""
+    return (u >> np.uint64(40)).astype(np.float64) * (1.0 / (1 << 24))
+
+
+def row_init(sign: int, dim: int, lo: float, hi: float) -> np.ndarray:
+    """Deterministic per-sign bounded-uniform init (reference seeds SmallRng
+    by sign — emb_entry.rs:35; we use splitmix64 streams, same both CPU/HIP)."""
+    seed = hashing.init_seed_for(np.array([sign], dtype=np.uint64))[0]
+    cols = np.arange(1, dim + 1, dtype=np.uint64)
+    u = hashing.splitmix64(np.uint64(seed) ^ cols)
+    return (lo + (hi - lo) * _u01_from_u64(u)).astype(np.float32)
+
+
+class EmbeddingStoreBase:
+    """One rank's shard for one dim-group of slots."""
+
+    def __init__(
+        self,
+        dim: int,
+        capacity: int,
+        optimizer: Optimizer,
+        hyper: EmbeddingConfig,
+        device: torch.device,
+    ):
+        self.dim = dim
+        self.optimizer = optimizer
+        self.hyper = hyper
+        self.device = device
+        self.opt_space = optimizer.require_space(dim)
+        self.row_width = dim + self.opt_space
+        self.n_buckets = _next_pow2(max(1, capacity // BUCKET_SIZE))
+        self.n_slots = self.n_buckets * BUCKET_SIZE
+        self.tick = 1  # current batch counter (0 reserved)
+        # Adam per-group beta powers (persia-common optim.rs:147-216); kept
+        # per-store and stepped once per update call (all slot-groups in a
+        # store receive every update batch in this architecture).
+        if isinstance(optimizer, Adam):
+            self.beta1_power = optimizer.betas[0]
+            self.beta2_power = optimizer.betas[1]
+
+    def __len__(self) -> int:
+        raise NotImplementedError
+
+    def lookup(self, keys: torch.Tensor, train: bool) -> torch.Tensor:
+        """keys: int64[n] (mixed u64 bit pattern) -> rows float32[n, dim].
+        train=True inserts on miss (admit-gated, seeded init);
+        train=False returns zeros on miss (reference parameter
+        mod.rs:231-251)."""
+        raise NotImplementedError
+
+    def update_gradients(self, keys: torch.Tensor, grads: torch.Tensor) -> int:
+        """Apply the sparse optimizer to rows addressed by keys; skip keys no
+        longer resident. Returns number skipped."""
+        raise NotImplementedError
+
+    def export_rows(self) -> Tuple[np.ndarray, np.ndarray]:
+        """-> (signs u64[n], inner f32[n, row_width]) of every resident row."""
+        raise NotImplementedError
+
+    def import_rows(self, signs: np.ndarray, inner: np.ndarray) -> None:
+        raise NotImplementedError
+
+    def clear(self) -> None:
+        raise NotImplementedError
+
+    def next_tick(self) -> int:
+        t = self.tick
+        self.tick += 1
+        return t
+
+    def _adam_step_powers(self):
+        o = self.optimizer
+        if isinstance(o, Adam):
+            b1p, b2p = self.beta1_power, self.beta2_power
+            self.beta1_power *= o.betas[0]
+            self.beta2_power *= o.betas[1]
+            return b1p, b2p
+        return None
+
+
+class CpuEmbeddingStore(EmbeddingStoreBase):
+    """Sequential oracle + CPU execution backend."""
+
+    def __init__(self, dim, capacity, optimizer, hyper, device=torch.device("cpu")):
+        super().__init__(dim, capacity, optimizer, hyper, device)
+        self.keys = np.zeros(self.n_slots, dtype=np.uint64)
+        self.ticks = np.zeros(self.n_slots, dtype=np.uint32)
+        self.arena = torch.zeros(self.n_slots, self.row_width, dtype=torch.float32)
+        self._count = 0
+
+    def __len__(self) -> int:
+        return self._count
+
+    # -- probing ------------------------------------------------------------
+    def _probe(self, k: np.uint64) -> int:
+        """Find slot of key k, or -1."""
+        mask = np.uint64(self.n_buckets - 1)
+        b0 = int(k & mask)
+        for p in range(PROBE_BUCKETS):
+            base = ((b0 + p) % self.n_buckets) * BUCKET_SIZE
+            for s in range(BUCKET_SIZE):
+                if self.keys[base + s] == k:
+                    return base + s
+        return -1
+
+    def _probe_or_claim(self, k: np.uint64, tick: int) -> Tuple[int, bool]:
+        """-> (slot, is_new). Claims an empty slot, else evicts the min-tick
+        slot of the probe window. Deterministic given serialized access."""
+        mask = np.uint64(self.n_buckets - 1)
+        b0 = int(k & mask)
+        empty = -1
+        victim, victim_tick = -1, None
+        for p in range(PROBE_BUCKETS):
+            base = ((b0 + p) % self.n_buckets) * BUCKET_SIZE
+            for s in range(BUCKET_SIZE):
+                i = base + s
+                ki = self.keys[i]
+                if ki == k:
+                    return i, False
+                if ki == EMPTY_KEY:
+                    if empty < 0:
+                        empty = i
+                elif victim_tick is None or self.ticks[i] < victim_tick:
+                    victim, victim_tick = i, self.ticks[i]
+        if empty >= 0:
+            self.keys[empty] = k
+            self._count += 1
+            return empty, True
+        self.keys[victim] = k  # evict (bounded-window LRU)
+        return victim, True
+
+    def _init_row(self, slot: int, k: np.uint64) -> None:
+        sign = int(hashing.splitmix64_inv(np.array([k], dtype=np.uint64))[0])
+        lo, hi = self.hyper.emb_initialization
+        emb = row_init(sign, self.dim, lo, hi)
+        row = torch.zeros(self.row_width, dtype=torch.float32)
+        row[: self.dim] = torch.from_numpy(emb)
+        if self.opt_space:
+            row[self.dim :] = self.optimizer.state_init(self.dim)
+        self.arena[slot] = row
+
+    def _admitted(self, k: np.uint64, tick: int) -> bool:
+        if self.hyper.admit_probability >= 1.0:
+            return True
+        u = hashing.splitmix64(np.array([k ^ np.uint64(tick)], dtype=np.uint64))
+        return float(_u01_from_u64(u)[0]) < self.hyper.admit_probability
+
+    # -- public ops ---------------------------------------------------------
+    def lookup(self, keys: torch.Tensor, train: bool) -> torch.Tensor:
+        tick = self.next_tick()
+        k_np = keys.cpu().numpy().view(np.uint64).copy()
+        k_np[k_np == EMPTY_KEY] = _ZERO_REMAP
+        out = torch.zeros(len(k_np), self.dim, dtype=torch.float32)
+        for i, k in enumerate(k_np):
+            if train:
+                if self._admitted(k, tick) or self._probe(k) >= 0:
+                    slot, is_new = self._probe_or_claim(k, tick)
+                    if is_new:
+                        self._init_row(slot, k)
+                    self.ticks[slot] = tick
+                    out[i] = self.arena[slot, : self.dim]
+            else:
+                slot = self._probe(k)
+                if slot >= 0:
+                    self.ticks[slot] = tick
+                    out[i] = self.arena[slot, : self.dim]
+        return out
+
+    def update_gradients(self, keys: torch.Tensor, grads: torch.Tensor) -> int:
+        powers = self._adam_step_powers()
+        k_np = keys.cpu().numpy().view(np.uint64).copy()
+        k_np[k_np == EMPTY_KEY] = _ZERO_REMAP
+        slots = np.array([self._probe(k) for k in k_np], dtype=np.int64)
+        present = slots >= 0
+        skipped = int((~present).sum())
+        if present.sum() == 0:
+            return skipped
+        idx = torch.from_numpy(slots[present])
+        g = grads.float().cpu()[torch.from_numpy(np.nonzero(present)[0])]
+        rows = self.arena[idx]  # copy [n, row_width]
+        emb = rows[:, : self.dim]
+        from persia_amd.ops import reference as R
+
+        o = self.optimizer
+        wb = self.hyper.weight_bound
+        if isinstance(o, SGD):
+            R.sgd_update(emb, g, o.lr, o.weight_decay, wb)
+        elif isinstance(o, Adagrad):
+            accum = rows[:, self.dim :] if not o.vectorwise_shared else rows[:, self.dim : self.dim + 1]
+            R.adagrad_update(
+                emb, accum, g, o.lr, o.g_square_momentum, o.eps, wb, o.vectorwise_shared
+            )
+        elif isinstance(o, Adam):
+            m = rows[:, self.dim : 2 * self.dim]
+            v = rows[:, 2 * self.dim :]
+            b1p, b2p = powers
+            R.adam_update(emb, m, v, g, b1p, b2p, o.lr, o.betas[0], o.betas[1], o.eps, wb)
+        else:
+            raise ValueError(f"unknown optimizer {o.kind}")
+        self.arena[idx] = rows
+        return skipped
+
+    def export_rows(self) -> Tuple[np.ndarray, np.ndarray]:
+        occ = np.nonzero(self.keys != EMPTY_KEY)[0]
+        signs = hashing.splitmix64_inv(self.keys[occ])
+        inner = self.arena[torch.from_numpy(occ)].numpy()
+        return signs, inner
+
+    def import_rows(self, signs: np.ndarray, inner: np.ndarray) -> None:
+        assert inner.shape[1] == self.row_width, (
+            f"checkpoint row width {inner.shape[1]} != store row width {self.row_width}"
+        )
+        ks = hashing.splitmix64(signs.astype(np.uint64))
+        ks[ks == EMPTY_KEY] = _ZERO_REMAP
+        tick = self.next_tick()
+        for i, k in enumerate(ks):
+            slot, _ = self._probe_or_claim(k, tick)
+            self.arena[slot] = torch.from_numpy(inner[i].astype(np.float32))
+            self.ticks[slot] = tick
+
+    def clear(self) -> None:
+        self.keys[:] = EMPTY_KEY
+        self.ticks[:] = 0
+        self.arena.zero_()
+        self._count = 0
+
+
+class HipEmbeddingStore(EmbeddingStoreBase):
+    """HBM-resident shard driven by HIP kernels (persia_amd._C).
+
+    All state lives in torch CUDA tensors (allocated through the caching
+    allocator so it coexists with the dense model); kernels mutate them in
+    place."""
+
+    def __init__(self, dim, capacity, optimizer, hyper, device):
+        super().__init__(dim, capacity, optimizer, hyper, device)
+        from persia_amd.ops import native
+
+        self._C = native()
+        assert device.type == "cuda"
+        self.keys = torch.zeros(self.n_slots, dtype=torch.int64, device=device)
+        self.ticks = torch.zeros(self.n_slots, dtype=torch.int32, device=device)
+        self.arena = torch.zeros(
+            self.n_slots, self.row_width, dtype=torch.float32, device=device
+        )
+        self._opt_code = {"sgd": 0, "adagrad": 1, "adam": 2}[optimizer.kind]
+
+    def _opt_params(self):
+        o = self.optimizer
+        if isinstance(o, SGD):
+            return [o.lr, o.weight_decay, 0.0, 0.0, 0.0, 0.0]
+        if isinstance(o, Adagrad):
+            return [
+                o.lr,
+                o.g_square_momentum,
+                o.eps,
+                1.0 if o.vectorwise_shared else 0.0,
+                0.0,
+                0.0,
+            ]
+        if isinstance(o, Adam):
+            return [o.lr, o.betas[0], o.betas[1], o.eps, 0.0, 0.0]
+        raise ValueError(o.kind)
+
+    def __len__(self) -> int:
+        return int(self._C.store_count(self.keys))
+
+    def lookup(self, keys: torch.Tensor, train: bool) -> torch.Tensor:
+        tick = self.next_tick()
+        out = torch.empty(keys.numel(), self.dim, dtype=torch.float32, device=self.device)
+        lo, hi = self.hyper.emb_initialization
+        self._C.store_lookup(
+            self.keys,
+            self.ticks,
+            self.arena,
+            keys,
+            out,
+            self.dim,
+            int(train),
+            tick,
+            float(lo),
+            float(hi),
+            float(self.hyper.admit_probability),
+            float(self.optimizer.state_init(self.dim)),
+            self.opt_space,
+        )
+        return out
+
+    def update_gradients(self, keys: torch.Tensor, grads: torch.Tensor) -> int:
+        powers = self._adam_step_powers()
+        b1p, b2p = powers if powers else (0.0, 0.0)
+        skipped = self._C.store_update(
+            self.keys,
+            self.ticks,
+            self.arena,
+            keys,
+            grads.float(),
+            self.dim,
+            self._opt_code,
+            self._opt_params(),
+            float(b1p),
+            float(b2p),
+            float(self.hyper.weight_bound),
+        )
+        return int(skipped)
+
+    def export_rows(self) -> Tuple[np.ndarray, np.ndarray]:
+        occ = torch.nonzero(self.keys != 0, as_tuple=False).view(-1)
+        keys = self.keys[occ].cpu().numpy().view(np.uint64)
+        signs = hashing.splitmix64_inv(keys)
+        inner = self.arena[occ].cpu().numpy()
+        return signs, inner
+
+    def import_rows(self, signs: np.ndarray, inner: np.ndarray) -> None:
+        assert inner.shape[1] == self.row_width
+        ks = hashing.splitmix64(signs.astype(np.uint64))
+        ks[ks == EMPTY_KEY] = _ZERO_REMAP
+        keys_t = torch.from_numpy(ks.view(np.int64)).to(self.device)
+        rows_t = torch.from_numpy(np.ascontiguousarray(inner, dtype=np.float32)).to(
+            self.device
+        )
+        tick = self.next_tick()
+        self._C.store_import(self.keys, self.ticks, self.arena, keys_t, rows_t, tick)
+
+    def clear(self) -> None:
+        self.keys.zero_()
+        self.ticks.zero_()
+        self.arena.zero_()
+
+
+def make_store(
+    dim: int,
+    capacity: int,
+    optimizer: Optimizer,
+    hyper: EmbeddingConfig,
+    device: torch.device,
+) -> EmbeddingStoreBase:
+    if device.type == "cuda":
+        return HipEmbeddingStore(dim, capacity, optimizer, hyper, device)
+    return CpuEmbeddingStore(dim, capacity, optimizer, hyper, device)

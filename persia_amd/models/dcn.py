@@ -1,0 +1,47 @@
+"""DCN-v2 style dense tower (BASELINE.json config 4)."""
+from typing import List
+
+import torch
+import torch.nn as nn
+
+
+class CrossLayerV2(nn.Module):
+    """x_{l+1} = x0 * (W x_l + b) + x_l (DCN-v2 full-rank cross)."""
+
+    def __init__(self, dim: int):
+        super().__init__()
+        self.w = nn.Linear(dim, dim)
+
+    def forward(self, x0: torch.Tensor, xl: torch.Tensor) -> torch.Tensor:
+        return x0 * self.w(xl) + xl
+
+
+class DCNv2(nn.Module):
+    def __init__(
+        self,
+        num_sparse: int = 26,
+        num_dense: int = 13,
+        dim: int = 64,
+        num_cross: int = 3,
+        deep: List[int] = (512, 256, 128),
+    ):
+        super().__init__()
+        in_dim = num_dense + num_sparse * dim
+        self.cross = nn.ModuleList([CrossLayerV2(in_dim) for _ in range(num_cross)])
+        layers: List[nn.Module] = []
+        sizes = [in_dim] + list(deep)
+        for i in range(len(sizes) - 1):
+            layers += [nn.Linear(sizes[i], sizes[i + 1]), nn.ReLU()]
+        self.deep = nn.Sequential(*layers)
+        self.head = nn.Linear(in_dim + sizes[-1], 1)
+
+    def forward(
+        self, non_id_tensors: List[torch.Tensor], embedding_tensors: List[torch.Tensor]
+    ) -> torch.Tensor:
+        dense = non_id_tensors[0].float()
+        x0 = torch.cat([dense] + [e.flatten(1).float() for e in embedding_tensors], dim=1)
+        xl = x0
+        for layer in self.cross:
+            xl = layer(x0, xl)
+        d = self.deep(x0)
+        return self.head(torch.cat([xl, d], dim=1)).squeeze(1)

@@ -1,0 +1,62 @@
+"""DLRM (Deep Learning Recommendation Model) dense tower.
+
+The benchmark model of BASELINE.json configs 2-3 (26 sparse slots, 13 dense
+features, dim=128 — the Criteo shape).  Bottom MLP embeds the dense
+features, a dot-product feature interaction crosses all embedding vectors,
+and a top MLP produces the CTR logit.
+"""
+from typing import List
+
+import torch
+import torch.nn as nn
+
+
+def _mlp(sizes: List[int], last_relu: bool = False) -> nn.Sequential:
+    layers: List[nn.Module] = []
+    for i in range(len(sizes) - 1):
+        layers.append(nn.Linear(sizes[i], sizes[i + 1]))
+        if i < len(sizes) - 2 or last_relu:
+            layers.append(nn.ReLU())
+    return nn.Sequential(*layers)
+
+
+class DotInteraction(nn.Module):
+    """Pairwise dot products of the (num_slots+1) feature vectors, lower
+    triangle (the DLRM interaction op; fused HIP kernel planned —
+    the matmul runs on MFMA via hipBLASLt through torch.bmm)."""
+
+    def forward(self, vectors: torch.Tensor) -> torch.Tensor:
+        # vectors: [B, F, D]
+        B, F, _D = vectors.shape
+        prod = torch.bmm(vectors, vectors.transpose(1, 2))  # [B, F, F]
+        li, lj = torch.tril_indices(F, F, offset=-1, device=vectors.device)
+        return prod[:, li, lj]  # [B, F*(F-1)/2]
+
+
+class DLRM(nn.Module):
+    def __init__(
+        self,
+        num_sparse: int = 26,
+        num_dense: int = 13,
+        dim: int = 128,
+        bottom_mlp: List[int] = (512, 256, 128),
+        top_mlp: List[int] = (1024, 1024, 512, 256),
+    ):
+        super().__init__()
+        self.dim = dim
+        self.num_sparse = num_sparse
+        self.bottom = _mlp([num_dense] + list(bottom_mlp) + [dim], last_relu=True)
+        self.interaction = DotInteraction()
+        n_f = num_sparse + 1
+        inter_out = n_f * (n_f - 1) // 2
+        self.top = _mlp([dim + inter_out] + list(top_mlp) + [1])
+
+    def forward(
+        self, non_id_tensors: List[torch.Tensor], embedding_tensors: List[torch.Tensor]
+    ) -> torch.Tensor:
+        dense = non_id_tensors[0].float()
+        x = self.bottom(dense)  # [B, D]
+        vectors = torch.stack([x] + [e.to(x.dtype) for e in embedding_tensors], dim=1)
+        inter = self.interaction(vectors)
+        out = self.top(torch.cat([x, inter], dim=1))
+        return out.squeeze(1)  # logits

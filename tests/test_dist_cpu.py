@@ -1,0 +1,113 @@
+"""Multi-process CPU (gloo) tests of the sharded embedding path.
+
+Load-bearing properties:
+* seeded-by-sign init + sign-sharded stores make lookup results INDEPENDENT
+  of world size (reference SURVEY §4 reproducibility fixture): a 2-rank run
+  produces bitwise the tensors of a 1-rank run;
+* gradients from several DP ranks are pre-aggregated per sign on the owner,
+  so a 2-rank run with identical replicas equals a 1-rank run with doubled
+  gradients.
+"""
+import os
+
+import numpy as np
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from persia_amd.utils import find_free_port
+
+WORLD = 2
+N_STEPS = 3
+
+
+def _schema():
+    from persia_amd.core.schema import EmbeddingSchema, SlotConfig
+
+    return EmbeddingSchema(
+        slots={
+            "a": SlotConfig(name="a", dim=8),
+            "b": SlotConfig(name="b", dim=8, sqrt_scaling=True),
+        }
+    )
+
+
+def _make_engine(dist_ctx):
+    from persia_amd.core.engine import EmbeddingEngine
+    from persia_amd.core.schema import GlobalConfig
+    from persia_amd.embedding import EmbeddingConfig
+    from persia_amd.embedding.optim import Adagrad
+
+    return EmbeddingEngine(
+        schema=_schema(),
+        hyper=EmbeddingConfig(emb_initialization=(-0.5, 0.5)),
+        optimizer=Adagrad(lr=0.1),
+        gconf=GlobalConfig(capacity=1 << 12),
+        device=torch.device("cpu"),
+        dist_ctx=dist_ctx,
+        wire_dtype=torch.float32,  # bitwise-comparable against 1-rank run
+    )
+
+
+def _batch(seed=0, B=16):
+    from persia_amd.embedding.data import IDTypeFeature, Label, PersiaBatch
+
+    rng = np.random.default_rng(seed)
+    feats = []
+    for name in ("a", "b"):
+        feats.append(
+            IDTypeFeature(
+                name,
+                [
+                    rng.integers(0, 200, size=rng.integers(1, 6), dtype=np.uint64)
+                    for _ in range(B)
+                ],
+            )
+        )
+    return PersiaBatch(feats, labels=[Label(np.ones((B, 1), np.float32))], requires_grad=True)
+
+
+def _grads(mult=1.0):
+    return {
+        "a": torch.full((16, 8), 0.25 * mult, dtype=torch.float32),
+        "b": torch.full((16, 8), -0.5 * mult, dtype=torch.float32),
+    }
+
+
+def _run_steps(eng, grad_mult=1.0):
+    outs = []
+    for step in range(N_STEPS):
+        tb = eng.process_batch(_batch(seed=step))
+        outs.append([p.sum_tensor.clone() for p in tb.payloads])
+        eng.apply_gradients(tb, _grads(grad_mult))
+    return outs
+
+
+def _worker(rank, port, result_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=WORLD)
+    from persia_amd.core.comm import DistContext
+
+    eng = _make_engine(DistContext.from_default_group())
+    # both ranks see identical batches/grads (worst-case total key overlap)
+    outs = _run_steps(eng, grad_mult=1.0)
+    if rank == 0:
+        torch.save(outs, os.path.join(result_dir, "dist_out.pt"))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_world2_bitwise_matches_world1(tmp_path):
+    from persia_amd.core.comm import DistContext
+
+    # 1-rank reference: two identical replicas pushing grads g merge into one
+    # per-sign application of 2g on the owner -> emulate with grad_mult=2
+    ref = _run_steps(_make_engine(DistContext(1, 0)), grad_mult=2.0)
+    port = find_free_port()
+    mp.spawn(_worker, args=(port, str(tmp_path)), nprocs=WORLD, join=True)
+    got = torch.load(tmp_path / "dist_out.pt")
+    assert len(got) == N_STEPS
+    for step in range(N_STEPS):
+        for p_ref, p_got in zip(ref[step], got[step]):
+            assert torch.equal(p_ref, p_got), f"mismatch at step {step}"
