@@ -1,0 +1,219 @@
+"""Flagship benchmark: DLRM-Criteo-shape hybrid training throughput.
+
+BASELINE.json metric: samples/sec (whole node) for a DLRM-style model —
+26 sparse slots (1e8-row ID space, dim=128, embeddings resident in HBM as a
+sharded hash table), 13 dense features, synthetic uniform ID/label data,
+random-init weights, bf16 dense compute, f16 embedding wire, Adagrad sparse
+optimizer, bounded-staleness async sparse pipeline + DDP dense.
+
+Contract (driver):
+  python bench.py --gpus N --steps K --warmup W
+N>1 is launched via torch.distributed.run (one rank per GPU over RCCL);
+weak scaling (per-GPU batch fixed).  Rank 0 prints ONE json line.
+"""
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch-size", type=int, default=4096, help="per-GPU batch")
+    p.add_argument("--num-sparse", type=int, default=26)
+    p.add_argument("--num-dense", type=int, default=13)
+    p.add_argument("--dim", type=int, default=128)
+    p.add_argument("--rows", type=float, default=1e8, help="total ID space")
+    p.add_argument("--capacity", type=float, default=None,
+                   help="resident rows per rank (default: rows/world)")
+    p.add_argument("--staleness", type=int, default=8)
+    p.add_argument("--model", type=str, default="dlrm", choices=["dlrm", "dcn"])
+    p.add_argument("--device", type=str, default=None)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    use_gpu = torch.cuda.is_available() and args.device != "cpu"
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+
+    if world > 1:
+        import torch.distributed as dist
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29500")
+        dist.init_process_group("nccl" if use_gpu else "gloo", rank=rank, world_size=world)
+
+    from persia_amd.core.comm import DistContext
+    from persia_amd.core.engine import EmbeddingEngine, ForwardPipeline
+    from persia_amd.core.schema import EmbeddingSchema, GlobalConfig, SlotConfig
+    from persia_amd.embedding import EmbeddingConfig
+    from persia_amd.embedding.data import IDTypeFeatureWithSingleID, Label, PersiaBatch
+    from persia_amd.embedding.optim import Adagrad
+    from persia_amd.models import DCNv2, DLRM
+
+    n_slots, dim, B = args.num_sparse, args.dim, args.batch_size
+    rows_total = int(args.rows)
+    vocab_per_slot = max(1, rows_total // n_slots)
+    capacity = int(args.capacity) if args.capacity else max(1 << 20, rows_total // world)
+
+    schema = EmbeddingSchema(
+        slots={f"f{i}": SlotConfig(name=f"f{i}", dim=dim) for i in range(n_slots)},
+        feature_index_prefix_bit=8,
+    )
+    engine = EmbeddingEngine(
+        schema=schema,
+        hyper=EmbeddingConfig(),
+        optimizer=Adagrad(lr=0.01),
+        gconf=GlobalConfig(capacity=capacity),
+        device=device,
+        dist_ctx=DistContext.from_default_group(),
+    )
+    if args.model == "dlrm":
+        model = DLRM(num_sparse=n_slots, num_dense=args.num_dense, dim=dim).to(device)
+    else:
+        model = DCNv2(num_sparse=n_slots, num_dense=args.num_dense, dim=dim).to(device)
+    if world > 1:
+        from torch.nn.parallel import DistributedDataParallel
+
+        model = DistributedDataParallel(
+            model,
+            device_ids=[local_rank] if use_gpu else None,
+            gradient_as_bucket_view=True,
+            bucket_cap_mb=50,
+        )
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    loss_fn = torch.nn.functional.binary_cross_entropy_with_logits
+
+    # ---- synthetic data: pre-generate host batches (excluded from timing)
+    n_batches = args.steps + args.warmup
+    rng = np.random.default_rng(1234 + rank)
+    host_batches = []
+    for _ in range(min(n_batches, 16)):
+        feats = [
+            IDTypeFeatureWithSingleID(
+                f"f{i}",
+                (rng.integers(0, vocab_per_slot, size=B, dtype=np.uint64)),
+            )
+            for i in range(n_slots)
+        ]
+        dense = rng.normal(size=(B, args.num_dense)).astype(np.float32)
+        label = rng.integers(0, 2, size=(B,)).astype(np.float32)
+        host_batches.append(
+            PersiaBatch(
+                feats,
+                non_id_type_features=[dense],
+                labels=[Label(label)],
+                requires_grad=True,
+            )
+        )
+
+    pipeline = ForwardPipeline(engine, staleness=args.staleness, out_buffer=args.staleness + 2)
+    pipeline.start()
+
+    amp_dtype = torch.bfloat16
+    amp_ctx = torch.autocast("cuda", dtype=amp_dtype) if use_gpu else _nullcontext()
+
+    def train_step(tb):
+        with amp_ctx:
+            embs = [p.sum_tensor for p in tb.payloads]
+            for e in embs:
+                e.requires_grad = True
+            logits = model(tb.non_id_type_tensors, embs)
+            loss = loss_fn(logits.float(), tb.label_tensors[0])
+        loss.backward()
+        engine.apply_gradients(tb, {p.name: e.grad for p, e in zip(tb.payloads, embs)})
+        opt.step()
+        opt.zero_grad(set_to_none=True)
+        pipeline.release_permit()
+        return loss
+
+    def feed(n):
+        for i in range(n):
+            pipeline.put(host_batches[i % len(host_batches)])
+
+    import threading
+
+    feeder = threading.Thread(
+        target=feed, args=(args.steps + args.warmup,), daemon=True
+    )
+    feeder.start()
+
+    for _ in range(args.warmup):
+        train_step(pipeline.get())
+
+    if world > 1:
+        import torch.distributed as dist
+
+        dist.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        train_step(pipeline.get())
+    if world > 1:
+        dist.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device if use_gpu else None)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    pipeline.stop()
+    samples = args.steps * B * world
+    if rank == 0:
+        result = {
+            "metric": "samples/sec (whole node), DLRM-style hybrid training",
+            "value": samples / elapsed,
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic (uniform ids, random labels, random-init weights)",
+            "config": {
+                "model": f"{args.model}-criteo {args.num_sparse}sparse/{args.num_dense}dense dim{dim}",
+                "global_batch": B * world,
+                "rows": rows_total,
+                "capacity_per_rank": capacity,
+                "staleness": args.staleness,
+                "parallelism": f"dp{world}+emb-a2a",
+            },
+        }
+        print(json.dumps(result))
+    if world > 1:
+        dist.destroy_process_group()
+
+
+class _nullcontext:
+    def __enter__(self):
+        return None
+
+    def __exit__(self, *a):
+        return False
+
+
+if __name__ == "__main__":
+    main()

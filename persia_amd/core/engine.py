@@ -54,6 +54,37 @@ def _owner_of_keys(keys: torch.Tensor, world_size: int) -> torch.Tensor:
     return (hi * world_size) >> 32
 
 
+def _dedup(keys_t: torch.Tensor):
+    """Sort-based dedup of u64-bit-pattern keys.
+
+    -> (uniq [U] ascending in u64 order, inverse [nnz] pos->uniq idx,
+        perm [nnz] sort permutation, ustarts [U+1] uniq boundaries in sorted
+        order).  perm/ustarts drive the ordered (deterministic, atomic-free)
+        gradient scatter kernel: a unique key's positions are contiguous in
+        sort order.  torch.sort on GPU is a hipCUB radix sort.
+    """
+    dev = keys_t.device
+    nnz = keys_t.numel()
+    if nnz == 0:
+        z = torch.zeros(0, dtype=torch.int64, device=dev)
+        return z, z, z, torch.zeros(1, dtype=torch.int64, device=dev)
+    flipped = keys_t ^ _FLIP  # signed sort order == unsigned key order
+    svals, perm = torch.sort(flipped)
+    neq = torch.ones(nnz, dtype=torch.bool, device=dev)
+    neq[1:] = svals[1:] != svals[:-1]
+    inv_sorted = torch.cumsum(neq, 0) - 1
+    inverse = torch.empty(nnz, dtype=torch.int64, device=dev)
+    inverse[perm] = inv_sorted
+    uniq = svals[neq] ^ _FLIP
+    ustarts = torch.cat(
+        [
+            torch.nonzero(neq, as_tuple=False).view(-1),
+            torch.tensor([nnz], dtype=torch.int64, device=dev),
+        ]
+    )
+    return uniq, inverse, perm, ustarts
+
+
 @dataclass
 class SlotPayload:
     """What the dense side consumes for one slot (reference
@@ -78,6 +109,7 @@ class _SlotCtx:
     cfg: SlotConfig
     pos_slice: Tuple[int, int]  # slice of group position space
     seg_offsets: torch.Tensor  # [B+1] per-slot CSR (into its own positions)
+    sum_seg_base: int = -1  # sum slots: base index in the group segment space
     slot_uniq_global: Optional[torch.Tensor] = None  # raw slots: local->global uniq
     slot_inverse: Optional[torch.Tensor] = None  # raw slots: pos->local distinct
 
@@ -87,7 +119,14 @@ class _GroupCtx:
     dim: int
     uniq_keys: torch.Tensor  # [U] int64 (u64 bit pattern)
     inverse: torch.Tensor  # [nnz] i64
+    perm: torch.Tensor  # [nnz] sort permutation (ordered grad scatter)
+    ustarts: torch.Tensor  # [U+1] unique boundaries in sorted order
     slots: List[_SlotCtx] = field(default_factory=list)
+    # fused sum-slot segment space (sum slots occupy positions [0, sum_end))
+    cat_offsets: Optional[torch.Tensor] = None  # [n_sum_segs+1]
+    seg_id: Optional[torch.Tensor] = None  # [nnz]: pos -> group seg idx, -1 raw
+    seg_lens: Optional[torch.Tensor] = None  # [n_sum_segs] f32
+    sqrt_mask: Optional[torch.Tensor] = None  # [n_sum_segs] bool
     send_counts: Optional[List[int]] = None
     recv_counts: Optional[List[int]] = None
 
@@ -226,61 +265,112 @@ class EmbeddingEngine:
 
     def _process_group(self, dim: int, feats, out: PersiaTrainingBatch, train: bool) -> _GroupCtx:
         dev = self.device
+        native = dev.type == "cuda"
+        if native:
+            from persia_amd.ops import native as _native
+
+            C = _native()
+        # sum slots first so they occupy a contiguous prefix of the position
+        # space (one fused segment-sum launch for the whole group)
+        feats = sorted(
+            feats, key=lambda f: 0 if self.schema.get_slot(f.name).embedding_summation else 1
+        )
         key_arrays = []
+        offset_arrays = []
         slot_ctxs: List[_SlotCtx] = []
         pos = 0
         for feat in feats:
             cfg = self.schema.get_slot(feat.name)
             keys, offsets = self._prepare_slot_keys(feat)
             key_arrays.append(keys)
-            sc = _SlotCtx(
-                name=feat.name,
-                cfg=cfg,
-                pos_slice=(pos, pos + len(keys)),
-                seg_offsets=torch.from_numpy(np.ascontiguousarray(offsets)).to(dev),
-            )
+            offset_arrays.append(offsets)
+            sc = _SlotCtx(name=feat.name, cfg=cfg, pos_slice=(pos, pos + len(keys)),
+                          seg_offsets=None)
             slot_ctxs.append(sc)
             pos += len(keys)
 
         all_keys = np.concatenate(key_arrays) if key_arrays else np.empty(0, np.uint64)
-        keys_t = torch.from_numpy(all_keys.view(np.int64)).to(dev)
-        flipped = keys_t ^ _FLIP
-        uniq_flipped, inverse = torch.unique(flipped, sorted=True, return_inverse=True)
-        uniq_keys = uniq_flipped ^ _FLIP
+        # single H2D upload of the whole group's keys + all offsets
+        keys_t = torch.from_numpy(all_keys.view(np.int64)).to(dev, non_blocking=False)
+        for sc, offs in zip(slot_ctxs, offset_arrays):
+            sc.seg_offsets = torch.from_numpy(np.ascontiguousarray(offs)).to(dev)
 
-        group = _GroupCtx(dim=dim, uniq_keys=uniq_keys, inverse=inverse, slots=slot_ctxs)
+        uniq_keys, inverse, perm, ustarts = _dedup(keys_t)
+        group = _GroupCtx(
+            dim=dim, uniq_keys=uniq_keys, inverse=inverse, perm=perm,
+            ustarts=ustarts, slots=slot_ctxs,
+        )
         rows = self._exchange_rows(group, train)  # [U, dim]
 
-        # --- postprocess into per-slot payloads
+        # ---- fused sum-slot postprocess
+        sum_slots = [sc for sc in slot_ctxs if sc.cfg.embedding_summation]
+        if sum_slots:
+            base = 0
+            offs_parts, sqrt_parts = [], []
+            for sc in sum_slots:
+                s, _e = sc.pos_slice
+                sc.sum_seg_base = base
+                nseg = sc.seg_offsets.numel() - 1
+                offs_parts.append(sc.seg_offsets[:-1] + s)
+                sqrt_parts.append(
+                    torch.full((nseg,), bool(sc.cfg.sqrt_scaling), dtype=torch.bool, device=dev)
+                )
+                base += nseg
+            sum_end = sum_slots[-1].pos_slice[1]
+            cat_offsets = torch.cat(
+                offs_parts + [torch.tensor([sum_end], dtype=torch.int64, device=dev)]
+            )
+            group.cat_offsets = cat_offsets
+            group.sqrt_mask = torch.cat(sqrt_parts)
+            lens = (cat_offsets[1:] - cat_offsets[:-1]).float()
+            group.seg_lens = lens
+            n_sum_segs = base
+            seg_id = torch.full((inverse.numel(),), -1, dtype=torch.int64, device=dev)
+            seg_id[:sum_end] = torch.repeat_interleave(
+                torch.arange(n_sum_segs, dtype=torch.int64, device=dev),
+                (cat_offsets[1:] - cat_offsets[:-1]),
+            )
+            group.seg_id = seg_id
+            fwd_scale = torch.where(
+                group.sqrt_mask, lens.clamp(min=1.0).rsqrt(), torch.ones_like(lens)
+            )
+            if native:
+                sums = C.segment_sum(rows.contiguous(), inverse, cat_offsets, fwd_scale)
+            else:
+                sums = (
+                    R.segment_sum_rows(rows, inverse[:sum_end], cat_offsets, False, torch.float32)
+                    * fwd_scale.unsqueeze(1)
+                ).to(torch.float16)
+            b0 = 0
+            for sc in sum_slots:
+                nseg = sc.seg_offsets.numel() - 1
+                p = SlotPayload(name=sc.name, cfg=sc.cfg, sum_tensor=sums[b0 : b0 + nseg])
+                out.payloads.append(p)
+                b0 += nseg
+
+        # ---- raw slots (torch path both backends: not in the flagship loop)
         for sc in slot_ctxs:
+            if sc.cfg.embedding_summation:
+                continue
             s, e = sc.pos_slice
             inv_slot = inverse[s:e]
-            payload = SlotPayload(name=sc.name, cfg=sc.cfg)
-            if sc.cfg.embedding_summation:
-                payload.sum_tensor = R.segment_sum_rows(
-                    rows, inv_slot, sc.seg_offsets, sc.cfg.sqrt_scaling, torch.float16
+            slot_uniq, slot_inv = torch.unique(inv_slot, sorted=True, return_inverse=True)
+            sc.slot_uniq_global = slot_uniq
+            sc.slot_inverse = slot_inv
+            scale = 1.0
+            rounds = sc.cfg.hash_stack_rounds
+            if sc.cfg.sqrt_scaling and rounds > 1:
+                scale = 1.0 / float(np.sqrt(rounds))
+            distinct, index, non_empty, num = R.raw_embedding_tensors(
+                rows[slot_uniq], slot_inv, sc.seg_offsets, sc.cfg.sample_fixed_size,
+                scale, torch.float16,
+            )
+            out.payloads.append(
+                SlotPayload(
+                    name=sc.name, cfg=sc.cfg, raw_distinct=distinct, raw_index=index,
+                    raw_non_empty_index=non_empty, raw_sample_id_num=num,
                 )
-            else:
-                slot_uniq, slot_inv = torch.unique(inv_slot, sorted=True, return_inverse=True)
-                sc.slot_uniq_global = slot_uniq
-                sc.slot_inverse = slot_inv
-                scale = 1.0
-                rounds = sc.cfg.hash_stack_rounds
-                if sc.cfg.sqrt_scaling and rounds > 1:
-                    scale = 1.0 / float(np.sqrt(rounds))
-                distinct, index, non_empty, num = R.raw_embedding_tensors(
-                    rows[slot_uniq],
-                    slot_inv,
-                    sc.seg_offsets,
-                    sc.cfg.sample_fixed_size,
-                    scale,
-                    torch.float16,
-                )
-                payload.raw_distinct = distinct
-                payload.raw_index = index
-                payload.raw_non_empty_index = non_empty
-                payload.raw_sample_id_num = num
-            out.payloads.append(payload)
+            )
         return group
 
     # --------------------------------------------------------- backward path
@@ -294,14 +384,53 @@ class EmbeddingEngine:
         """grads: slot name -> grad tensor ((B,dim) f16 for sum slots,
         (U_slot, dim) f32 for raw slots — the contract of ctx._on_backward,
         reference persia/ctx.py:926-1005) or None (skipped)."""
+        native = self.device.type == "cuda"
+        if native:
+            from persia_amd.ops import native as _native
+
+            C = _native()
         for group in training_batch._groups:
             U = group.uniq_keys.numel()
             buf = torch.zeros(U, group.dim, dtype=torch.float32, device=self.device)
             any_grad = False
+            sum_slots = [sc for sc in group.slots if sc.cfg.embedding_summation]
+            if native and sum_slots:
+                # fused ordered scatter over ALL sum slots in one launch.
+                # Per-slot NaN skip (reference mod.rs:731-746) without a host
+                # sync: zero the slot's per-segment scale via a device flag
+                # and sanitize the grads (NaN*0 would still be NaN).
+                g_parts, flag_parts = [], []
+                B = training_batch.batch_size
+                for sc in sum_slots:
+                    g = grads.get(sc.name)
+                    if g is None:
+                        g = torch.zeros(B, group.dim, dtype=torch.float16, device=self.device)
+                        flag = torch.ones((), dtype=torch.bool, device=self.device)
+                    else:
+                        any_grad = True
+                        flag = torch.isnan(g).any()
+                    g_parts.append(g.to(torch.float16))
+                    flag_parts.append(flag.expand(sc.seg_offsets.numel() - 1))
+                grads_cat = torch.nan_to_num(torch.cat(g_parts, dim=0))
+                skip = torch.cat(flag_parts)
+                lens = group.seg_lens
+                scale = torch.where(
+                    group.sqrt_mask, lens.clamp(min=1.0).rsqrt(), torch.ones_like(lens)
+                )
+                if loss_scale != 1.0:
+                    scale = scale / loss_scale
+                scale = torch.where(skip, torch.zeros_like(scale), scale)
+                C.grad_scatter(
+                    grads_cat.contiguous(), group.perm, group.ustarts,
+                    group.seg_id, scale.contiguous(), buf,
+                )
+                any_grad = True  # kernels launched regardless (no host sync)
             for sc in group.slots:
                 g = grads.get(sc.name)
                 if g is None:
                     continue
+                if native and sc.cfg.embedding_summation:
+                    continue  # handled by the fused scatter above
                 # NaN filter (reference mod.rs:731-746: skip the whole slot)
                 if bool(torch.isnan(g).any()):
                     self.nan_grad_batches += 1

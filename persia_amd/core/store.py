@@ -289,6 +289,7 @@ class HipEmbeddingStore(EmbeddingStoreBase):
         self.arena = torch.zeros(
             self.n_slots, self.row_width, dtype=torch.float32, device=device
         )
+        self._skipped = torch.zeros(1, dtype=torch.int32, device=device)
         self._opt_code = {"sgd": 0, "adagrad": 1, "adam": 2}[optimizer.kind]
 
     def _opt_params(self):
@@ -309,7 +310,7 @@ class HipEmbeddingStore(EmbeddingStoreBase):
         raise ValueError(o.kind)
 
     def __len__(self) -> int:
-        return int(self._C.store_count(self.keys))
+        return int((self.keys != 0).sum().item())
 
     def lookup(self, keys: torch.Tensor, train: bool) -> torch.Tensor:
         tick = self.next_tick()
@@ -335,20 +336,24 @@ class HipEmbeddingStore(EmbeddingStoreBase):
     def update_gradients(self, keys: torch.Tensor, grads: torch.Tensor) -> int:
         powers = self._adam_step_powers()
         b1p, b2p = powers if powers else (0.0, 0.0)
-        skipped = self._C.store_update(
+        self._C.store_update(
             self.keys,
             self.ticks,
             self.arena,
             keys,
-            grads.float(),
+            grads.float().contiguous(),
             self.dim,
             self._opt_code,
             self._opt_params(),
             float(b1p),
             float(b2p),
             float(self.hyper.weight_bound),
+            self._skipped,
         )
-        return int(skipped)
+        return 0  # skipped count is accumulated device-side (no hot-path sync)
+
+    def skipped_count(self) -> int:
+        return int(self._skipped.item())
 
     def export_rows(self) -> Tuple[np.ndarray, np.ndarray]:
         occ = torch.nonzero(self.keys != 0, as_tuple=False).view(-1)

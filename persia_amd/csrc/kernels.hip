@@ -1,0 +1,475 @@
+// persia_amd HIP kernels (gfx950 / CDNA4, wave64).
+//
+// MI355X-native replacements for the reference's CPU hot loops:
+//  * set-associative HBM hash table  <- EvictionMap LRU + per-sign locks
+//    (rust/persia-embedding-holder/src/eviction_map.rs,
+//     embedding_parameter_service/mod.rs:162-262)
+//  * fused gather + segment-sum      <- add_assign_avx2 summation
+//    (embedding_worker_service/mod.rs:486-629, persia-simd lib.rs:4-18)
+//  * ordered gradient scatter        <- per-sign HashMap grad accumulation
+//    (embedding_worker_service/mod.rs:782-814)
+//  * fused row-wise sparse optimizers + weight bound
+//    (persia-simd lib.rs:21-251, exact-math variants)
+//
+// Layout: keys u64[n_slots] (0 = empty), ticks u32[n_slots], arena
+// f32[n_slots, row_width], row = [emb(dim) | opt_state].  Buckets of 8
+// slots (one 64B cache line of keys); probe scans 4 consecutive buckets.
+//
+// Concurrency contract (matches the engine's stream discipline):
+//  * keys within one lookup/update call are unique (deduped upstream);
+//  * lookup calls are serialized on one stream (ForwardPipeline);
+//  * an update may overlap a lookup; claims use atomicCAS on the key word,
+//    so the only hazard is an eviction racing an in-flight update of the
+//    evicted row — bounded-staleness noise the PERSIA algorithm tolerates
+//    (SURVEY §7 hard part 2); it cannot corrupt the table structure.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#include "common.h"
+
+using ull = unsigned long long;
+
+namespace {
+
+__device__ __forceinline__ float pa_to_float(float x) { return x; }
+__device__ __forceinline__ float pa_to_float(__half x) { return __half2float(x); }
+
+// ---------------------------------------------------------------- store probe
+
+// Find the slot holding key k, or -1.  Bucket loads hit one cache line.
+__device__ __forceinline__ long long probe_find(const ull* __restrict__ keys,
+                                                int64_t n_buckets, ull k) {
+  const int64_t mask = n_buckets - 1;
+  int64_t b = (int64_t)(k & (ull)mask);
+  for (int p = 0; p < PA_PROBE_BUCKETS; ++p) {
+    const int64_t base = ((b + p) & mask) * PA_BUCKET_SIZE;
+#pragma unroll
+    for (int s = 0; s < PA_BUCKET_SIZE; ++s) {
+      if (keys[base + s] == k) return base + s;
+    }
+  }
+  return -1;
+}
+
+// Probe-or-claim with bounded-window LRU eviction (store.py _probe_or_claim).
+// Returns slot; *is_new = 1 when this thread claimed it (must init the row).
+__device__ long long probe_claim(ull* __restrict__ keys,
+                                 const unsigned* __restrict__ ticks,
+                                 int64_t n_buckets, ull k, int* is_new) {
+  const int64_t mask = n_buckets - 1;
+  const int64_t b = (int64_t)(k & (ull)mask);
+  for (int attempt = 0; attempt < 16; ++attempt) {
+    long long empty = -1;
+    long long victim = -1;
+    unsigned victim_tick = 0xFFFFFFFFu;
+    ull victim_key = 0;
+    for (int p = 0; p < PA_PROBE_BUCKETS; ++p) {
+      const int64_t base = ((b + p) & mask) * PA_BUCKET_SIZE;
+#pragma unroll
+      for (int s = 0; s < PA_BUCKET_SIZE; ++s) {
+        const int64_t i = base + s;
+        const ull ki = keys[i];
+        if (ki == k) { *is_new = 0; return i; }
+        if (ki == PA_EMPTY_KEY) {
+          if (empty < 0) empty = i;
+        } else if (victim < 0 || ticks[i] < victim_tick) {
+          victim = i; victim_tick = ticks[i]; victim_key = ki;
+        }
+      }
+    }
+    if (empty >= 0) {
+      if (atomicCAS((ull*)&keys[empty], PA_EMPTY_KEY, k) == PA_EMPTY_KEY) {
+        *is_new = 1; return empty;
+      }
+      continue;  // lost the race; rescan
+    }
+    if (victim >= 0) {
+      if (atomicCAS((ull*)&keys[victim], victim_key, k) == victim_key) {
+        *is_new = 1; return victim;
+      }
+      continue;
+    }
+  }
+  *is_new = 0;
+  return -1;  // give up: treated as a non-admitted miss
+}
+
+__global__ void probe_claim_kernel(ull* __restrict__ table_keys,
+                                   unsigned* __restrict__ ticks,
+                                   const ull* __restrict__ query,
+                                   int64_t n, int64_t n_buckets, int train,
+                                   unsigned tick, float admit_prob,
+                                   long long* __restrict__ out_slot,
+                                   int* __restrict__ out_new) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const ull k = query[i];
+  long long slot;
+  int is_new = 0;
+  if (train) {
+    bool admitted = true;
+    if (admit_prob < 1.0f) {
+      // stochastic admission re-rolled per tick (store.py _admitted)
+      const double u = pa_u01(pa_splitmix64(k ^ (ull)tick));
+      admitted = u < (double)admit_prob;
+    }
+    if (admitted) {
+      slot = probe_claim(table_keys, ticks, n_buckets, k, &is_new);
+    } else {
+      slot = probe_find(table_keys, n_buckets, k);
+    }
+  } else {
+    slot = probe_find(table_keys, n_buckets, k);
+  }
+  if (slot >= 0) ticks[slot] = tick;
+  out_slot[i] = slot;
+  out_new[i] = is_new;
+}
+
+// one wave per key: init freshly claimed rows, then gather emb -> out (f32)
+__global__ void init_gather_kernel(float* __restrict__ arena,
+                                   const ull* __restrict__ query,
+                                   const long long* __restrict__ slots,
+                                   const int* __restrict__ is_new,
+                                   float* __restrict__ out, int64_t n, int dim,
+                                   int row_width, double lo, double hi,
+                                   float state_init) {
+  const int wave = threadIdx.x / PA_WAVE;
+  const int lane = threadIdx.x % PA_WAVE;
+  const int waves_per_block = blockDim.x / PA_WAVE;
+  for (int64_t i = (int64_t)blockIdx.x * waves_per_block + wave; i < n;
+       i += (int64_t)gridDim.x * waves_per_block) {
+    const long long slot = slots[i];
+    float* dst = out + i * dim;
+    if (slot < 0) {
+      for (int c = lane; c < dim; c += PA_WAVE) dst[c] = 0.0f;
+      continue;
+    }
+    float* row = arena + (int64_t)slot * row_width;
+    if (is_new[i]) {
+      const uint64_t sign = pa_splitmix64_inv((uint64_t)query[i]);
+      const uint64_t seed = pa_init_seed(sign);
+      for (int c = lane; c < dim; c += PA_WAVE)
+        row[c] = pa_init_val(seed, c, lo, hi);
+      for (int c = dim + lane; c < row_width; c += PA_WAVE)
+        row[c] = state_init;
+    }
+    for (int c = lane; c < dim; c += PA_WAVE) dst[c] = row[c];
+  }
+}
+
+// ------------------------------------------------------------- sparse update
+
+// one wave per key: probe (lane-parallel over the 32-slot window) + fused
+// row-wise optimizer + weight bound.
+// opt: 0 = SGD, 1 = Adagrad, 2 = Adam.   params: see store.py _opt_params.
+__global__ void update_kernel(ull* __restrict__ table_keys,
+                              unsigned* __restrict__ ticks,
+                              float* __restrict__ arena,
+                              const ull* __restrict__ query,
+                              const float* __restrict__ grads, int64_t n,
+                              int dim, int row_width, int64_t n_buckets,
+                              int opt, float p0, float p1, float p2, float p3,
+                              float b1_power, float b2_power,
+                              float weight_bound,
+                              int* __restrict__ skipped) {
+  const int wave = threadIdx.x / PA_WAVE;
+  const int lane = threadIdx.x % PA_WAVE;
+  const int waves_per_block = blockDim.x / PA_WAVE;
+  const int64_t mask = n_buckets - 1;
+  for (int64_t i = (int64_t)blockIdx.x * waves_per_block + wave; i < n;
+       i += (int64_t)gridDim.x * waves_per_block) {
+    const ull k = query[i];
+    // lane-parallel probe: lanes 0..31 each inspect one slot of the window
+    long long slot = -1;
+    const int64_t b = (int64_t)(k & (ull)mask);
+    if (lane < PA_PROBE_BUCKETS * PA_BUCKET_SIZE) {
+      const int p = lane / PA_BUCKET_SIZE;
+      const int s = lane % PA_BUCKET_SIZE;
+      const int64_t j = ((b + p) & mask) * PA_BUCKET_SIZE + s;
+      if (table_keys[j] == k) slot = j;
+    }
+    const unsigned long long found = __ballot(slot >= 0);
+    if (found == 0) {
+      if (lane == 0) atomicAdd(skipped, 1);
+      continue;
+    }
+    const int src = __ffsll((long long)found) - 1;
+    slot = __shfl(slot, src);
+    float* row = arena + (int64_t)slot * row_width;
+    const float* g = grads + i * dim;
+    if (opt == 0) {  // SGD: w -= lr*(g + wd*w)         p0=lr p1=wd
+      for (int c = lane; c < dim; c += PA_WAVE) {
+        float w = row[c] - p0 * (g[c] + p1 * row[c]);
+        if (weight_bound > 0.0f) w = fminf(fmaxf(w, -weight_bound), weight_bound);
+        row[c] = w;
+      }
+    } else if (opt == 1) {  // Adagrad  p0=lr p1=g_square_momentum p2=eps p3=vectorwise
+      if (p3 > 0.5f) {
+        // shared scalar accumulator at row[dim]
+        const float acc = row[dim];
+        float gsq = 0.0f;
+        for (int c = lane; c < dim; c += PA_WAVE) {
+          float w = row[c] - p0 * g[c] * rsqrtf(acc + p2);
+          if (weight_bound > 0.0f) w = fminf(fmaxf(w, -weight_bound), weight_bound);
+          row[c] = w;
+          gsq += g[c] * g[c];
+        }
+#pragma unroll
+        for (int off = PA_WAVE / 2; off > 0; off >>= 1)
+          gsq += __shfl_down(gsq, off);
+        if (lane == 0) row[dim] = acc * p1 + gsq / (float)dim;
+      } else {
+        for (int c = lane; c < dim; c += PA_WAVE) {
+          const float acc = row[dim + c];
+          float w = row[c] - p0 * g[c] * rsqrtf(acc + p2);
+          if (weight_bound > 0.0f) w = fminf(fmaxf(w, -weight_bound), weight_bound);
+          row[c] = w;
+          row[dim + c] = acc * p1 + g[c] * g[c];
+        }
+      }
+    } else {  // Adam  p0=lr p1=beta1 p2=beta2 p3=eps
+      const float om1 = 1.0f - p1, om2 = 1.0f - p2;
+      const float c1 = 1.0f / (1.0f - b1_power), c2 = 1.0f / (1.0f - b2_power);
+      for (int c = lane; c < dim; c += PA_WAVE) {
+        const float m = p1 * row[dim + c] + om1 * g[c];
+        const float v = p2 * row[2 * dim + c] + om2 * g[c] * g[c];
+        float w = row[c] - p0 * (m * c1) / (p3 + sqrtf(v * c2));
+        if (weight_bound > 0.0f) w = fminf(fmaxf(w, -weight_bound), weight_bound);
+        row[c] = w;
+        row[dim + c] = m;
+        row[2 * dim + c] = v;
+      }
+    }
+  }
+}
+
+// --------------------------------------------------------------- import rows
+
+__global__ void import_kernel(ull* __restrict__ table_keys,
+                              unsigned* __restrict__ ticks,
+                              float* __restrict__ arena,
+                              const ull* __restrict__ query,
+                              const float* __restrict__ rows, int64_t n,
+                              int row_width, int64_t n_buckets, unsigned tick) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  int is_new;
+  const long long slot = probe_claim(table_keys, ticks, n_buckets, query[i], &is_new);
+  if (slot < 0) return;
+  ticks[slot] = tick;
+  float* dst = arena + (int64_t)slot * row_width;
+  const float* src = rows + i * row_width;
+  for (int c = 0; c < row_width; ++c) dst[c] = src[c];
+}
+
+// ----------------------------------------------------- fused gather + seg-sum
+
+// out[s, :] = seg_scale[s] * sum_{k in [off[s], off[s+1])} rows[inverse[k], :]
+// One wave per segment (all sum-slots of a dim-group in ONE launch);
+// f32 accumulate; f16 store (reference wire dtype, persia-common lib.rs:88-99).
+template <typename RowT>
+__global__ void segment_sum_kernel(const RowT* __restrict__ rows,
+                                   const int64_t* __restrict__ inverse,
+                                   const int64_t* __restrict__ seg_offsets,
+                                   const float* __restrict__ seg_scale,
+                                   __half* __restrict__ out, int64_t n_seg,
+                                   int dim) {
+  const int wave = threadIdx.x / PA_WAVE;
+  const int lane = threadIdx.x % PA_WAVE;
+  const int waves_per_block = blockDim.x / PA_WAVE;
+  for (int64_t s = (int64_t)blockIdx.x * waves_per_block + wave; s < n_seg;
+       s += (int64_t)gridDim.x * waves_per_block) {
+    const int64_t lo = seg_offsets[s], hi = seg_offsets[s + 1];
+    const float scale = seg_scale ? seg_scale[s] : 1.0f;
+    __half* dst = out + s * dim;
+    for (int c = lane; c < dim; c += PA_WAVE) {
+      float acc = 0.0f;
+      for (int64_t k = lo; k < hi; ++k) {
+        acc += pa_to_float(rows[inverse[k] * dim + c]);
+      }
+      dst[c] = __float2half(acc * scale);
+    }
+  }
+}
+
+// --------------------------------------------------- ordered gradient scatter
+
+// out[u, :] += sum_{p in [ustart[u], ustart[u+1])} grads[seg_id[perm[p]], :]
+//              * seg_scale[seg_id[perm[p]]]
+// One wave per unique sign — deterministic (ordered), atomic-free: the
+// positions of a unique key are contiguous in sort order.  Positions with
+// seg_id < 0 (raw-slot positions, handled by a separate indexed add) are
+// skipped.
+template <typename GradT>
+__global__ void grad_scatter_kernel(const GradT* __restrict__ grads,
+                                    const int64_t* __restrict__ perm,
+                                    const int64_t* __restrict__ ustarts,
+                                    const int64_t* __restrict__ seg_id,
+                                    const float* __restrict__ seg_scale,
+                                    float* __restrict__ out, int64_t n_unique,
+                                    int dim) {
+  const int wave = threadIdx.x / PA_WAVE;
+  const int lane = threadIdx.x % PA_WAVE;
+  const int waves_per_block = blockDim.x / PA_WAVE;
+  for (int64_t u = (int64_t)blockIdx.x * waves_per_block + wave; u < n_unique;
+       u += (int64_t)gridDim.x * waves_per_block) {
+    const int64_t lo = ustarts[u], hi = ustarts[u + 1];
+    float* dst = out + u * dim;
+    for (int c = lane; c < dim; c += PA_WAVE) {
+      float acc = 0.0f;
+      for (int64_t p = lo; p < hi; ++p) {
+        const int64_t s = seg_id[perm[p]];
+        if (s >= 0) acc += pa_to_float(grads[s * dim + c]) * seg_scale[s];
+      }
+      dst[c] += acc;
+    }
+  }
+}
+
+inline int n_blocks_for(int64_t work_items, int per_block) {
+  int64_t b = (work_items + per_block - 1) / per_block;
+  // >> 256 CUs needed to fill the chip; cap and grid-stride beyond
+  if (b > 16384) b = 16384;
+  if (b < 1) b = 1;
+  return (int)b;
+}
+
+}  // namespace
+
+// ============================================================ torch bindings
+
+static hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+void store_lookup(torch::Tensor table_keys, torch::Tensor ticks,
+                  torch::Tensor arena, torch::Tensor query, torch::Tensor out,
+                  int64_t dim, int64_t train, int64_t tick, double lo,
+                  double hi, double admit_prob, double state_init,
+                  int64_t opt_space) {
+  const int64_t n = query.numel();
+  if (n == 0) return;
+  const int64_t n_buckets = table_keys.numel() / PA_BUCKET_SIZE;
+  const int row_width = (int)(dim + opt_space);
+  auto opts = torch::TensorOptions().dtype(torch::kInt64).device(query.device());
+  auto slots = torch::empty({n}, opts);
+  auto is_new = torch::empty({n}, opts.dtype(torch::kInt32));
+  hipStream_t st = cur_stream();
+  hipLaunchKernelGGL(probe_claim_kernel, dim3(n_blocks_for(n, 256)), dim3(256),
+                     0, st, (ull*)table_keys.data_ptr<int64_t>(),
+                     (unsigned*)ticks.data_ptr<int32_t>(),
+                     (const ull*)query.data_ptr<int64_t>(), n, n_buckets,
+                     (int)train, (unsigned)tick, (float)admit_prob,
+                     (long long*)slots.data_ptr<int64_t>(),
+                     is_new.data_ptr<int32_t>());
+  hipLaunchKernelGGL(init_gather_kernel, dim3(n_blocks_for(n, 4)), dim3(256),
+                     0, st, arena.data_ptr<float>(),
+                     (const ull*)query.data_ptr<int64_t>(),
+                     (const long long*)slots.data_ptr<int64_t>(),
+                     is_new.data_ptr<int32_t>(), out.data_ptr<float>(), n,
+                     (int)dim, row_width, lo, hi, (float)state_init);
+}
+
+void store_update(torch::Tensor table_keys, torch::Tensor ticks,
+                  torch::Tensor arena, torch::Tensor query,
+                  torch::Tensor grads, int64_t dim, int64_t opt,
+                  std::vector<double> params, double b1_power,
+                  double b2_power, double weight_bound,
+                  torch::Tensor skipped /* persistent i32[1], no sync */) {
+  const int64_t n = query.numel();
+  if (n == 0) return;
+  const int64_t n_buckets = table_keys.numel() / PA_BUCKET_SIZE;
+  const int row_width = (int)arena.size(1);
+  hipStream_t st = cur_stream();
+  hipLaunchKernelGGL(update_kernel, dim3(n_blocks_for(n, 4)), dim3(256), 0, st,
+                     (ull*)table_keys.data_ptr<int64_t>(),
+                     (unsigned*)ticks.data_ptr<int32_t>(),
+                     arena.data_ptr<float>(),
+                     (const ull*)query.data_ptr<int64_t>(),
+                     grads.data_ptr<float>(), n, (int)dim, row_width,
+                     n_buckets, (int)opt, (float)params[0], (float)params[1],
+                     (float)params[2], (float)params[3], (float)b1_power,
+                     (float)b2_power, (float)weight_bound,
+                     skipped.data_ptr<int32_t>());
+}
+
+void store_import(torch::Tensor table_keys, torch::Tensor ticks,
+                  torch::Tensor arena, torch::Tensor query, torch::Tensor rows,
+                  int64_t tick) {
+  const int64_t n = query.numel();
+  if (n == 0) return;
+  const int64_t n_buckets = table_keys.numel() / PA_BUCKET_SIZE;
+  hipLaunchKernelGGL(import_kernel, dim3(n_blocks_for(n, 256)), dim3(256), 0,
+                     cur_stream(), (ull*)table_keys.data_ptr<int64_t>(),
+                     (unsigned*)ticks.data_ptr<int32_t>(),
+                     arena.data_ptr<float>(),
+                     (const ull*)query.data_ptr<int64_t>(),
+                     rows.data_ptr<float>(), n, (int)rows.size(1), n_buckets,
+                     (unsigned)tick);
+}
+
+torch::Tensor segment_sum(torch::Tensor rows, torch::Tensor inverse,
+                          torch::Tensor seg_offsets, torch::Tensor seg_scale) {
+  const int64_t n_seg = seg_offsets.numel() - 1;
+  const int64_t dim = rows.size(1);
+  auto out = torch::empty(
+      {n_seg, dim},
+      torch::TensorOptions().dtype(torch::kFloat16).device(rows.device()));
+  if (n_seg == 0) return out;
+  const float* scale_ptr =
+      seg_scale.numel() ? seg_scale.data_ptr<float>() : nullptr;
+  hipStream_t st = cur_stream();
+  const dim3 grid(n_blocks_for(n_seg, 4)), block(256);
+  if (rows.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL(segment_sum_kernel<float>, grid, block, 0, st,
+                       rows.data_ptr<float>(), inverse.data_ptr<int64_t>(),
+                       seg_offsets.data_ptr<int64_t>(), scale_ptr,
+                       (__half*)out.data_ptr<at::Half>(), n_seg, (int)dim);
+  } else if (rows.scalar_type() == torch::kFloat16) {
+    hipLaunchKernelGGL(segment_sum_kernel<__half>, grid, block, 0, st,
+                       (const __half*)rows.data_ptr<at::Half>(),
+                       inverse.data_ptr<int64_t>(),
+                       seg_offsets.data_ptr<int64_t>(), scale_ptr,
+                       (__half*)out.data_ptr<at::Half>(), n_seg, (int)dim);
+  } else {
+    TORCH_CHECK(false, "segment_sum: rows must be f32 or f16");
+  }
+  return out;
+}
+
+void grad_scatter(torch::Tensor grads, torch::Tensor perm,
+                  torch::Tensor ustarts, torch::Tensor seg_id,
+                  torch::Tensor seg_scale, torch::Tensor out) {
+  const int64_t n_unique = ustarts.numel() - 1;
+  const int64_t dim = out.size(1);
+  if (n_unique == 0) return;
+  hipStream_t st = cur_stream();
+  const dim3 grid(n_blocks_for(n_unique, 4)), block(256);
+  if (grads.scalar_type() == torch::kFloat16) {
+    hipLaunchKernelGGL(grad_scatter_kernel<__half>, grid, block, 0, st,
+                       (const __half*)grads.data_ptr<at::Half>(),
+                       perm.data_ptr<int64_t>(), ustarts.data_ptr<int64_t>(),
+                       seg_id.data_ptr<int64_t>(), seg_scale.data_ptr<float>(),
+                       out.data_ptr<float>(), n_unique, (int)dim);
+  } else if (grads.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL(grad_scatter_kernel<float>, grid, block, 0, st,
+                       grads.data_ptr<float>(), perm.data_ptr<int64_t>(),
+                       ustarts.data_ptr<int64_t>(), seg_id.data_ptr<int64_t>(),
+                       seg_scale.data_ptr<float>(), out.data_ptr<float>(),
+                       n_unique, (int)dim);
+  } else {
+    TORCH_CHECK(false, "grad_scatter: grads must be f16 or f32");
+  }
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "persia_amd HIP kernels (gfx950)";
+  m.def("store_lookup", &store_lookup, "hash-table lookup/insert + gather");
+  m.def("store_update", &store_update, "fused sparse optimizer update");
+  m.def("store_import", &store_import, "bulk insert rows (checkpoint load)");
+  m.def("segment_sum", &segment_sum, "fused gather + per-sample summation");
+  m.def("grad_scatter", &grad_scatter, "ordered per-sign gradient scatter");
+}
