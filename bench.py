@@ -23,8 +23,8 @@ import torch
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=20)
-    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--steps", type=int, default=50)
+    p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--batch-size", type=int, default=4096, help="per-GPU batch")
     p.add_argument("--num-sparse", type=int, default=26)
     p.add_argument("--num-dense", type=int, default=13)
@@ -142,17 +142,16 @@ def main():
         pipeline.release_permit()
         return loss
 
-    def feed(n):
-        for i in range(n):
-            pipeline.put(host_batches[i % len(host_batches)])
-
     import threading
 
-    feeder = threading.Thread(
-        target=feed, args=(args.steps + args.warmup,), daemon=True
-    )
-    feeder.start()
+    def feed(start, n):
+        for i in range(start, start + n):
+            pipeline.put(host_batches[i % len(host_batches)])
 
+    # warmup: feed + consume, then drain so NO timed batch has its sparse work
+    # pre-done before t0 (the staleness window only overlaps WITHIN the timed
+    # region — keeps the measurement honest for any step count)
+    threading.Thread(target=feed, args=(0, args.warmup), daemon=True).start()
     for _ in range(args.warmup):
         train_step(pipeline.get())
 
@@ -163,6 +162,8 @@ def main():
     if use_gpu:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
+    feeder = threading.Thread(target=feed, args=(args.warmup, args.steps), daemon=True)
+    feeder.start()
     for _ in range(args.steps):
         train_step(pipeline.get())
     if world > 1:

@@ -151,7 +151,9 @@ class CpuEmbeddingStore(EmbeddingStoreBase):
 
     def _probe_or_claim(self, k: np.uint64, tick: int) -> Tuple[int, bool]:
         """-> (slot, is_new). Claims an empty slot, else evicts the min-tick
-        slot of the probe window. Deterministic given serialized access."""
+        slot of the probe window; rows touched at the CURRENT tick are never
+        victims (same rule as the HIP kernel, so concurrent claims within a
+        batch cannot evict each other). slot=-1 = overflow miss."""
         mask = np.uint64(self.n_buckets - 1)
         b0 = int(k & mask)
         empty = -1
@@ -166,13 +168,19 @@ class CpuEmbeddingStore(EmbeddingStoreBase):
                 if ki == EMPTY_KEY:
                     if empty < 0:
                         empty = i
-                elif victim_tick is None or self.ticks[i] < victim_tick:
+                elif self.ticks[i] != tick and (
+                    victim_tick is None or self.ticks[i] < victim_tick
+                ):
                     victim, victim_tick = i, self.ticks[i]
         if empty >= 0:
             self.keys[empty] = k
+            self.ticks[empty] = tick
             self._count += 1
             return empty, True
+        if victim < 0:
+            return -1, False  # whole window is current-tick: overflow
         self.keys[victim] = k  # evict (bounded-window LRU)
+        self.ticks[victim] = tick
         return victim, True
 
     def _init_row(self, slot: int, k: np.uint64) -> None:
@@ -201,6 +209,8 @@ class CpuEmbeddingStore(EmbeddingStoreBase):
             if train:
                 if self._admitted(k, tick) or self._probe(k) >= 0:
                     slot, is_new = self._probe_or_claim(k, tick)
+                    if slot < 0:
+                        continue  # overflow miss: zeros
                     if is_new:
                         self._init_row(slot, k)
                     self.ticks[slot] = tick
@@ -261,6 +271,8 @@ class CpuEmbeddingStore(EmbeddingStoreBase):
         tick = self.next_tick()
         for i, k in enumerate(ks):
             slot, _ = self._probe_or_claim(k, tick)
+            if slot < 0:
+                continue  # shard over capacity: drop (bounded table)
             self.arena[slot] = torch.from_numpy(inner[i].astype(np.float32))
             self.ticks[slot] = tick
 

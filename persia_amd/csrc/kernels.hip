@@ -55,9 +55,13 @@ __device__ __forceinline__ long long probe_find(const ull* __restrict__ keys,
 
 // Probe-or-claim with bounded-window LRU eviction (store.py _probe_or_claim).
 // Returns slot; *is_new = 1 when this thread claimed it (must init the row).
+// Rows touched at the CURRENT tick are never eviction victims, so concurrent
+// claims within one batch cannot evict each other; when the whole window is
+// current-tick the key overflows to a miss (bounded churn).
 __device__ long long probe_claim(ull* __restrict__ keys,
-                                 const unsigned* __restrict__ ticks,
-                                 int64_t n_buckets, ull k, int* is_new) {
+                                 unsigned* __restrict__ ticks,
+                                 int64_t n_buckets, ull k, unsigned cur_tick,
+                                 int* is_new) {
   const int64_t mask = n_buckets - 1;
   const int64_t b = (int64_t)(k & (ull)mask);
   for (int attempt = 0; attempt < 16; ++attempt) {
@@ -74,23 +78,29 @@ __device__ long long probe_claim(ull* __restrict__ keys,
         if (ki == k) { *is_new = 0; return i; }
         if (ki == PA_EMPTY_KEY) {
           if (empty < 0) empty = i;
-        } else if (victim < 0 || ticks[i] < victim_tick) {
-          victim = i; victim_tick = ticks[i]; victim_key = ki;
+        } else {
+          const unsigned t = ticks[i];
+          if (t != cur_tick && (victim < 0 || t < victim_tick)) {
+            victim = i; victim_tick = t; victim_key = ki;
+          }
         }
       }
     }
     if (empty >= 0) {
       if (atomicCAS((ull*)&keys[empty], PA_EMPTY_KEY, k) == PA_EMPTY_KEY) {
+        ticks[empty] = cur_tick;  // publish before peers pick victims
         *is_new = 1; return empty;
       }
       continue;  // lost the race; rescan
     }
     if (victim >= 0) {
       if (atomicCAS((ull*)&keys[victim], victim_key, k) == victim_key) {
+        ticks[victim] = cur_tick;
         *is_new = 1; return victim;
       }
       continue;
     }
+    break;  // every slot is current-tick: overflow
   }
   *is_new = 0;
   return -1;  // give up: treated as a non-admitted miss
@@ -116,7 +126,7 @@ __global__ void probe_claim_kernel(ull* __restrict__ table_keys,
       admitted = u < (double)admit_prob;
     }
     if (admitted) {
-      slot = probe_claim(table_keys, ticks, n_buckets, k, &is_new);
+      slot = probe_claim(table_keys, ticks, n_buckets, k, tick, &is_new);
     } else {
       slot = probe_find(table_keys, n_buckets, k);
     }
@@ -257,7 +267,8 @@ __global__ void import_kernel(ull* __restrict__ table_keys,
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   int is_new;
-  const long long slot = probe_claim(table_keys, ticks, n_buckets, query[i], &is_new);
+  const long long slot =
+      probe_claim(table_keys, ticks, n_buckets, query[i], tick, &is_new);
   if (slot < 0) return;
   ticks[slot] = tick;
   float* dst = arena + (int64_t)slot * row_width;

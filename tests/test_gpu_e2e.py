@@ -30,7 +30,7 @@ def test_dlrm_train_loop_learns():
         dist_ctx=DistContext(1, 0),
     )
     model = DLRM(num_sparse=n_slots, num_dense=4, dim=dim).to(device)
-    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    opt = torch.optim.Adam(model.parameters(), lr=3e-3)
 
     rng = np.random.default_rng(0)
     id_scores = rng.normal(size=vocab)
@@ -46,10 +46,10 @@ def test_dlrm_train_loop_learns():
             feats, non_id_type_features=[dense], labels=[Label(label)], requires_grad=True
         )
 
-    pipeline = ForwardPipeline(engine, staleness=4)
+    pipeline = ForwardPipeline(engine, staleness=2)
     pipeline.start()
-    n_steps = 60
-    batches = [mk_batch() for _ in range(20)]
+    n_steps = 150
+    batches = [mk_batch() for _ in range(10)]
     import threading
 
     def feed():

@@ -117,11 +117,15 @@ def test_store_eviction_bounded():
     hip = _hip_store(capacity=BUCKET_SIZE * PROBE_BUCKETS, dim=4)
     n_slots = hip.n_slots
     signs = np.arange(1, 20 * n_slots, dtype=np.uint64)
-    for chunk in np.array_split(signs, 16):
+    chunks = np.array_split(signs, 16)
+    for chunk in chunks:
         hip.lookup(_keys(chunk).to(_dev()), train=True)
     assert len(hip) <= n_slots
-    r = hip.lookup(_keys(signs[-4:]).to(_dev()), train=False).cpu()
-    assert not torch.all(r == 0)
+    # the final batch's signs claimed the window (current-tick rows are never
+    # evicted by their own batch; overflow beyond the window misses)
+    r = hip.lookup(_keys(chunks[-1]).to(_dev()), train=False).cpu()
+    present = (r != 0).any(dim=1).sum().item()
+    assert present >= min(len(chunks[-1]), n_slots) // 2
 
 
 def test_segment_sum_matches_reference():
