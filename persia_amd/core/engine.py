@@ -275,13 +275,19 @@ class EmbeddingEngine:
         feats = sorted(
             feats, key=lambda f: 0 if self.schema.get_slot(f.name).embedding_summation else 1
         )
+        gpu_prep = native and not any(
+            self.schema.get_slot(f.name).hash_stack_rounds > 0 for f in feats
+        )
         key_arrays = []
         offset_arrays = []
         slot_ctxs: List[_SlotCtx] = []
         pos = 0
         for feat in feats:
             cfg = self.schema.get_slot(feat.name)
-            keys, offsets = self._prepare_slot_keys(feat)
+            if gpu_prep:
+                keys, offsets = feat.values, feat.offsets  # raw; mixed on-GPU
+            else:
+                keys, offsets = self._prepare_slot_keys(feat)
             key_arrays.append(keys)
             offset_arrays.append(offsets)
             sc = _SlotCtx(name=feat.name, cfg=cfg, pos_slice=(pos, pos + len(keys)),
@@ -290,10 +296,32 @@ class EmbeddingEngine:
             pos += len(keys)
 
         all_keys = np.concatenate(key_arrays) if key_arrays else np.empty(0, np.uint64)
-        # single H2D upload of the whole group's keys + all offsets
+        # single H2D upload of the whole group's values + all offsets
         keys_t = torch.from_numpy(all_keys.view(np.int64)).to(dev, non_blocking=False)
+        all_offs = np.concatenate(
+            [o for o in offset_arrays] or [np.zeros(1, np.int64)]
+        )
+        all_offs_t = torch.from_numpy(all_offs).to(dev)
+        o0 = 0
         for sc, offs in zip(slot_ctxs, offset_arrays):
-            sc.seg_offsets = torch.from_numpy(np.ascontiguousarray(offs)).to(dev)
+            sc.seg_offsets = all_offs_t[o0 : o0 + len(offs)]
+            o0 += len(offs)
+        if gpu_prep:
+            slot_starts = torch.tensor(
+                [sc.pos_slice[0] for sc in slot_ctxs] + [pos],
+                dtype=torch.int64, device=dev,
+            )
+            prefixes = torch.from_numpy(
+                np.array([sc.cfg.index_prefix for sc in slot_ctxs], dtype=np.uint64)
+                .view(np.int64)
+            ).to(dev)
+            spacing = self.schema.feature_spacing
+            from persia_amd.ops import native as _native2
+
+            keys_t = _native2().sign_prep(
+                keys_t, slot_starts, prefixes,
+                spacing if spacing < (1 << 63) else -1,
+            )
 
         uniq_keys, inverse, perm, ustarts = _dedup(keys_t)
         group = _GroupCtx(
@@ -575,4 +603,14 @@ class ForwardPipeline:
         self._sem.release()
 
     def stop(self):
+        """Deterministic shutdown: wake and join the worker so no background
+        thread is inside a HIP call at interpreter teardown."""
         self._stop.set()
+        self._sem.release()  # unblock a worker waiting on the staleness gate
+        try:
+            self._in.put_nowait(None)
+        except queue.Full:
+            pass
+        if self._thread is not None:
+            self._thread.join(timeout=5.0)
+            self._thread = None

@@ -256,6 +256,32 @@ __global__ void update_kernel(ull* __restrict__ table_keys,
   }
 }
 
+// ------------------------------------------------------------------ sign prep
+
+// raw ids -> routed keys: fold into the slot's feature-group prefix space
+// (indices_add_prefix, mod.rs:403-429) then splitmix64-mix (the mixed key is
+// the dedup sort key AND the shard router).  Slot of position i found by
+// binary search over slot_starts (S is small).
+__global__ void sign_prep_kernel(const ull* __restrict__ values,
+                                 const int64_t* __restrict__ slot_starts,
+                                 const ull* __restrict__ prefixes, int n_slots,
+                                 ull spacing, ull* __restrict__ out,
+                                 int64_t n) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  int lo = 0, hi = n_slots;  // find slot: slot_starts[s] <= i < slot_starts[s+1]
+  while (hi - lo > 1) {
+    const int mid = (lo + hi) >> 1;
+    if (i >= slot_starts[mid]) lo = mid; else hi = mid;
+  }
+  const ull prefix = prefixes[lo];
+  ull sign = values[i];
+  if (prefix != 0) sign = sign % spacing + prefix;
+  ull k = pa_splitmix64(sign);
+  if (k == PA_EMPTY_KEY) k = 0xD1B54A32D192ED03ull;  // empty-sentinel remap
+  out[i] = k;
+}
+
 // --------------------------------------------------------------- import rows
 
 __global__ void import_kernel(ull* __restrict__ table_keys,
@@ -476,6 +502,20 @@ void grad_scatter(torch::Tensor grads, torch::Tensor perm,
   }
 }
 
+torch::Tensor sign_prep(torch::Tensor values, torch::Tensor slot_starts,
+                        torch::Tensor prefixes, int64_t spacing) {
+  const int64_t n = values.numel();
+  auto out = torch::empty_like(values);
+  if (n == 0) return out;
+  hipLaunchKernelGGL(sign_prep_kernel, dim3(n_blocks_for(n, 256)), dim3(256),
+                     0, cur_stream(), (const ull*)values.data_ptr<int64_t>(),
+                     slot_starts.data_ptr<int64_t>(),
+                     (const ull*)prefixes.data_ptr<int64_t>(),
+                     (int)(slot_starts.numel() - 1), (ull)spacing,
+                     (ull*)out.data_ptr<int64_t>(), n);
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "persia_amd HIP kernels (gfx950)";
   m.def("store_lookup", &store_lookup, "hash-table lookup/insert + gather");
@@ -483,4 +523,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("store_import", &store_import, "bulk insert rows (checkpoint load)");
   m.def("segment_sum", &segment_sum, "fused gather + per-sample summation");
   m.def("grad_scatter", &grad_scatter, "ordered per-sign gradient scatter");
+  m.def("sign_prep", &sign_prep, "prefix-fold + splitmix64 key mixing");
 }
