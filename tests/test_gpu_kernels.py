@@ -182,6 +182,31 @@ def test_grad_scatter_matches_reference():
     assert torch.allclose(out.cpu(), ref, atol=0.05, rtol=0.01)
 
 
+def test_sign_prep_matches_numpy():
+    from persia_amd.core import hashing
+    from persia_amd.ops import native
+
+    C = native()
+    rng = np.random.default_rng(3)
+    sizes = [100, 57, 301]
+    prefixes = np.array([1 << 56, 2 << 56, 0], dtype=np.uint64)
+    spacing = (1 << 56) - 1
+    vals = [rng.integers(0, 2 ** 63, size=s, dtype=np.uint64) for s in sizes]
+    expected = []
+    for v, p in zip(vals, prefixes):
+        s = hashing.apply_prefix(v, int(p), spacing)
+        k = hashing.splitmix64(s)
+        k[k == 0] = np.uint64(0xD1B54A32D192ED03)
+        expected.append(k)
+    expected = np.concatenate(expected)
+    allv = torch.from_numpy(np.concatenate(vals).view(np.int64)).to(_dev())
+    starts = torch.tensor([0, 100, 157, 458], dtype=torch.int64, device=_dev())
+    pref_t = torch.from_numpy(prefixes.view(np.int64)).to(_dev())
+    out = C.sign_prep(allv, starts, pref_t, spacing)
+    got = out.cpu().numpy().view(np.uint64)
+    assert np.array_equal(got, expected)
+
+
 def test_engine_gpu_matches_cpu_engine():
     from persia_amd.core.comm import DistContext
     from persia_amd.core.engine import EmbeddingEngine
