@@ -130,13 +130,11 @@ def main():
 
     def train_step(tb):
         with amp_ctx:
-            embs = [p.sum_tensor for p in tb.payloads]
-            for e in embs:
-                e.requires_grad = True
+            embs = tb.training_embeddings()
             logits = model(tb.non_id_type_tensors, embs)
             loss = loss_fn(logits.float(), tb.label_tensors[0])
         loss.backward()
-        engine.apply_gradients(tb, {p.name: e.grad for p, e in zip(tb.payloads, embs)})
+        engine.apply_gradients_base(tb)
         opt.step()
         opt.zero_grad(set_to_none=True)
         pipeline.release_permit()

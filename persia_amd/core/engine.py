@@ -127,8 +127,30 @@ class _GroupCtx:
     seg_id: Optional[torch.Tensor] = None  # [nnz]: pos -> group seg idx, -1 raw
     seg_lens: Optional[torch.Tensor] = None  # [n_sum_segs] f32
     sqrt_mask: Optional[torch.Tensor] = None  # [n_sum_segs] bool
+    sum_base: Optional[torch.Tensor] = None  # [n_sum_segs, dim] f16 fused output
+    n_sum_slots: int = 0
     send_counts: Optional[List[int]] = None
     recv_counts: Optional[List[int]] = None
+
+
+class _GroupPlan:
+    """Static per-(group, batch-size) device constants for the all-single-ID
+    fast path (the DLRM shape: every slot one id per sample).  Everything here
+    is shape-derived, so one plan serves every batch of that shape."""
+
+    def __init__(self, dim: int, names, prefixes_np: np.ndarray, B: int,
+                 device: torch.device):
+        S = len(names)
+        self.dim = dim
+        self.names = names
+        self.B = B
+        self.S = S
+        self.slot_starts = torch.arange(0, (S + 1) * B, B, dtype=torch.int64,
+                                        device=device)
+        self.prefixes = torch.from_numpy(prefixes_np.view(np.int64)).to(device)
+        self.cat_offsets = torch.arange(S * B + 1, dtype=torch.int64, device=device)
+        self.seg_id = torch.arange(S * B, dtype=torch.int64, device=device)
+        self.empty_scale = torch.empty(0, dtype=torch.float32, device=device)
 
 
 class PersiaTrainingBatch:
@@ -145,6 +167,28 @@ class PersiaTrainingBatch:
         self.batch_id: Optional[int] = None
         self._groups: List[_GroupCtx] = []
         self._engine: Optional["EmbeddingEngine"] = None
+
+    def enable_training_views(self) -> Dict[str, torch.Tensor]:
+        """Mark each group's sum base requires_grad and return fresh per-slot
+        views connected to it (autograd accumulates into ``sum_base.grad`` —
+        consumed by ``EmbeddingEngine.apply_gradients_base``)."""
+        views: Dict[str, torch.Tensor] = {}
+        for group in self._groups:
+            if group.sum_base is None:
+                continue
+            group.sum_base.requires_grad_(True)
+            b0 = 0
+            for sc in group.slots:
+                if not sc.cfg.embedding_summation:
+                    continue
+                views[sc.name] = group.sum_base[b0 : b0 + self.batch_size]
+                b0 += self.batch_size
+        return views
+
+    def training_embeddings(self) -> List[torch.Tensor]:
+        """payload-ordered embedding tensors for a manual train loop."""
+        views = self.enable_training_views()
+        return [views.get(p.name, p.sum_tensor) for p in self.payloads]
 
 
 class EmbeddingEngine:
@@ -181,6 +225,12 @@ class EmbeddingEngine:
         }
         self.skipped_grad_signs = 0
         self.nan_grad_batches = 0
+        self._plans = {}
+        self._nan_dev = (
+            torch.zeros(1, dtype=torch.int64, device=device)
+            if device.type == "cuda"
+            else None
+        )
 
     # ------------------------------------------------------------- sign prep
 
@@ -263,6 +313,53 @@ class EmbeddingEngine:
         out.payloads.sort(key=lambda p: order[p.name])
         return out
 
+    def _process_group_fast(self, dim: int, feats, out: PersiaTrainingBatch,
+                            train: bool) -> _GroupCtx:
+        """All-single-ID sum-slot fast path (flagship DLRM shape): one H2D
+        upload, ~15 kernel launches, static device-side plan."""
+        from persia_amd.ops import native as _native
+
+        C = _native()
+        dev = self.device
+        B = len(feats[0].values)
+        names = tuple(f.name for f in feats)
+        plan = self._plans.get((dim, names, B))
+        if plan is None:
+            prefixes_np = np.array(
+                [self.schema.get_slot(n).index_prefix for n in names], dtype=np.uint64
+            )
+            plan = _GroupPlan(dim, names, prefixes_np, B, dev)
+            self._plans[(dim, names, B)] = plan
+        values_np = np.concatenate([f.values for f in feats])
+        vals_t = torch.from_numpy(values_np.view(np.int64)).to(dev)
+        spacing = self.schema.feature_spacing
+        keys_t = C.sign_prep(
+            vals_t, plan.slot_starts, plan.prefixes,
+            spacing if spacing < (1 << 63) else -1,
+        )
+        uniq_keys, inverse, perm, ustarts = _dedup(keys_t)
+        slot_ctxs = [
+            _SlotCtx(
+                name=f.name, cfg=self.schema.get_slot(f.name),
+                pos_slice=(i * B, (i + 1) * B), seg_offsets=None,
+                sum_seg_base=i * B,
+            )
+            for i, f in enumerate(feats)
+        ]
+        group = _GroupCtx(
+            dim=dim, uniq_keys=uniq_keys, inverse=inverse, perm=perm,
+            ustarts=ustarts, slots=slot_ctxs, cat_offsets=plan.cat_offsets,
+            seg_id=plan.seg_id, n_sum_slots=plan.S,
+        )
+        rows = self._exchange_rows(group, train)
+        sums = C.segment_sum(rows.contiguous(), inverse, plan.cat_offsets, plan.empty_scale)
+        group.sum_base = sums
+        for i, sc in enumerate(slot_ctxs):
+            out.payloads.append(
+                SlotPayload(name=sc.name, cfg=sc.cfg, sum_tensor=sums[i * B : (i + 1) * B])
+            )
+        return group
+
     def _process_group(self, dim: int, feats, out: PersiaTrainingBatch, train: bool) -> _GroupCtx:
         dev = self.device
         native = dev.type == "cuda"
@@ -270,6 +367,12 @@ class EmbeddingEngine:
             from persia_amd.ops import native as _native
 
             C = _native()
+            if all(getattr(f, "is_single", False) for f in feats) and all(
+                self.schema.get_slot(f.name).embedding_summation
+                and self.schema.get_slot(f.name).hash_stack_rounds == 0
+                for f in feats
+            ):
+                return self._process_group_fast(dim, feats, out, train)
         # sum slots first so they occupy a contiguous prefix of the position
         # space (one fused segment-sum launch for the whole group)
         feats = sorted(
@@ -369,6 +472,8 @@ class EmbeddingEngine:
                     R.segment_sum_rows(rows, inverse[:sum_end], cat_offsets, False, torch.float32)
                     * fwd_scale.unsqueeze(1)
                 ).to(torch.float16)
+            group.sum_base = sums
+            group.n_sum_slots = len(sum_slots)
             b0 = 0
             for sc in sum_slots:
                 nseg = sc.seg_offsets.numel() - 1
@@ -450,7 +555,7 @@ class EmbeddingEngine:
                 scale = torch.where(skip, torch.zeros_like(scale), scale)
                 C.grad_scatter(
                     grads_cat.contiguous(), group.perm, group.ustarts,
-                    group.seg_id, scale.contiguous(), buf,
+                    group.seg_id, scale.contiguous(), buf, 1,
                 )
                 any_grad = True  # kernels launched regardless (no host sync)
             for sc in group.slots:
@@ -487,32 +592,116 @@ class EmbeddingEngine:
                     buf.index_add_(0, sc.slot_uniq_global, gf)
             if not any_grad:
                 continue
-            store = self.stores[group.dim]
-            if not self.dist.distributed:
-                self.skipped_grad_signs += store.update_gradients(group.uniq_keys, buf)
+            self._route_and_update(group, buf)
+
+    def _route_and_update(self, group: _GroupCtx, buf: torch.Tensor) -> None:
+        store = self.stores[group.dim]
+        if not self.dist.distributed:
+            self.skipped_grad_signs += store.update_gradients(group.uniq_keys, buf)
+            return
+        send_counts, recv_counts = group.send_counts, group.recv_counts
+        if send_counts is None:
+            owner = _owner_of_keys(group.uniq_keys, self.dist.world_size)
+            send_counts = torch.bincount(owner, minlength=self.dist.world_size).tolist()
+            recv_counts = self.dist.all_to_all_lengths(send_counts)
+        keys_recv = self.dist.all_to_all(group.uniq_keys, send_counts, recv_counts)
+        grads_recv = self.dist.all_to_all(
+            buf.to(self.wire_dtype), send_counts, recv_counts
+        )
+        # merge duplicate keys across source ranks: pre-aggregate into one
+        # optimizer application per sign (the reference applies each rank's
+        # RPC sequentially under a per-sign lock — summing first is the
+        # synchronous-equivalent, deterministic, and race-free on GPU)
+        uniq_f, inv = torch.unique(keys_recv ^ _FLIP, sorted=True, return_inverse=True)
+        merged = torch.zeros(
+            uniq_f.numel(), group.dim, dtype=torch.float32, device=self.device
+        )
+        merged.index_add_(0, inv, grads_recv.float())
+        self.skipped_grad_signs += store.update_gradients(uniq_f ^ _FLIP, merged)
+
+    def apply_gradients_base(
+        self,
+        training_batch: PersiaTrainingBatch,
+        raw_grads: Optional[Dict[str, Optional[torch.Tensor]]] = None,
+        loss_scale: float = 1.0,
+    ) -> None:
+        """Backward fast path: sum-slot gradients are read directly from
+        ``group.sum_base.grad`` (the per-group base tensor that
+        ``prepare_features`` marked requires_grad), so there is no per-slot
+        cat/NaN-reduce cascade — one ordered scatter per group.
+
+        ``raw_grads``: slot name -> (U_slot, dim) f32 for raw slots (the
+        index_add_ de-dup output of ctx._on_backward)."""
+        native = self.device.type == "cuda"
+        if not native:
+            # CPU: slice the base into the generic per-slot dict
+            grads: Dict[str, Optional[torch.Tensor]] = dict(raw_grads or {})
+            for group in training_batch._groups:
+                gbase = (
+                    group.sum_base.grad if group.sum_base is not None else None
+                )
+                b0 = 0
+                for sc in group.slots:
+                    if not sc.cfg.embedding_summation:
+                        continue
+                    nseg = training_batch.batch_size
+                    grads[sc.name] = (
+                        gbase[b0 : b0 + nseg] if gbase is not None else None
+                    )
+                    b0 += nseg
+            return self.apply_gradients(training_batch, grads, loss_scale)
+
+        from persia_amd.ops import native as _native
+
+        C = _native()
+        for group in training_batch._groups:
+            U = group.uniq_keys.numel()
+            buf = torch.empty(U, group.dim, dtype=torch.float32, device=self.device)
+            gbase = group.sum_base.grad if group.sum_base is not None else None
+            if gbase is not None:
+                S = group.n_sum_slots
+                flags = torch.isnan(gbase).view(S, -1).any(dim=1)  # [S]
+                slot_scale = torch.where(
+                    flags,
+                    torch.zeros((), device=self.device),
+                    torch.full((), 1.0 / loss_scale, device=self.device),
+                )
+                B = gbase.shape[0] // S
+                seg_scale = slot_scale.unsqueeze(1).expand(S, B).reshape(-1)
+                if group.sqrt_mask is not None and bool(group.sqrt_mask.any()):
+                    sqrt_f = torch.where(
+                        group.sqrt_mask,
+                        group.seg_lens.clamp(min=1.0).rsqrt(),
+                        torch.ones_like(group.seg_lens),
+                    )
+                    seg_scale = seg_scale * sqrt_f
+                C.grad_scatter(
+                    gbase.contiguous(), group.perm, group.ustarts, group.seg_id,
+                    seg_scale.contiguous(), buf, 0,
+                )
+                if self._nan_dev is not None:
+                    self._nan_dev += flags.sum()
             else:
-                send_counts, recv_counts = group.send_counts, group.recv_counts
-                if send_counts is None:
-                    owner = _owner_of_keys(group.uniq_keys, self.dist.world_size)
-                    send_counts = torch.bincount(owner, minlength=self.dist.world_size).tolist()
-                    recv_counts = self.dist.all_to_all_lengths(send_counts)
-                keys_recv = self.dist.all_to_all(group.uniq_keys, send_counts, recv_counts)
-                grads_recv = self.dist.all_to_all(
-                    buf.to(self.wire_dtype), send_counts, recv_counts
-                )
-                # merge duplicate keys across source ranks: pre-aggregate into
-                # one optimizer application per sign (the reference applies
-                # each rank's RPC sequentially under a per-sign lock — summing
-                # first is the synchronous-equivalent, deterministic, and
-                # race-free on GPU)
-                uniq_f, inv = torch.unique(
-                    keys_recv ^ _FLIP, sorted=True, return_inverse=True
-                )
-                merged = torch.zeros(
-                    uniq_f.numel(), group.dim, dtype=torch.float32, device=self.device
-                )
-                merged.index_add_(0, inv, grads_recv.float())
-                self.skipped_grad_signs += store.update_gradients(uniq_f ^ _FLIP, merged)
+                buf.zero_()
+            # raw slots add on top (buf fully written above)
+            if raw_grads:
+                for sc in group.slots:
+                    if sc.cfg.embedding_summation:
+                        continue
+                    g = raw_grads.get(sc.name)
+                    if g is None:
+                        continue
+                    if bool(torch.isnan(g).any()):
+                        self.nan_grad_batches += 1
+                        continue
+                    gf = g.float()
+                    if loss_scale != 1.0:
+                        gf = gf / loss_scale
+                    rounds = sc.cfg.hash_stack_rounds
+                    if sc.cfg.sqrt_scaling and rounds > 0:
+                        gf = gf / float(np.sqrt(rounds))
+                    buf.index_add_(0, sc.slot_uniq_global, gf)
+            self._route_and_update(group, buf)
 
     # ------------------------------------------------------------ checkpoint
 

@@ -26,6 +26,7 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 
 #include "common.h"
 
@@ -35,6 +36,9 @@ namespace {
 
 __device__ __forceinline__ float pa_to_float(float x) { return x; }
 __device__ __forceinline__ float pa_to_float(__half x) { return __half2float(x); }
+__device__ __forceinline__ float pa_to_float(__hip_bfloat16 x) {
+  return __bfloat162float(x);
+}
 
 // ---------------------------------------------------------------- store probe
 
@@ -340,7 +344,7 @@ __global__ void segment_sum_kernel(const RowT* __restrict__ rows,
 // positions of a unique key are contiguous in sort order.  Positions with
 // seg_id < 0 (raw-slot positions, handled by a separate indexed add) are
 // skipped.
-template <typename GradT>
+template <typename GradT, bool ACCUMULATE>
 __global__ void grad_scatter_kernel(const GradT* __restrict__ grads,
                                     const int64_t* __restrict__ perm,
                                     const int64_t* __restrict__ ustarts,
@@ -359,9 +363,14 @@ __global__ void grad_scatter_kernel(const GradT* __restrict__ grads,
       float acc = 0.0f;
       for (int64_t p = lo; p < hi; ++p) {
         const int64_t s = seg_id[perm[p]];
-        if (s >= 0) acc += pa_to_float(grads[s * dim + c]) * seg_scale[s];
+        if (s >= 0) {
+          // scale 0 skips the slot entirely (NaN-slot mask: never multiply a
+          // NaN gradient, even by zero)
+          const float sc = seg_scale[s];
+          if (sc != 0.0f) acc += pa_to_float(grads[s * dim + c]) * sc;
+        }
       }
-      dst[c] += acc;
+      if (ACCUMULATE) dst[c] += acc; else dst[c] = acc;
     }
   }
 }
@@ -479,27 +488,29 @@ torch::Tensor segment_sum(torch::Tensor rows, torch::Tensor inverse,
 
 void grad_scatter(torch::Tensor grads, torch::Tensor perm,
                   torch::Tensor ustarts, torch::Tensor seg_id,
-                  torch::Tensor seg_scale, torch::Tensor out) {
+                  torch::Tensor seg_scale, torch::Tensor out,
+                  int64_t accumulate) {
   const int64_t n_unique = ustarts.numel() - 1;
   const int64_t dim = out.size(1);
   if (n_unique == 0) return;
   hipStream_t st = cur_stream();
   const dim3 grid(n_blocks_for(n_unique, 4)), block(256);
+#define PA_GS(T, ACC)                                                         \
+  hipLaunchKernelGGL((grad_scatter_kernel<T, ACC>), grid, block, 0, st,       \
+                     (const T*)grads.data_ptr(), perm.data_ptr<int64_t>(),    \
+                     ustarts.data_ptr<int64_t>(), seg_id.data_ptr<int64_t>(), \
+                     seg_scale.data_ptr<float>(), out.data_ptr<float>(),      \
+                     n_unique, (int)dim)
   if (grads.scalar_type() == torch::kFloat16) {
-    hipLaunchKernelGGL(grad_scatter_kernel<__half>, grid, block, 0, st,
-                       (const __half*)grads.data_ptr<at::Half>(),
-                       perm.data_ptr<int64_t>(), ustarts.data_ptr<int64_t>(),
-                       seg_id.data_ptr<int64_t>(), seg_scale.data_ptr<float>(),
-                       out.data_ptr<float>(), n_unique, (int)dim);
+    if (accumulate) PA_GS(__half, true); else PA_GS(__half, false);
   } else if (grads.scalar_type() == torch::kFloat32) {
-    hipLaunchKernelGGL(grad_scatter_kernel<float>, grid, block, 0, st,
-                       grads.data_ptr<float>(), perm.data_ptr<int64_t>(),
-                       ustarts.data_ptr<int64_t>(), seg_id.data_ptr<int64_t>(),
-                       seg_scale.data_ptr<float>(), out.data_ptr<float>(),
-                       n_unique, (int)dim);
+    if (accumulate) PA_GS(float, true); else PA_GS(float, false);
+  } else if (grads.scalar_type() == torch::kBFloat16) {
+    if (accumulate) PA_GS(__hip_bfloat16, true); else PA_GS(__hip_bfloat16, false);
   } else {
-    TORCH_CHECK(false, "grad_scatter: grads must be f16 or f32");
+    TORCH_CHECK(false, "grad_scatter: grads must be f16/bf16/f32");
   }
+#undef PA_GS
 }
 
 torch::Tensor sign_prep(torch::Tensor values, torch::Tensor slot_starts,

@@ -191,6 +191,10 @@ class EmbeddingCtx(BaseCtx):
             labels = batch.label_tensors
         is_training = mode == PreprocessMode.TRAIN and batch.requires_grad
 
+        # sum slots: gradients flow into the per-group BASE tensor (one fused
+        # scatter in backward instead of a per-slot cascade)
+        sum_views = batch.enable_training_views() if is_training else {}
+
         emb_tensors: List[torch.Tensor] = []
         cache = []  # (name, distinct_id_tensor, index, non_empty_index, emb_tensor)
         for p in batch.payloads:
@@ -216,8 +220,7 @@ class EmbeddingCtx(BaseCtx):
                     (p.name, distinct_id_tensor, index_tensor, non_empty_index, index_select_raw_tensor)
                 )
             else:
-                t = p.sum_tensor
-                t.requires_grad = is_training
+                t = sum_views.get(p.name, p.sum_tensor)
                 emb_tensors.append(t)
                 cache.append((p.name, None, None, None, t))
         batch._emb_cache = cache
@@ -376,33 +379,37 @@ class TrainCtx(EmbeddingCtx):
         batch = self.current_batch
         finite = True
         if self.mixed_precision and self.update_times % check_frequency == 0:
-            finite = _check_finite([c[-1].grad for c in batch._emb_cache])
+            finite = _check_finite(
+                [g.sum_base.grad for g in batch._groups if g.sum_base is not None]
+                + [
+                    c[-1].grad
+                    for c in batch._emb_cache
+                    if c[1] is not None  # raw slots: the index-selected tensor
+                ]
+            )
         self.update_times += 1
 
-        grads: Dict[str, Optional[torch.Tensor]] = {}
-        empty = []
+        # raw slots: per-slot de-dup scatter (reference ctx.py:968-976);
+        # sum slots: gradients are already in each group's sum_base.grad
+        raw_grads: Dict[str, Optional[torch.Tensor]] = {}
         for (name, distinct_id_tensor, index, non_empty_index, emb_tensor) in batch._emb_cache:
-            if emb_tensor.grad is None:
-                grads[name] = None
-                empty.append(name)
+            if distinct_id_tensor is None:
                 continue
-            if distinct_id_tensor is not None:
-                if distinct_id_tensor.shape[0] > 1:
-                    grad = torch.zeros_like(distinct_id_tensor, dtype=torch.float32)
-                    nz = non_empty_index.view(-1)
-                    non_zero_grad = emb_tensor.grad.index_select(0, nz).float()
-                    dst = index.view(-1)[nz]
-                    grad.index_add_(0, dst, non_zero_grad)
-                    grads[name] = grad[1:, :]
-                else:
-                    grads[name] = None
+            if emb_tensor.grad is None:
+                raw_grads[name] = None
+                continue
+            if distinct_id_tensor.shape[0] > 1:
+                grad = torch.zeros_like(distinct_id_tensor, dtype=torch.float32)
+                nz = non_empty_index.view(-1)
+                non_zero_grad = emb_tensor.grad.index_select(0, nz).float()
+                dst = index.view(-1)[nz]
+                grad.index_add_(0, dst, non_zero_grad)
+                raw_grads[name] = grad[1:, :]
             else:
-                grads[name] = emb_tensor.grad  # f16 (B, dim)
-        self.engine.apply_gradients(batch, grads, loss_scale)
+                raw_grads[name] = None
+        self.engine.apply_gradients_base(batch, raw_grads, loss_scale)
         if self._pipeline is not None and batch.requires_grad:
             self._pipeline.release_permit()
-        if empty:
-            _logger.warning(f"Current batch has empty gradient tensors: {empty}")
         return finite
 
 

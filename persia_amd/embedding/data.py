@@ -142,14 +142,20 @@ class Label(NdarrayDataBase):
 
 
 class _FlatIDFeature:
-    """Internal flat CSR form of one slot's ID batch."""
+    """Internal flat CSR form of one slot's ID batch.
 
-    __slots__ = ("name", "values", "offsets")
+    ``is_single`` marks exactly-one-id-per-sample slots (offsets == arange):
+    the engine's static-plan fast path applies to batches of only such slots.
+    """
 
-    def __init__(self, name: str, values: np.ndarray, offsets: np.ndarray):
+    __slots__ = ("name", "values", "offsets", "is_single")
+
+    def __init__(self, name: str, values: np.ndarray, offsets: np.ndarray,
+                 is_single: bool = False):
         self.name = name
         self.values = values
         self.offsets = offsets
+        self.is_single = is_single
 
     @property
     def batch_size(self) -> int:
@@ -196,7 +202,12 @@ class PersiaBatch:
             assert f.name not in seen, f"duplicate id_type_feature name: {f.name}"
             seen.add(f.name)
             values, offsets = f.flatten()
-            self.id_type_features.append(_FlatIDFeature(f.name, values, offsets))
+            self.id_type_features.append(
+                _FlatIDFeature(
+                    f.name, values, offsets,
+                    is_single=isinstance(f, IDTypeFeatureWithSingleID),
+                )
+            )
 
         self.non_id_type_features: List[NonIDTypeFeature] = []
         for i, x in enumerate(non_id_type_features or []):
@@ -298,7 +309,12 @@ class PersiaBatch:
             name = get_str()
             values = get_arr()
             offsets = get_arr()
-            obj.id_type_features.append(_FlatIDFeature(name, values, offsets))
+            obj.id_type_features.append(
+                _FlatIDFeature(
+                    name, values, offsets,
+                    is_single=bool(np.all(np.diff(offsets) == 1)),
+                )
+            )
         obj.non_id_type_features = []
         for _ in range(n_nid):
             name = get_str()

@@ -61,15 +61,13 @@ def test_dlrm_train_loop_learns():
     losses = []
     for _ in range(n_steps):
         tb = pipeline.get(timeout=120)
-        embs = [p.sum_tensor for p in tb.payloads]
-        for e in embs:
-            e.requires_grad = True
+        embs = tb.training_embeddings()
         logits = model(tb.non_id_type_tensors, embs)
         loss = torch.nn.functional.binary_cross_entropy_with_logits(
             logits.float(), tb.label_tensors[0]
         )
         loss.backward()
-        engine.apply_gradients(tb, {p.name: e.grad for p, e in zip(tb.payloads, embs)})
+        engine.apply_gradients_base(tb)
         opt.step()
         opt.zero_grad(set_to_none=True)
         pipeline.release_permit()
