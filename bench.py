@@ -35,6 +35,8 @@ def parse_args():
     p.add_argument("--staleness", type=int, default=8)
     p.add_argument("--model", type=str, default="dlrm", choices=["dlrm", "dcn"])
     p.add_argument("--device", type=str, default=None)
+    p.add_argument("--graph", type=int, default=1,
+                   help="capture dense fwd+bwd in a hipGraph (1 GPU only)")
     return p.parse_args()
 
 
@@ -128,11 +130,37 @@ def main():
     amp_dtype = torch.bfloat16
     amp_ctx = torch.autocast("cuda", dtype=amp_dtype) if use_gpu else _nullcontext()
 
+    # the flagship path feeds the model the PACKED per-group sum tensor
+    # ([S*B, dim] slot-major) — one reshape in the model, gradients land in
+    # sum_base.grad with no per-slot autograd traffic; hipGraph-capture the
+    # dense fwd+bwd (launch-bound otherwise)
+    graphed = None
+    if args.graph and use_gpu and world == 1 and args.model == "dlrm":
+        sample_dense = torch.zeros(B, args.num_dense, device=device)
+        sample_base = torch.zeros(
+            n_slots * B, dim, dtype=torch.float16, device=device, requires_grad=True
+        )
+        try:
+            with torch.autocast("cuda", dtype=amp_dtype):
+                graphed = torch.cuda.make_graphed_callables(
+                    model, (sample_dense, sample_base), allow_unused_input=True
+                )
+        except Exception as e:  # pragma: no cover - fall back to eager
+            print(f"# hipGraph capture failed ({e}); running eager", flush=True)
+            graphed = None
+
     def train_step(tb):
-        with amp_ctx:
-            embs = tb.training_embeddings()
-            logits = model(tb.non_id_type_tensors, embs)
-            loss = loss_fn(logits.float(), tb.label_tensors[0])
+        if graphed is not None:
+            base = tb._groups[0].sum_base
+            base.requires_grad_(True)
+            with amp_ctx:
+                logits = graphed(tb.non_id_type_tensors[0], base)
+                loss = loss_fn(logits.float(), tb.label_tensors[0])
+        else:
+            with amp_ctx:
+                embs = tb.training_embeddings()
+                logits = model(tb.non_id_type_tensors, embs)
+                loss = loss_fn(logits.float(), tb.label_tensors[0])
         loss.backward()
         engine.apply_gradients_base(tb)
         opt.step()

@@ -153,6 +153,28 @@ class _GroupPlan:
         self.empty_scale = torch.empty(0, dtype=torch.float32, device=device)
 
 
+class _SplitSlots(torch.autograd.Function):
+    """Split a group's fused sum tensor into per-slot views with ONE backward
+    op: autograd's native per-view SliceBackward would zero-fill and add a
+    full-base-sized buffer per slot (2×n_slots kernels per step)."""
+
+    @staticmethod
+    def forward(ctx, base: torch.Tensor, n_slots: int, B: int):
+        ctx.n_slots, ctx.B, ctx.dim = n_slots, B, base.shape[1]
+        ctx.dev, ctx.dtype = base.device, base.dtype
+        return tuple(base[i * B : (i + 1) * B] for i in range(n_slots))
+
+    @staticmethod
+    def backward(ctx, *grads):
+        parts = [
+            g
+            if g is not None
+            else torch.zeros(ctx.B, ctx.dim, dtype=ctx.dtype, device=ctx.dev)
+            for g in grads
+        ]
+        return torch.cat(parts, dim=0), None, None
+
+
 class PersiaTrainingBatch:
     """Device-side batch ready for the dense model (reference
     PersiaTrainingBatch, persia-core/src/forward.rs:256-331)."""
@@ -177,12 +199,9 @@ class PersiaTrainingBatch:
             if group.sum_base is None:
                 continue
             group.sum_base.requires_grad_(True)
-            b0 = 0
-            for sc in group.slots:
-                if not sc.cfg.embedding_summation:
-                    continue
-                views[sc.name] = group.sum_base[b0 : b0 + self.batch_size]
-                b0 += self.batch_size
+            sum_names = [sc.name for sc in group.slots if sc.cfg.embedding_summation]
+            parts = _SplitSlots.apply(group.sum_base, len(sum_names), self.batch_size)
+            views.update(zip(sum_names, parts))
         return views
 
     def training_embeddings(self) -> List[torch.Tensor]:

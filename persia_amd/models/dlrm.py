@@ -52,11 +52,25 @@ class DLRM(nn.Module):
         self.top = _mlp([dim + inter_out] + list(top_mlp) + [1])
 
     def forward(
-        self, non_id_tensors: List[torch.Tensor], embedding_tensors: List[torch.Tensor]
+        self, non_id_tensors, embedding_tensors
     ) -> torch.Tensor:
-        dense = non_id_tensors[0].float()
+        if torch.is_tensor(non_id_tensors):
+            dense = non_id_tensors.float()
+        else:
+            dense = non_id_tensors[0].float()
         x = self.bottom(dense)  # [B, D]
-        vectors = torch.stack([x] + [e.to(x.dtype) for e in embedding_tensors], dim=1)
+        if torch.is_tensor(embedding_tensors):
+            # packed slot-major [S*B, D] (the engine's fused sum output) —
+            # ONE reshape instead of a 26-way stack
+            B = x.shape[0]
+            emb = (
+                embedding_tensors.view(self.num_sparse, B, self.dim)
+                .permute(1, 0, 2)
+                .to(x.dtype)
+            )
+            vectors = torch.cat([x.unsqueeze(1), emb], dim=1)
+        else:
+            vectors = torch.stack([x] + [e.to(x.dtype) for e in embedding_tensors], dim=1)
         inter = self.interaction(vectors)
         out = self.top(torch.cat([x, inter], dim=1))
         return out.squeeze(1)  # logits

@@ -174,7 +174,11 @@ def test_grad_scatter_matches_reference():
     lens = (seg_offsets[1:] - seg_offsets[:-1]).float()
     scale = lens.clamp(min=1.0).rsqrt() / 2.0  # sqrt scaling + loss scale 2
     out = torch.zeros(U, dim, device=_dev())
-    C.grad_scatter(grads, perm, ustarts, seg_id, scale, out)
+    C.grad_scatter(grads, perm, ustarts, seg_id, scale, out, 1)
+    # write-mode must agree with accumulate-into-zeros
+    out_w = torch.empty(U, dim, device=_dev())
+    C.grad_scatter(grads, perm, ustarts, seg_id, scale, out_w, 0)
+    assert torch.equal(out, out_w)
     ref = R.segment_grad_scatter(
         grads.cpu(), inverse.cpu(), seg_offsets.cpu(), U, scale_factor=2.0,
         sqrt_scaling=True,
