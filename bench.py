@@ -128,7 +128,12 @@ def main():
     pipeline.start()
 
     amp_dtype = torch.bfloat16
-    amp_ctx = torch.autocast("cuda", dtype=amp_dtype) if use_gpu else _nullcontext()
+    # cache_enabled=False: required for hipGraph capture of autocast regions
+    amp_ctx = (
+        torch.autocast("cuda", dtype=amp_dtype, cache_enabled=False)
+        if use_gpu
+        else _nullcontext()
+    )
 
     # the flagship path feeds the model the PACKED per-group sum tensor
     # ([S*B, dim] slot-major) — one reshape in the model, gradients land in
@@ -141,7 +146,7 @@ def main():
             n_slots * B, dim, dtype=torch.float16, device=device, requires_grad=True
         )
         try:
-            with torch.autocast("cuda", dtype=amp_dtype):
+            with torch.autocast("cuda", dtype=amp_dtype, cache_enabled=False):
                 graphed = torch.cuda.make_graphed_callables(
                     model, (sample_dense, sample_base), allow_unused_input=True
                 )
