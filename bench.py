@@ -169,7 +169,9 @@ def main():
         loss.backward()
         engine.apply_gradients_base(tb)
         opt.step()
-        opt.zero_grad(set_to_none=True)
+        # graphed backward writes into static .grad buffers: zeroing must keep
+        # them alive (set_to_none would free the graph's output addresses)
+        opt.zero_grad(set_to_none=graphed is None)
         pipeline.release_permit()
         return loss
 
