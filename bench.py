@@ -137,41 +137,66 @@ def main():
 
     # the flagship path feeds the model the PACKED per-group sum tensor
     # ([S*B, dim] slot-major) — one reshape in the model, gradients land in
-    # sum_base.grad with no per-slot autograd traffic; hipGraph-capture the
-    # dense fwd+bwd (launch-bound otherwise)
-    graphed = None
+    # sum_base.grad with no per-slot autograd traffic.  The whole dense
+    # iteration (zero-grads + fwd + bwd + SGD step) is captured into ONE
+    # hipGraph with static in/out buffers; each step copies inputs in and
+    # replays (the dense side was launch-bound, ~200 kernels/step).
+    graph = None
+    static = {}
     if args.graph and use_gpu and world == 1 and args.model == "dlrm":
-        sample_dense = torch.zeros(B, args.num_dense, device=device)
-        sample_base = torch.zeros(
-            n_slots * B, dim, dtype=torch.float16, device=device, requires_grad=True
-        )
         try:
-            with torch.autocast("cuda", dtype=amp_dtype, cache_enabled=False):
-                graphed = torch.cuda.make_graphed_callables(
-                    model, (sample_dense, sample_base), allow_unused_input=True
-                )
+            static = {
+                "dense": torch.zeros(B, args.num_dense, device=device),
+                "base": torch.zeros(
+                    n_slots * B, dim, dtype=torch.float16, device=device,
+                    requires_grad=True,
+                ),
+                "label": torch.zeros(B, device=device),
+            }
+
+            def iteration():
+                static["base"].grad.zero_()
+                with torch.autocast("cuda", dtype=amp_dtype, cache_enabled=False):
+                    logits = model(static["dense"], static["base"])
+                    loss = loss_fn(logits.float(), static["label"])
+                loss.backward()
+                opt.step()
+                for p in model.parameters():
+                    p.grad.zero_()
+                return loss
+
+            # warmup on a side stream (allocator state, autotuned GEMMs)
+            static["base"].grad = torch.zeros_like(static["base"], dtype=torch.float16)
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(3):
+                    iteration()
+            torch.cuda.current_stream().wait_stream(s)
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                static["loss"] = iteration()
         except Exception as e:  # pragma: no cover - fall back to eager
             print(f"# hipGraph capture failed ({e}); running eager", flush=True)
-            graphed = None
+            graph = None
 
     def train_step(tb):
-        if graphed is not None:
-            base = tb._groups[0].sum_base
-            base.requires_grad_(True)
-            with amp_ctx:
-                logits = graphed(tb.non_id_type_tensors[0], base)
-                loss = loss_fn(logits.float(), tb.label_tensors[0])
-        else:
-            with amp_ctx:
-                embs = tb.training_embeddings()
-                logits = model(tb.non_id_type_tensors, embs)
-                loss = loss_fn(logits.float(), tb.label_tensors[0])
+        if graph is not None:
+            static["dense"].copy_(tb.non_id_type_tensors[0], non_blocking=True)
+            static["base"].copy_(tb._groups[0].sum_base.detach(), non_blocking=True)
+            static["label"].copy_(tb.label_tensors[0], non_blocking=True)
+            graph.replay()
+            engine.apply_gradients_base(tb, sum_base_grads=[static["base"].grad])
+            pipeline.release_permit()
+            return static["loss"]
+        with amp_ctx:
+            embs = tb.training_embeddings()
+            logits = model(tb.non_id_type_tensors, embs)
+            loss = loss_fn(logits.float(), tb.label_tensors[0])
         loss.backward()
         engine.apply_gradients_base(tb)
         opt.step()
-        # graphed backward writes into static .grad buffers: zeroing must keep
-        # them alive (set_to_none would free the graph's output addresses)
-        opt.zero_grad(set_to_none=graphed is None)
+        opt.zero_grad(set_to_none=True)
         pipeline.release_permit()
         return loss
 

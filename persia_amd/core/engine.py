@@ -643,6 +643,7 @@ class EmbeddingEngine:
         training_batch: PersiaTrainingBatch,
         raw_grads: Optional[Dict[str, Optional[torch.Tensor]]] = None,
         loss_scale: float = 1.0,
+        sum_base_grads: Optional[List[torch.Tensor]] = None,
     ) -> None:
         """Backward fast path: sum-slot gradients are read directly from
         ``group.sum_base.grad`` (the per-group base tensor that
@@ -673,10 +674,13 @@ class EmbeddingEngine:
         from persia_amd.ops import native as _native
 
         C = _native()
-        for group in training_batch._groups:
+        for gi, group in enumerate(training_batch._groups):
             U = group.uniq_keys.numel()
             buf = torch.empty(U, group.dim, dtype=torch.float32, device=self.device)
-            gbase = group.sum_base.grad if group.sum_base is not None else None
+            if sum_base_grads is not None:
+                gbase = sum_base_grads[gi]
+            else:
+                gbase = group.sum_base.grad if group.sum_base is not None else None
             if gbase is not None:
                 S = group.n_sum_slots
                 flags = torch.isnan(gbase).view(S, -1).any(dim=1)  # [S]
