@@ -182,9 +182,10 @@ def main():
 
     def train_step(tb):
         if graph is not None:
-            static["dense"].copy_(tb.non_id_type_tensors[0], non_blocking=True)
-            static["base"].copy_(tb._groups[0].sum_base.detach(), non_blocking=True)
-            static["label"].copy_(tb.label_tensors[0], non_blocking=True)
+            with torch.no_grad():
+                static["dense"].copy_(tb.non_id_type_tensors[0], non_blocking=True)
+                static["base"].copy_(tb._groups[0].sum_base, non_blocking=True)
+                static["label"].copy_(tb.label_tensors[0], non_blocking=True)
             graph.replay()
             engine.apply_gradients_base(tb, sum_base_grads=[static["base"].grad])
             pipeline.release_permit()
