@@ -209,6 +209,33 @@ class PersiaTrainingBatch:
         views = self.enable_training_views()
         return [views.get(p.name, p.sum_tensor) for p in self.payloads]
 
+    def record_stream(self, stream: "torch.cuda.Stream") -> None:
+        """Tag every device tensor produced on the lookup stream as in-use by
+        ``stream`` so the caching allocator cannot recycle the memory for new
+        lookup-stream allocations while consumer kernels still read it
+        (mandatory for cross-stream tensor hand-off)."""
+        for t in self._device_tensors():
+            t.record_stream(stream)
+
+    def _device_tensors(self):
+        for t in self.non_id_type_tensors + self.label_tensors:
+            if t.is_cuda:
+                yield t
+        for g in self._groups:
+            for t in (g.uniq_keys, g.inverse, g.perm, g.ustarts, g.sum_base,
+                      g.cat_offsets, g.seg_id, g.seg_lens, g.sqrt_mask):
+                if t is not None and t.is_cuda:
+                    yield t
+            for sc in g.slots:
+                for t in (sc.seg_offsets, sc.slot_uniq_global, sc.slot_inverse):
+                    if t is not None and t.is_cuda:
+                        yield t
+        for p in self.payloads:
+            for t in (p.raw_distinct, p.raw_index, p.raw_non_empty_index,
+                      p.raw_sample_id_num):
+                if t is not None and t.is_cuda:
+                    yield t
+
 
 class EmbeddingEngine:
     def __init__(
@@ -806,7 +833,9 @@ class ForwardPipeline:
         if tb is None and self._exc is not None:
             raise self._exc
         if tb is not None and getattr(tb, "_ready_event", None) is not None:
-            torch.cuda.current_stream().wait_event(tb._ready_event)
+            cur = torch.cuda.current_stream()
+            cur.wait_event(tb._ready_event)
+            tb.record_stream(cur)
         return tb
 
     def release_permit(self):
