@@ -272,11 +272,12 @@ class EmbeddingEngine:
         self.skipped_grad_signs = 0
         self.nan_grad_batches = 0
         self._plans = {}
-        self._nan_dev = (
-            torch.zeros(1, dtype=torch.int64, device=device)
-            if device.type == "cuda"
-            else None
-        )
+        self._empty_scale = None
+
+    def _empty_f32(self) -> torch.Tensor:
+        if self._empty_scale is None:
+            self._empty_scale = torch.empty(0, dtype=torch.float32, device=self.device)
+        return self._empty_scale
 
     # ------------------------------------------------------------- sign prep
 
@@ -709,28 +710,26 @@ class EmbeddingEngine:
             else:
                 gbase = group.sum_base.grad if group.sum_base is not None else None
             if gbase is not None:
-                S = group.n_sum_slots
-                flags = torch.isnan(gbase).view(S, -1).any(dim=1)  # [S]
-                slot_scale = torch.where(
-                    flags,
-                    torch.zeros((), device=self.device),
-                    torch.full((), 1.0 / loss_scale, device=self.device),
-                )
-                B = gbase.shape[0] // S
-                seg_scale = slot_scale.unsqueeze(1).expand(S, B).reshape(-1)
-                if group.sqrt_mask is not None and bool(group.sqrt_mask.any()):
-                    sqrt_f = torch.where(
-                        group.sqrt_mask,
-                        group.seg_lens.clamp(min=1.0).rsqrt(),
-                        torch.ones_like(group.seg_lens),
+                # NaN gradients flow into buf and are skipped ROW-wise by the
+                # update kernel (store.update_gradients) — no isnan pass here
+                has_sqrt = group.sqrt_mask is not None and bool(group.sqrt_mask.any())
+                if loss_scale == 1.0 and not has_sqrt:
+                    seg_scale = self._empty_f32()
+                else:
+                    n_segs = gbase.shape[0]
+                    seg_scale = torch.full(
+                        (n_segs,), 1.0 / loss_scale, device=self.device
                     )
-                    seg_scale = seg_scale * sqrt_f
+                    if has_sqrt:
+                        seg_scale = seg_scale * torch.where(
+                            group.sqrt_mask,
+                            group.seg_lens.clamp(min=1.0).rsqrt(),
+                            torch.ones_like(group.seg_lens),
+                        )
                 C.grad_scatter(
                     gbase.contiguous(), group.perm, group.ustarts, group.seg_id,
-                    seg_scale.contiguous(), buf, 0,
+                    seg_scale, buf, 0,
                 )
-                if self._nan_dev is not None:
-                    self._nan_dev += flags.sum()
             else:
                 buf.zero_()
             # raw slots add on top (buf fully written above)

@@ -228,7 +228,10 @@ class CpuEmbeddingStore(EmbeddingStoreBase):
         k_np[k_np == EMPTY_KEY] = _ZERO_REMAP
         slots = np.array([self._probe(k) for k in k_np], dtype=np.int64)
         present = slots >= 0
-        skipped = int((~present).sum())
+        # row-level NaN skip (same semantics as the HIP update kernel)
+        nan_rows = torch.isnan(grads.float()).any(dim=1).cpu().numpy()
+        present = present & ~nan_rows
+        skipped = int((slots < 0).sum())
         if present.sum() == 0:
             return skipped
         idx = torch.from_numpy(slots[present])
@@ -301,7 +304,7 @@ class HipEmbeddingStore(EmbeddingStoreBase):
         self.arena = torch.zeros(
             self.n_slots, self.row_width, dtype=torch.float32, device=device
         )
-        self._skipped = torch.zeros(1, dtype=torch.int32, device=device)
+        self._skipped = torch.zeros(2, dtype=torch.int32, device=device)  # miss, nan
         self._opt_code = {"sgd": 0, "adagrad": 1, "adam": 2}[optimizer.kind]
 
     def _opt_params(self):
@@ -365,7 +368,10 @@ class HipEmbeddingStore(EmbeddingStoreBase):
         return 0  # skipped count is accumulated device-side (no hot-path sync)
 
     def skipped_count(self) -> int:
-        return int(self._skipped.item())
+        return int(self._skipped[0].item())
+
+    def nan_row_count(self) -> int:
+        return int(self._skipped[1].item())
 
     def export_rows(self) -> Tuple[np.ndarray, np.ndarray]:
         occ = torch.nonzero(self.keys != 0, as_tuple=False).view(-1)

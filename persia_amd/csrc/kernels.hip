@@ -207,13 +207,21 @@ __global__ void update_kernel(ull* __restrict__ table_keys,
     }
     const unsigned long long found = __ballot(slot >= 0);
     if (found == 0) {
-      if (lane == 0) atomicAdd(skipped, 1);
+      if (lane == 0) atomicAdd(&skipped[0], 1);
       continue;
     }
     const int src = __ffsll((long long)found) - 1;
     slot = __shfl(slot, src);
     float* row = arena + (int64_t)slot * row_width;
     const float* g = grads + i * dim;
+    // row-level NaN skip (finer-grained than the reference's per-slot skip,
+    // mod.rs:731-746: a NaN gradient never touches the table)
+    bool has_nan = false;
+    for (int c = lane; c < dim; c += PA_WAVE) has_nan |= isnan(g[c]);
+    if (__ballot(has_nan) != 0) {
+      if (lane == 0) atomicAdd(&skipped[1], 1);
+      continue;
+    }
     if (opt == 0) {  // SGD: w -= lr*(g + wd*w)         p0=lr p1=wd
       for (int c = lane; c < dim; c += PA_WAVE) {
         float w = row[c] - p0 * (g[c] + p1 * row[c]);
@@ -365,8 +373,8 @@ __global__ void grad_scatter_kernel(const GradT* __restrict__ grads,
         const int64_t s = seg_id[perm[p]];
         if (s >= 0) {
           // scale 0 skips the slot entirely (NaN-slot mask: never multiply a
-          // NaN gradient, even by zero)
-          const float sc = seg_scale[s];
+          // NaN gradient, even by zero); null scale = 1
+          const float sc = seg_scale ? seg_scale[s] : 1.0f;
           if (sc != 0.0f) acc += pa_to_float(grads[s * dim + c]) * sc;
         }
       }
@@ -424,7 +432,7 @@ void store_update(torch::Tensor table_keys, torch::Tensor ticks,
                   torch::Tensor grads, int64_t dim, int64_t opt,
                   std::vector<double> params, double b1_power,
                   double b2_power, double weight_bound,
-                  torch::Tensor skipped /* persistent i32[1], no sync */) {
+                  torch::Tensor skipped /* persistent i32[2]: miss, nan */) {
   const int64_t n = query.numel();
   if (n == 0) return;
   const int64_t n_buckets = table_keys.numel() / PA_BUCKET_SIZE;
@@ -495,11 +503,13 @@ void grad_scatter(torch::Tensor grads, torch::Tensor perm,
   if (n_unique == 0) return;
   hipStream_t st = cur_stream();
   const dim3 grid(n_blocks_for(n_unique, 4)), block(256);
+  const float* scale_ptr =
+      seg_scale.numel() ? seg_scale.data_ptr<float>() : nullptr;
 #define PA_GS(T, ACC)                                                         \
   hipLaunchKernelGGL((grad_scatter_kernel<T, ACC>), grid, block, 0, st,       \
                      (const T*)grads.data_ptr(), perm.data_ptr<int64_t>(),    \
                      ustarts.data_ptr<int64_t>(), seg_id.data_ptr<int64_t>(), \
-                     seg_scale.data_ptr<float>(), out.data_ptr<float>(),      \
+                     scale_ptr, out.data_ptr<float>(),                        \
                      n_unique, (int)dim)
   if (grads.scalar_type() == torch::kFloat16) {
     if (accumulate) PA_GS(__half, true); else PA_GS(__half, false);
