@@ -273,6 +273,10 @@ class EmbeddingEngine:
         self.nan_grad_batches = 0
         self._plans = {}
         self._empty_scale = None
+        from persia_amd.core.metrics import EngineMetrics
+
+        self.metrics_enabled = bool(gconf.enable_metrics)
+        self.metrics = EngineMetrics(self.metrics_enabled)
 
     def _empty_f32(self) -> torch.Tensor:
         if self._empty_scale is None:
@@ -783,6 +787,7 @@ class ForwardPipeline:
         self._thread: Optional[threading.Thread] = None
         self._stop = threading.Event()
         self._exc: Optional[BaseException] = None
+        self.inflight = 0  # lookups ahead of their gradient push (staleness)
 
     def start(self):
         if self._thread is None:
@@ -804,6 +809,9 @@ class ForwardPipeline:
                 self._out.put(None)
                 break
             self._sem.acquire()
+            self.inflight += 1
+            if self.engine.metrics_enabled:
+                self.engine.metrics.staleness.set(self.inflight)
             try:
                 if stream is not None:
                     with torch.cuda.stream(stream):
@@ -822,6 +830,8 @@ class ForwardPipeline:
     def put(self, batch: PersiaBatch):
         self.start()
         self._in.put(batch)
+        if self.engine.metrics_enabled:
+            self.engine.metrics.num_pending_batches.set(self._in.qsize())
 
     def finish(self):
         self.start()
@@ -840,6 +850,7 @@ class ForwardPipeline:
     def release_permit(self):
         """Called after the batch's gradients were applied (or the batch was
         dropped) — the staleness window slides."""
+        self.inflight = max(0, self.inflight - 1)
         self._sem.release()
 
     def stop(self):

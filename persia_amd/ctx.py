@@ -79,6 +79,9 @@ class BaseCtx:
     """Bootstraps rank info and (for embedding-carrying ctxs) the engine."""
 
     def __init__(self, threadpool_worker_size: int = 10, device_id: Optional[int] = None):
+        from persia_amd.core.watchdog import maybe_start_deadlock_detection
+
+        maybe_start_deadlock_detection()
         if device_id is None:
             device_id = _env.get_local_rank() if torch.cuda.is_available() else None
         if device_id is not None and device_id >= 0:

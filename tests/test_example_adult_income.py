@@ -1,0 +1,37 @@
+"""The reference's correctness anchor: deterministic-mode AUC reproducibility
+on the adult-income-style workload (SURVEY §4 / BASELINE.md)."""
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run_once():
+    env = dict(os.environ)
+    env.update(REPRODUCIBLE="1", EMBEDDING_STALENESS="1")
+    code = (
+        "import sys; sys.path.insert(0, r'%s');"
+        "sys.path.insert(0, r'%s');"
+        "import train; a, b = train.main(epochs=1); print('AUCS', repr(a), repr(b))"
+        % (os.path.join(REPO, "examples", "adult_income"), REPO)
+    )
+    out = subprocess.run(
+        [sys.executable, "-c", code],
+        cwd=os.path.join(REPO, "examples", "adult_income"),
+        env=env, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("AUCS")][0]
+    _, train_auc, test_auc = line.split()
+    return train_auc, test_auc
+
+
+@pytest.mark.timeout(1200)
+def test_deterministic_auc_reproduces():
+    a1 = _run_once()
+    a2 = _run_once()
+    assert a1 == a2, f"deterministic mode must reproduce bitwise: {a1} != {a2}"
+    assert float(a1[0].split("(")[-1].rstrip(")")) > 0.80
