@@ -20,24 +20,48 @@ import numpy as np
 import torch
 
 
+# The five BASELINE.json configs. "criteo" is the headline.
+PRESETS = {
+    "ctr-smoke": dict(model="dlrm", num_sparse=2, num_dense=5, dim=8, rows=2e4,
+                      batch_size=128),
+    "criteo": dict(model="dlrm", num_sparse=26, num_dense=13, dim=128, rows=1e8),
+    "terabyte": dict(model="dlrm", num_sparse=26, num_dense=13, dim=128, rows=1e10),
+    "dcn-spill": dict(model="dcn", num_sparse=26, num_dense=13, dim=64, rows=1e11,
+                      spill_capacity=2e8),
+    "100t": dict(model="dlrm", num_sparse=64, num_dense=13, dim=8, rows=1e12),
+}
+
+
 def parse_args():
     p = argparse.ArgumentParser()
+    p.add_argument("--preset", type=str, default="criteo", choices=list(PRESETS))
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=50)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--batch-size", type=int, default=4096, help="per-GPU batch")
-    p.add_argument("--num-sparse", type=int, default=26)
-    p.add_argument("--num-dense", type=int, default=13)
-    p.add_argument("--dim", type=int, default=128)
-    p.add_argument("--rows", type=float, default=1e8, help="total ID space")
+    p.add_argument("--batch-size", type=int, default=None, help="per-GPU batch")
+    p.add_argument("--num-sparse", type=int, default=None)
+    p.add_argument("--num-dense", type=int, default=None)
+    p.add_argument("--dim", type=int, default=None)
+    p.add_argument("--rows", type=float, default=None, help="total ID space")
     p.add_argument("--capacity", type=float, default=None,
-                   help="resident rows per rank (default: rows/world)")
+                   help="resident rows per rank (default: min(rows/world, HBM budget))")
+    p.add_argument("--hbm-budget-gb", type=float, default=200.0,
+                   help="HBM bytes budget for the table per rank")
+    p.add_argument("--spill-capacity", type=float, default=None,
+                   help="host-DRAM rows per rank (0 = off)")
     p.add_argument("--staleness", type=int, default=8)
-    p.add_argument("--model", type=str, default="dlrm", choices=["dlrm", "dcn"])
+    p.add_argument("--model", type=str, default=None, choices=["dlrm", "dcn"])
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--graph", type=int, default=1,
                    help="capture dense fwd+bwd in a hipGraph (1 GPU only)")
-    return p.parse_args()
+    args = p.parse_args()
+    preset = dict(PRESETS[args.preset])
+    preset.setdefault("batch_size", 4096)
+    preset.setdefault("spill_capacity", 0)
+    for k, v in preset.items():
+        if getattr(args, k, None) is None:
+            setattr(args, k, v)
+    return args
 
 
 def main():
@@ -71,7 +95,12 @@ def main():
     n_slots, dim, B = args.num_sparse, args.dim, args.batch_size
     rows_total = int(args.rows)
     vocab_per_slot = max(1, rows_total // n_slots)
-    capacity = int(args.capacity) if args.capacity else max(1 << 20, rows_total // world)
+    # Adagrad doubles the row (dim emb + dim accumulator) + 12B key/tick
+    hbm_rows = int(args.hbm_budget_gb * 1e9 / (dim * 2 * 4 + 12))
+    capacity = (
+        int(args.capacity) if args.capacity
+        else max(1 << 20, min(rows_total // world, hbm_rows))
+    )
 
     schema = EmbeddingSchema(
         slots={f"f{i}": SlotConfig(name=f"f{i}", dim=dim) for i in range(n_slots)},
@@ -81,7 +110,9 @@ def main():
         schema=schema,
         hyper=EmbeddingConfig(),
         optimizer=Adagrad(lr=0.01),
-        gconf=GlobalConfig(capacity=capacity),
+        gconf=GlobalConfig(
+            capacity=capacity, spill_capacity=int(args.spill_capacity or 0)
+        ),
         device=device,
         dist_ctx=DistContext.from_default_group(),
     )
@@ -254,7 +285,8 @@ def main():
             "dtype": "bf16",
             "data": "synthetic (uniform ids, random labels, random-init weights)",
             "config": {
-                "model": f"{args.model}-criteo {args.num_sparse}sparse/{args.num_dense}dense dim{dim}",
+                "preset": args.preset,
+                "model": f"{args.model} {args.num_sparse}sparse/{args.num_dense}dense dim{dim}",
                 "global_batch": B * world,
                 "rows": rows_total,
                 "capacity_per_rank": capacity,

@@ -266,7 +266,9 @@ class EmbeddingEngine:
             optimizer = SGD(lr=0.0)
             self.optimizer = optimizer
         self.stores = {
-            dim: make_store(dim, gconf.capacity, optimizer, hyper, device)
+            dim: make_store(
+                dim, gconf.capacity, optimizer, hyper, device, gconf.spill_capacity
+            )
             for dim in self.groups
         }
         self.skipped_grad_signs = 0

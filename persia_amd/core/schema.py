@@ -157,6 +157,7 @@ class GlobalConfig:
 
     job_type: str = "train"  # train | eval | infer
     capacity: int = 1 << 24
+    spill_capacity: int = 0  # host-DRAM rows per rank (0 = spill disabled)
     buckets_pow2: bool = True
     enable_metrics: bool = False
     checkpointing_workers: int = 4

@@ -57,6 +57,46 @@ def row_init(sign: int, dim: int, lo: float, hi: float) -> np.ndarray:
     return (lo + (hi - lo) * _u01_from_u64(u)).astype(np.float32)
 
 
+class HostTier:
+    """Host-DRAM spill tier: rows evicted from the HBM table park here and
+    come back on their next lookup (exclusive cache: a fetch removes the
+    entry, an eviction re-inserts the freshest copy).  Bounded LRU.
+
+    Replaces the reference's remote-CPU parameter servers as the capacity
+    extension beyond HBM (BASELINE config 4: 1e11-row tables)."""
+
+    def __init__(self, capacity: int, row_width: int):
+        from collections import OrderedDict
+
+        self.capacity = capacity
+        self.row_width = row_width
+        self._map: "OrderedDict[int, np.ndarray]" = OrderedDict()
+
+    def __len__(self) -> int:
+        return len(self._map)
+
+    def insert(self, keys: np.ndarray, rows: np.ndarray) -> None:
+        for i, k in enumerate(keys):
+            k = int(k)
+            if k in self._map:
+                self._map.move_to_end(k)
+            self._map[k] = rows[i].copy()
+        while len(self._map) > self.capacity:
+            self._map.popitem(last=False)
+
+    def fetch(self, keys: np.ndarray):
+        """-> (rows f32 [k, row_width], found bool [k]); found entries are
+        removed (they move to the HBM tier)."""
+        rows = np.zeros((len(keys), self.row_width), dtype=np.float32)
+        found = np.zeros(len(keys), dtype=bool)
+        for i, k in enumerate(keys):
+            row = self._map.pop(int(k), None)
+            if row is not None:
+                rows[i] = row
+                found[i] = True
+        return rows, found
+
+
 class EmbeddingStoreBase:
     """One rank's shard for one dim-group of slots."""
 
@@ -67,6 +107,7 @@ class EmbeddingStoreBase:
         optimizer: Optimizer,
         hyper: EmbeddingConfig,
         device: torch.device,
+        spill_capacity: int = 0,
     ):
         self.dim = dim
         self.optimizer = optimizer
@@ -77,6 +118,7 @@ class EmbeddingStoreBase:
         self.n_buckets = _next_pow2(max(1, capacity // BUCKET_SIZE))
         self.n_slots = self.n_buckets * BUCKET_SIZE
         self.tick = 1  # current batch counter (0 reserved)
+        self.spill = HostTier(spill_capacity, self.row_width) if spill_capacity > 0 else None
         # Adam per-group beta powers (persia-common optim.rs:147-216); kept
         # per-store and stepped once per update call (all slot-groups in a
         # store receive every update batch in this architecture).
@@ -127,8 +169,9 @@ class EmbeddingStoreBase:
 class CpuEmbeddingStore(EmbeddingStoreBase):
     """Sequential oracle + CPU execution backend."""
 
-    def __init__(self, dim, capacity, optimizer, hyper, device=torch.device("cpu")):
-        super().__init__(dim, capacity, optimizer, hyper, device)
+    def __init__(self, dim, capacity, optimizer, hyper, device=torch.device("cpu"),
+                 spill_capacity: int = 0):
+        super().__init__(dim, capacity, optimizer, hyper, device, spill_capacity)
         self.keys = np.zeros(self.n_slots, dtype=np.uint64)
         self.ticks = np.zeros(self.n_slots, dtype=np.uint32)
         self.arena = torch.zeros(self.n_slots, self.row_width, dtype=torch.float32)
@@ -179,6 +222,11 @@ class CpuEmbeddingStore(EmbeddingStoreBase):
             return empty, True
         if victim < 0:
             return -1, False  # whole window is current-tick: overflow
+        if self.spill is not None:  # save the victim row to the host tier
+            self.spill.insert(
+                np.array([self.keys[victim]], dtype=np.uint64),
+                self.arena[victim].numpy().reshape(1, -1),
+            )
         self.keys[victim] = k  # evict (bounded-window LRU)
         self.ticks[victim] = tick
         return victim, True
@@ -212,7 +260,14 @@ class CpuEmbeddingStore(EmbeddingStoreBase):
                     if slot < 0:
                         continue  # overflow miss: zeros
                     if is_new:
-                        self._init_row(slot, k)
+                        restored = False
+                        if self.spill is not None:
+                            rows, found = self.spill.fetch(np.array([k], dtype=np.uint64))
+                            if found[0]:
+                                self.arena[slot] = torch.from_numpy(rows[0])
+                                restored = True
+                        if not restored:
+                            self._init_row(slot, k)
                     self.ticks[slot] = tick
                     out[i] = self.arena[slot, : self.dim]
             else:
@@ -293,8 +348,8 @@ class HipEmbeddingStore(EmbeddingStoreBase):
     allocator so it coexists with the dense model); kernels mutate them in
     place."""
 
-    def __init__(self, dim, capacity, optimizer, hyper, device):
-        super().__init__(dim, capacity, optimizer, hyper, device)
+    def __init__(self, dim, capacity, optimizer, hyper, device, spill_capacity: int = 0):
+        super().__init__(dim, capacity, optimizer, hyper, device, spill_capacity)
         from persia_amd.ops import native
 
         self._C = native()
@@ -306,6 +361,31 @@ class HipEmbeddingStore(EmbeddingStoreBase):
         )
         self._skipped = torch.zeros(2, dtype=torch.int32, device=device)  # miss, nan
         self._opt_code = {"sgd": 0, "adagrad": 1, "adam": 2}[optimizer.kind]
+        self._no_evict = (
+            torch.empty(0, dtype=torch.int64, device=device),
+            torch.empty(0, dtype=torch.int32, device=device),
+            torch.empty(0, dtype=torch.float32, device=device),
+        )
+
+    def _evict_buffers(self, n: int):
+        if self.spill is None:
+            return self._no_evict
+        return (
+            torch.zeros(n, dtype=torch.int64, device=self.device),
+            torch.zeros(1, dtype=torch.int32, device=self.device),
+            torch.empty(n, self.row_width, dtype=torch.float32, device=self.device),
+        )
+
+    def _drain_evictions(self, ev):
+        keys_t, count_t, rows_t = ev
+        if keys_t.numel() == 0:
+            return
+        cnt = int(count_t.item())  # sync (spill path only)
+        if cnt == 0:
+            return
+        keys_np = keys_t[:cnt].cpu().numpy().view(np.uint64)
+        rows_np = rows_t[:cnt].cpu().numpy()
+        self.spill.insert(keys_np, rows_np)
 
     def _opt_params(self):
         o = self.optimizer
@@ -329,7 +409,28 @@ class HipEmbeddingStore(EmbeddingStoreBase):
 
     def lookup(self, keys: torch.Tensor, train: bool) -> torch.Tensor:
         tick = self.next_tick()
-        out = torch.empty(keys.numel(), self.dim, dtype=torch.float32, device=self.device)
+        n = keys.numel()
+        ev = self._no_evict
+        if self.spill is not None and train and n:
+            # spill phase 1: restore missing-but-spilled rows into HBM first
+            slots = self._C.store_probe(self.keys, self.ticks, keys, tick)
+            miss_keys = keys[slots < 0]
+            if miss_keys.numel():
+                miss_np = miss_keys.cpu().numpy().view(np.uint64)
+                rows, found = self.spill.fetch(miss_np)
+                if found.any():
+                    ev = self._evict_buffers(n + int(found.sum()))
+                    found_keys = torch.from_numpy(
+                        miss_np[found].view(np.int64).copy()
+                    ).to(self.device)
+                    rows_t = torch.from_numpy(rows[found]).to(self.device)
+                    self._C.store_import(
+                        self.keys, self.ticks, self.arena, found_keys, rows_t,
+                        tick, *ev,
+                    )
+            if ev is self._no_evict:
+                ev = self._evict_buffers(n)
+        out = torch.empty(n, self.dim, dtype=torch.float32, device=self.device)
         lo, hi = self.hyper.emb_initialization
         self._C.store_lookup(
             self.keys,
@@ -345,7 +446,9 @@ class HipEmbeddingStore(EmbeddingStoreBase):
             float(self.hyper.admit_probability),
             float(self.optimizer.state_init(self.dim)),
             self.opt_space,
+            *ev,
         )
+        self._drain_evictions(ev)
         return out
 
     def update_gradients(self, keys: torch.Tensor, grads: torch.Tensor) -> int:
@@ -389,7 +492,11 @@ class HipEmbeddingStore(EmbeddingStoreBase):
             self.device
         )
         tick = self.next_tick()
-        self._C.store_import(self.keys, self.ticks, self.arena, keys_t, rows_t, tick)
+        ev = self._evict_buffers(keys_t.numel())
+        self._C.store_import(
+            self.keys, self.ticks, self.arena, keys_t, rows_t, tick, *ev
+        )
+        self._drain_evictions(ev)
 
     def clear(self) -> None:
         self.keys.zero_()
@@ -403,7 +510,8 @@ def make_store(
     optimizer: Optimizer,
     hyper: EmbeddingConfig,
     device: torch.device,
+    spill_capacity: int = 0,
 ) -> EmbeddingStoreBase:
     if device.type == "cuda":
-        return HipEmbeddingStore(dim, capacity, optimizer, hyper, device)
-    return CpuEmbeddingStore(dim, capacity, optimizer, hyper, device)
+        return HipEmbeddingStore(dim, capacity, optimizer, hyper, device, spill_capacity)
+    return CpuEmbeddingStore(dim, capacity, optimizer, hyper, device, spill_capacity)

@@ -65,7 +65,7 @@ __device__ __forceinline__ long long probe_find(const ull* __restrict__ keys,
 __device__ long long probe_claim(ull* __restrict__ keys,
                                  unsigned* __restrict__ ticks,
                                  int64_t n_buckets, ull k, unsigned cur_tick,
-                                 int* is_new) {
+                                 int* is_new, ull* victim_out) {
   const int64_t mask = n_buckets - 1;
   const int64_t b = (int64_t)(k & (ull)mask);
   for (int attempt = 0; attempt < 16; ++attempt) {
@@ -100,7 +100,9 @@ __device__ long long probe_claim(ull* __restrict__ keys,
     if (victim >= 0) {
       if (atomicCAS((ull*)&keys[victim], victim_key, k) == victim_key) {
         ticks[victim] = cur_tick;
-        *is_new = 1; return victim;
+        *is_new = 1;
+        if (victim_out) *victim_out = victim_key;
+        return victim;
       }
       continue;
     }
@@ -110,18 +112,26 @@ __device__ long long probe_claim(ull* __restrict__ keys,
   return -1;  // give up: treated as a non-admitted miss
 }
 
+// Eviction log (host-DRAM spill tier): a claim that evicts appends the
+// victim's key to evict_keys and remembers the log index so init_gather can
+// copy the victim ROW out before overwriting it — the engine drains the log
+// into the host tier (SURVEY §7.7 spill design).
 __global__ void probe_claim_kernel(ull* __restrict__ table_keys,
                                    unsigned* __restrict__ ticks,
                                    const ull* __restrict__ query,
                                    int64_t n, int64_t n_buckets, int train,
                                    unsigned tick, float admit_prob,
                                    long long* __restrict__ out_slot,
-                                   int* __restrict__ out_new) {
+                                   int* __restrict__ out_new,
+                                   ull* __restrict__ evict_keys,
+                                   int* __restrict__ evict_count,
+                                   long long* __restrict__ out_evict_idx) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   const ull k = query[i];
   long long slot;
   int is_new = 0;
+  ull victim = 0;
   if (train) {
     bool admitted = true;
     if (admit_prob < 1.0f) {
@@ -130,7 +140,8 @@ __global__ void probe_claim_kernel(ull* __restrict__ table_keys,
       admitted = u < (double)admit_prob;
     }
     if (admitted) {
-      slot = probe_claim(table_keys, ticks, n_buckets, k, tick, &is_new);
+      slot = probe_claim(table_keys, ticks, n_buckets, k, tick, &is_new,
+                         evict_keys ? &victim : nullptr);
     } else {
       slot = probe_find(table_keys, n_buckets, k);
     }
@@ -140,6 +151,12 @@ __global__ void probe_claim_kernel(ull* __restrict__ table_keys,
   if (slot >= 0) ticks[slot] = tick;
   out_slot[i] = slot;
   out_new[i] = is_new;
+  long long eidx = -1;
+  if (evict_keys && victim != 0) {
+    eidx = (long long)atomicAdd(evict_count, 1);
+    evict_keys[eidx] = victim;
+  }
+  if (out_evict_idx) out_evict_idx[i] = eidx;
 }
 
 // one wave per key: init freshly claimed rows, then gather emb -> out (f32)
@@ -149,7 +166,9 @@ __global__ void init_gather_kernel(float* __restrict__ arena,
                                    const int* __restrict__ is_new,
                                    float* __restrict__ out, int64_t n, int dim,
                                    int row_width, double lo, double hi,
-                                   float state_init) {
+                                   float state_init,
+                                   const long long* __restrict__ evict_idx,
+                                   float* __restrict__ evict_rows) {
   const int wave = threadIdx.x / PA_WAVE;
   const int lane = threadIdx.x % PA_WAVE;
   const int waves_per_block = blockDim.x / PA_WAVE;
@@ -162,6 +181,11 @@ __global__ void init_gather_kernel(float* __restrict__ arena,
       continue;
     }
     float* row = arena + (int64_t)slot * row_width;
+    if (evict_rows && is_new[i] && evict_idx[i] >= 0) {
+      // spill: save the victim's full row before overwriting it
+      float* spill = evict_rows + evict_idx[i] * row_width;
+      for (int c = lane; c < row_width; c += PA_WAVE) spill[c] = row[c];
+    }
     if (is_new[i]) {
       const uint64_t sign = pa_splitmix64_inv((uint64_t)query[i]);
       const uint64_t seed = pa_init_seed(sign);
@@ -301,15 +325,26 @@ __global__ void import_kernel(ull* __restrict__ table_keys,
                               float* __restrict__ arena,
                               const ull* __restrict__ query,
                               const float* __restrict__ rows, int64_t n,
-                              int row_width, int64_t n_buckets, unsigned tick) {
+                              int row_width, int64_t n_buckets, unsigned tick,
+                              ull* __restrict__ evict_keys,
+                              int* __restrict__ evict_count,
+                              float* __restrict__ evict_rows) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   int is_new;
+  ull victim = 0;
   const long long slot =
-      probe_claim(table_keys, ticks, n_buckets, query[i], tick, &is_new);
+      probe_claim(table_keys, ticks, n_buckets, query[i], tick, &is_new,
+                  evict_keys ? &victim : nullptr);
   if (slot < 0) return;
   ticks[slot] = tick;
   float* dst = arena + (int64_t)slot * row_width;
+  if (evict_keys && victim != 0) {
+    const long long eidx = (long long)atomicAdd(evict_count, 1);
+    evict_keys[eidx] = victim;
+    float* spill = evict_rows + eidx * row_width;
+    for (int c = 0; c < row_width; ++c) spill[c] = dst[c];
+  }
   const float* src = rows + i * row_width;
   for (int c = 0; c < row_width; ++c) dst[c] = src[c];
 }
@@ -403,14 +438,18 @@ void store_lookup(torch::Tensor table_keys, torch::Tensor ticks,
                   torch::Tensor arena, torch::Tensor query, torch::Tensor out,
                   int64_t dim, int64_t train, int64_t tick, double lo,
                   double hi, double admit_prob, double state_init,
-                  int64_t opt_space) {
+                  int64_t opt_space, torch::Tensor evict_keys,
+                  torch::Tensor evict_count, torch::Tensor evict_rows) {
   const int64_t n = query.numel();
   if (n == 0) return;
   const int64_t n_buckets = table_keys.numel() / PA_BUCKET_SIZE;
   const int row_width = (int)(dim + opt_space);
+  const bool spill = evict_keys.numel() > 0;
   auto opts = torch::TensorOptions().dtype(torch::kInt64).device(query.device());
   auto slots = torch::empty({n}, opts);
   auto is_new = torch::empty({n}, opts.dtype(torch::kInt32));
+  torch::Tensor evict_idx;
+  if (spill) evict_idx = torch::empty({n}, opts);
   hipStream_t st = cur_stream();
   hipLaunchKernelGGL(probe_claim_kernel, dim3(n_blocks_for(n, 256)), dim3(256),
                      0, st, (ull*)table_keys.data_ptr<int64_t>(),
@@ -418,13 +457,36 @@ void store_lookup(torch::Tensor table_keys, torch::Tensor ticks,
                      (const ull*)query.data_ptr<int64_t>(), n, n_buckets,
                      (int)train, (unsigned)tick, (float)admit_prob,
                      (long long*)slots.data_ptr<int64_t>(),
-                     is_new.data_ptr<int32_t>());
+                     is_new.data_ptr<int32_t>(),
+                     spill ? (ull*)evict_keys.data_ptr<int64_t>() : nullptr,
+                     spill ? evict_count.data_ptr<int32_t>() : nullptr,
+                     spill ? (long long*)evict_idx.data_ptr<int64_t>() : nullptr);
   hipLaunchKernelGGL(init_gather_kernel, dim3(n_blocks_for(n, 4)), dim3(256),
                      0, st, arena.data_ptr<float>(),
                      (const ull*)query.data_ptr<int64_t>(),
                      (const long long*)slots.data_ptr<int64_t>(),
                      is_new.data_ptr<int32_t>(), out.data_ptr<float>(), n,
-                     (int)dim, row_width, lo, hi, (float)state_init);
+                     (int)dim, row_width, lo, hi, (float)state_init,
+                     spill ? (const long long*)evict_idx.data_ptr<int64_t>() : nullptr,
+                     spill ? evict_rows.data_ptr<float>() : nullptr);
+}
+
+torch::Tensor store_probe(torch::Tensor table_keys, torch::Tensor ticks,
+                          torch::Tensor query, int64_t tick) {
+  const int64_t n = query.numel();
+  auto opts = torch::TensorOptions().dtype(torch::kInt64).device(query.device());
+  auto slots = torch::empty({n}, opts);
+  if (n == 0) return slots;
+  auto is_new = torch::empty({n}, opts.dtype(torch::kInt32));
+  const int64_t n_buckets = table_keys.numel() / PA_BUCKET_SIZE;
+  hipLaunchKernelGGL(probe_claim_kernel, dim3(n_blocks_for(n, 256)), dim3(256),
+                     0, cur_stream(), (ull*)table_keys.data_ptr<int64_t>(),
+                     (unsigned*)ticks.data_ptr<int32_t>(),
+                     (const ull*)query.data_ptr<int64_t>(), n, n_buckets,
+                     /*train=*/0, (unsigned)tick, 1.0f,
+                     (long long*)slots.data_ptr<int64_t>(),
+                     is_new.data_ptr<int32_t>(), nullptr, nullptr, nullptr);
+  return slots;
 }
 
 void store_update(torch::Tensor table_keys, torch::Tensor ticks,
@@ -452,17 +514,22 @@ void store_update(torch::Tensor table_keys, torch::Tensor ticks,
 
 void store_import(torch::Tensor table_keys, torch::Tensor ticks,
                   torch::Tensor arena, torch::Tensor query, torch::Tensor rows,
-                  int64_t tick) {
+                  int64_t tick, torch::Tensor evict_keys,
+                  torch::Tensor evict_count, torch::Tensor evict_rows) {
   const int64_t n = query.numel();
   if (n == 0) return;
   const int64_t n_buckets = table_keys.numel() / PA_BUCKET_SIZE;
+  const bool spill = evict_keys.numel() > 0;
   hipLaunchKernelGGL(import_kernel, dim3(n_blocks_for(n, 256)), dim3(256), 0,
                      cur_stream(), (ull*)table_keys.data_ptr<int64_t>(),
                      (unsigned*)ticks.data_ptr<int32_t>(),
                      arena.data_ptr<float>(),
                      (const ull*)query.data_ptr<int64_t>(),
                      rows.data_ptr<float>(), n, (int)rows.size(1), n_buckets,
-                     (unsigned)tick);
+                     (unsigned)tick,
+                     spill ? (ull*)evict_keys.data_ptr<int64_t>() : nullptr,
+                     spill ? evict_count.data_ptr<int32_t>() : nullptr,
+                     spill ? evict_rows.data_ptr<float>() : nullptr);
 }
 
 torch::Tensor segment_sum(torch::Tensor rows, torch::Tensor inverse,
@@ -540,6 +607,7 @@ torch::Tensor sign_prep(torch::Tensor values, torch::Tensor slot_starts,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "persia_amd HIP kernels (gfx950)";
   m.def("store_lookup", &store_lookup, "hash-table lookup/insert + gather");
+  m.def("store_probe", &store_probe, "probe-only (spill-tier miss detection)");
   m.def("store_update", &store_update, "fused sparse optimizer update");
   m.def("store_import", &store_import, "bulk insert rows (checkpoint load)");
   m.def("segment_sum", &segment_sum, "fused gather + per-sample summation");

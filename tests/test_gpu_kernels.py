@@ -128,6 +128,40 @@ def test_store_eviction_bounded():
     assert present >= min(len(chunks[-1]), n_slots) // 2
 
 
+def test_spill_roundtrip_gpu():
+    """Evicted rows park in host DRAM and come back with optimizer state."""
+    from persia_amd.core.store import BUCKET_SIZE, PROBE_BUCKETS, HipEmbeddingStore
+    from persia_amd.embedding import EmbeddingConfig
+    from persia_amd.embedding.optim import Adagrad
+
+    store = HipEmbeddingStore(
+        4, BUCKET_SIZE * PROBE_BUCKETS, Adagrad(lr=0.1, initial_accumulator_value=0.01),
+        EmbeddingConfig(emb_initialization=(-0.5, 0.5)), _dev(), spill_capacity=10000,
+    )
+    k7 = _keys([7]).to(_dev())
+    store.lookup(k7, train=True)
+    store.update_gradients(k7, torch.full((1, 4), 2.0, device=_dev()))
+    updated = store.lookup(k7, train=False).cpu().clone()
+
+    from persia_amd.core import hashing as H
+
+    key7 = int(H.splitmix64(np.array([7], np.uint64)).view(np.int64)[0])
+    sign = 1000
+    while True:
+        slots = store._C.store_probe(
+            store.keys, store.ticks,
+            torch.tensor([key7], dtype=torch.int64, device=_dev()), store.tick
+        )
+        if int(slots.item()) < 0:
+            break
+        store.lookup(_keys(list(range(sign, sign + 8))).to(_dev()), train=True)
+        sign += 8
+        assert sign < 100000, "sign 7 never evicted?"
+    assert len(store.spill) > 0
+    restored = store.lookup(k7, train=True).cpu()
+    assert torch.equal(restored, updated)
+
+
 def test_segment_sum_matches_reference():
     from persia_amd.ops import native
     from persia_amd.ops import reference as R

@@ -1,0 +1,63 @@
+"""Host-DRAM spill tier semantics (CPU model; the HIP path is covered by
+tests/test_gpu_kernels.py::test_spill_roundtrip_gpu)."""
+import numpy as np
+import torch
+
+from persia_amd.core import hashing
+from persia_amd.core.store import BUCKET_SIZE, PROBE_BUCKETS, CpuEmbeddingStore
+from persia_amd.embedding import EmbeddingConfig
+from persia_amd.embedding.optim import Adagrad
+
+
+def _keys(signs):
+    h = hashing.splitmix64(np.asarray(signs, dtype=np.uint64))
+    return torch.from_numpy(h.view(np.int64))
+
+
+def test_evicted_row_survives_in_host_tier():
+    store = CpuEmbeddingStore(
+        4,
+        BUCKET_SIZE * PROBE_BUCKETS,  # one probe window = whole table
+        Adagrad(lr=0.1, initial_accumulator_value=0.01),
+        EmbeddingConfig(emb_initialization=(-0.5, 0.5)),
+        spill_capacity=10_000,
+    )
+    k7 = _keys([7])
+    before = store.lookup(k7, train=True).clone()
+    store.update_gradients(k7, torch.full((1, 4), 2.0))
+    updated = store.lookup(k7, train=False).clone()
+    assert not torch.equal(before, updated)
+
+    # flood the table with other signs until sign 7 gets evicted
+    sign = 1000
+    while store._probe(np.uint64(int(hashing.splitmix64(np.array([7], np.uint64))[0]))) >= 0:
+        store.lookup(_keys(list(range(sign, sign + 8))), train=True)
+        sign += 8
+        assert sign < 100000, "sign 7 never evicted?"
+    assert len(store.spill) > 0
+
+    # next training lookup restores the UPDATED row (incl. optimizer state)
+    restored = store.lookup(k7, train=True)
+    assert torch.equal(restored, updated)
+    # and the accumulator survived: identical second update on both paths
+    store2 = CpuEmbeddingStore(
+        4, 1 << 12, Adagrad(lr=0.1, initial_accumulator_value=0.01),
+        EmbeddingConfig(emb_initialization=(-0.5, 0.5)),
+    )
+    store2.lookup(k7, train=True)
+    store2.update_gradients(k7, torch.full((1, 4), 2.0))
+    store.update_gradients(k7, torch.full((1, 4), 0.5))
+    store2.update_gradients(k7, torch.full((1, 4), 0.5))
+    assert torch.allclose(
+        store.lookup(k7, train=False), store2.lookup(k7, train=False), atol=1e-6
+    )
+
+
+def test_spill_capacity_bounded():
+    store = CpuEmbeddingStore(
+        2, BUCKET_SIZE * PROBE_BUCKETS, Adagrad(lr=0.1), EmbeddingConfig(),
+        spill_capacity=16,
+    )
+    for s in range(0, 2000, 8):
+        store.lookup(_keys(list(range(s, s + 8))), train=True)
+    assert len(store.spill) <= 16
