@@ -174,7 +174,7 @@ def main():
     # replays (the dense side was launch-bound, ~200 kernels/step).
     graph = None
     static = {}
-    if args.graph and use_gpu and world == 1 and args.model == "dlrm":
+    if args.graph and use_gpu and world == 1:
         try:
             static = {
                 "dense": torch.zeros(B, args.num_dense, device=device),

@@ -36,9 +36,11 @@ EMPTY_KEY = np.uint64(0)
 _ZERO_REMAP = np.uint64(0xD1B54A32D192ED03)  # mixed-key 0 is remapped here
 
 
-def _next_pow2(x: int) -> int:
+def _floor_pow2(x: int) -> int:
+    """Largest power of two <= x (capacity is a BUDGET: never round up —
+    a ceil on a 2.6e9-row request would double a 128 GB arena)."""
     p = 1
-    while p < x:
+    while (p << 1) <= x:
         p <<= 1
     return p
 
@@ -115,7 +117,7 @@ class EmbeddingStoreBase:
         self.device = device
         self.opt_space = optimizer.require_space(dim)
         self.row_width = dim + self.opt_space
-        self.n_buckets = _next_pow2(max(1, capacity // BUCKET_SIZE))
+        self.n_buckets = _floor_pow2(max(1, capacity // BUCKET_SIZE))
         self.n_slots = self.n_buckets * BUCKET_SIZE
         self.tick = 1  # current batch counter (0 reserved)
         self.spill = HostTier(spill_capacity, self.row_width) if spill_capacity > 0 else None

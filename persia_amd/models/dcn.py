@@ -35,11 +35,24 @@ class DCNv2(nn.Module):
         self.deep = nn.Sequential(*layers)
         self.head = nn.Linear(in_dim + sizes[-1], 1)
 
-    def forward(
-        self, non_id_tensors: List[torch.Tensor], embedding_tensors: List[torch.Tensor]
-    ) -> torch.Tensor:
-        dense = non_id_tensors[0].float()
-        x0 = torch.cat([dense] + [e.flatten(1).float() for e in embedding_tensors], dim=1)
+    def forward(self, non_id_tensors, embedding_tensors) -> torch.Tensor:
+        dense = (
+            non_id_tensors.float()
+            if torch.is_tensor(non_id_tensors)
+            else non_id_tensors[0].float()
+        )
+        if torch.is_tensor(embedding_tensors):
+            # packed slot-major [S*B, D] from the engine's fused sum output
+            B = dense.shape[0]
+            S = embedding_tensors.shape[0] // B
+            emb = (
+                embedding_tensors.view(S, B, -1).permute(1, 0, 2).reshape(B, -1).float()
+            )
+            x0 = torch.cat([dense, emb], dim=1)
+        else:
+            x0 = torch.cat(
+                [dense] + [e.flatten(1).float() for e in embedding_tensors], dim=1
+            )
         xl = x0
         for layer in self.cross:
             xl = layer(x0, xl)

@@ -145,19 +145,14 @@ def test_spill_roundtrip_gpu():
 
     from persia_amd.core import hashing as H
 
-    key7 = int(H.splitmix64(np.array([7], np.uint64)).view(np.int64)[0])
+    key7 = int(H.splitmix64(np.array([7], np.uint64))[0])
     sign = 1000
-    while True:
-        slots = store._C.store_probe(
-            store.keys, store.ticks,
-            torch.tensor([key7], dtype=torch.int64, device=_dev()), store.tick
-        )
-        if int(slots.item()) < 0:
-            break
+    # flood until key 7 lands in the host tier (don't probe it — probing
+    # refreshes its tick and protects it from eviction)
+    while key7 not in store.spill._map:
         store.lookup(_keys(list(range(sign, sign + 8))).to(_dev()), train=True)
         sign += 8
         assert sign < 100000, "sign 7 never evicted?"
-    assert len(store.spill) > 0
     restored = store.lookup(k7, train=True).cpu()
     assert torch.equal(restored, updated)
 
