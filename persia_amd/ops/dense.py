@@ -1,0 +1,97 @@
+"""Fused dense layers over the CDNA4 MFMA kernels (csrc/dense.hip).
+
+``FusedLinear`` = bf16 GEMM + bias + ReLU in ONE kernel (forward), with
+hand-written dgrad / wgrad / bias-grad kernels in backward — replacing a
+hipBLASLt GEMM plus an elementwise cascade per layer.  f32 master weights,
+bf16 activations (the bench's dtype contract).
+
+Constraints (asserted): hidden widths N % 32 == 0 (dgrad reuses the GEMM
+kernel with K=N); input features are zero-padded to a multiple of 32.
+"""
+from typing import List, Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+def _pad32(x: torch.Tensor, dim: int) -> torch.Tensor:
+    k = x.shape[dim]
+    pad = (-k) % 32
+    if pad == 0:
+        return x
+    padding = [0, 0] * (x.dim() - 1 - dim) + [0, pad]
+    return F.pad(x, padding)
+
+
+class FusedLinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, act: int):
+        from persia_amd.ops import native
+
+        C = native()
+        x_bf = _pad32(x.to(torch.bfloat16), 1).contiguous()
+        w_bf = _pad32(weight.to(torch.bfloat16), 1).contiguous()  # [N, Kp]
+        w_t = w_bf.t().contiguous()  # [Kp, N]
+        out = C.gemm_bias_act(x_bf, w_t, bias.float(), act, 0)
+        ctx.save_for_backward(x_bf, w_bf, out)
+        ctx.act = act
+        ctx.k = x.shape[1]
+        return out
+
+    @staticmethod
+    def backward(ctx, g):
+        from persia_amd.ops import native
+
+        C = native()
+        x_bf, w_bf, out = ctx.saved_tensors
+        g = g.to(torch.bfloat16).contiguous()
+        if ctx.act == 1:
+            g = C.relu_bwd(g, out)
+        db = C.bias_grad(g)
+        # dX = g @ W  (W [N, Kp] row-major IS the [N->Kp] operand)
+        dx = C.gemm_bias_act(g, w_bf, torch.empty(0, device=g.device), 0, 0)
+        if dx.shape[1] != ctx.k:
+            dx = dx[:, : ctx.k]
+        dw = C.wgrad(g, x_bf)
+        if dw.shape[1] != ctx.k:
+            dw = dw[:, : ctx.k].contiguous()
+        return dx, dw, db, None
+
+
+class FusedLinear(nn.Module):
+    def __init__(self, in_features: int, out_features: int, relu: bool = True):
+        super().__init__()
+        assert out_features % 32 == 0, "FusedLinear needs out_features % 32 == 0"
+        self.in_features = in_features
+        self.out_features = out_features
+        self.relu = relu
+        self.weight = nn.Parameter(torch.empty(out_features, in_features))
+        self.bias = nn.Parameter(torch.zeros(out_features))
+        nn.init.kaiming_uniform_(self.weight, a=5 ** 0.5)
+
+    def forward(self, x):
+        return FusedLinearFn.apply(x, self.weight, self.bias, 1 if self.relu else 0)
+
+
+class FusedMLP(nn.Module):
+    """Stack of FusedLinear layers (+ optional plain final projection when the
+    last width isn't a multiple of 32, e.g. the 1-logit head)."""
+
+    def __init__(self, sizes: List[int], last_relu: bool = False):
+        super().__init__()
+        layers: List[nn.Module] = []
+        for i in range(len(sizes) - 1):
+            is_last = i == len(sizes) - 2
+            relu = (not is_last) or last_relu
+            if sizes[i + 1] % 32 == 0:
+                layers.append(FusedLinear(sizes[i], sizes[i + 1], relu=relu))
+            else:
+                lin = nn.Linear(sizes[i], sizes[i + 1])
+                layers.append(lin if not relu else nn.Sequential(lin, nn.ReLU()))
+        self.layers = nn.ModuleList(layers)
+
+    def forward(self, x):
+        for l in self.layers:
+            x = l(x)
+        return x

@@ -1,0 +1,107 @@
+"""Numerics of the fused MFMA dense kernels vs plain-torch fp32 references."""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _dev():
+    return torch.device("cuda", 0)
+
+
+@pytest.mark.parametrize("shape", [(256, 64, 128), (512, 96, 256), (4096, 480, 1024)])
+@pytest.mark.parametrize("act", [0, 1])
+def test_gemm_bias_act_matches_torch(shape, act):
+    from persia_amd.ops import native
+
+    C = native()
+    M, K, N = shape
+    torch.manual_seed(0)
+    # asymmetric operands (transpose-detecting — guide §5.4 rule 16)
+    A = (torch.randn(M, K, device=_dev()) * 0.5).to(torch.bfloat16)
+    B = (torch.randn(K, N, device=_dev()) * 0.5).to(torch.bfloat16)
+    bias = torch.randn(N, device=_dev())
+    out = C.gemm_bias_act(A.contiguous(), B.contiguous(), bias, act, 0)
+    ref = A.float() @ B.float() + bias
+    if act == 1:
+        ref = torch.relu(ref)
+    assert torch.allclose(out.float(), ref, atol=0.1 + 0.02 * np.sqrt(K), rtol=0.02), (
+        f"max err {(out.float() - ref).abs().max()}"
+    )
+
+
+def test_wgrad_matches_torch():
+    from persia_amd.ops import native
+
+    C = native()
+    torch.manual_seed(1)
+    for M, N, K in [(4096, 512, 480), (1024, 64, 32), (4096, 1024, 1024)]:
+        dC = (torch.randn(M, N, device=_dev()) * 0.1).to(torch.bfloat16)
+        A = (torch.randn(M, K, device=_dev()) * 0.1).to(torch.bfloat16)
+        dW = C.wgrad(dC.contiguous(), A.contiguous())
+        ref = dC.float().t() @ A.float()
+        assert torch.allclose(dW, ref, atol=0.5, rtol=0.02), (
+            f"{M}x{N}x{K}: max err {(dW - ref).abs().max()}"
+        )
+
+
+def test_bias_grad_and_relu_bwd():
+    from persia_amd.ops import native
+
+    C = native()
+    torch.manual_seed(2)
+    dC = torch.randn(1000, 256, device=_dev()).to(torch.bfloat16)
+    db = C.bias_grad(dC.contiguous())
+    assert torch.allclose(db, dC.float().sum(0), atol=0.5, rtol=0.02)
+    out = torch.randn(1000, 256, device=_dev()).to(torch.bfloat16)
+    geff = C.relu_bwd(dC.contiguous(), out.contiguous())
+    ref = dC.float() * (out.float() > 0)
+    assert torch.allclose(geff.float(), ref, atol=1e-2)
+
+
+def test_fused_linear_autograd_matches_torch():
+    from persia_amd.ops.dense import FusedLinearFn
+
+    torch.manual_seed(3)
+    M, K, N = 512, 200, 256  # K not %32: exercises padding
+    x = torch.randn(M, K, device=_dev(), dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(N, K, device=_dev(), requires_grad=True) * 0.05
+    w.retain_grad()
+    b = torch.randn(N, device=_dev(), requires_grad=True)
+    out = FusedLinearFn.apply(x, w, b, 1)
+    g = torch.randn_like(out.float()).to(torch.bfloat16)
+    out.backward(g)
+
+    x2 = x.detach().float().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    ref = torch.relu(x2 @ w2.t() + b2)
+    ref.backward(g.float())
+    assert torch.allclose(out.float(), ref, atol=0.3, rtol=0.05)
+    assert torch.allclose(x.grad.float(), x2.grad, atol=0.3, rtol=0.05), (
+        f"dx max err {(x.grad.float() - x2.grad).abs().max()}"
+    )
+    assert torch.allclose(w.grad, w2.grad, atol=0.5, rtol=0.05), (
+        f"dw max err {(w.grad - w2.grad).abs().max()}"
+    )
+    assert torch.allclose(b.grad, b2.grad, atol=0.5, rtol=0.05)
+
+
+def test_fused_mlp_learns():
+    from persia_amd.ops.dense import FusedMLP
+
+    torch.manual_seed(4)
+    mlp = FusedMLP([64, 256, 128, 32], last_relu=False).to(_dev())
+    opt = torch.optim.SGD(mlp.parameters(), lr=0.05)
+    x = torch.randn(2048, 64, device=_dev(), dtype=torch.bfloat16)
+    target = torch.randn(2048, 32, device=_dev())
+    losses = []
+    for _ in range(50):
+        out = mlp(x)
+        loss = ((out.float() - target) ** 2).mean()
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0] * 0.7, losses[::10]
