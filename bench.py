@@ -54,6 +54,8 @@ def parse_args():
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--graph", type=int, default=1,
                    help="capture dense fwd+bwd in a hipGraph (1 GPU only)")
+    p.add_argument("--fused-dense", type=int, default=1,
+                   help="use the hand-written MFMA GEMM+bias+ReLU layers")
     args = p.parse_args()
     preset = dict(PRESETS[args.preset])
     preset.setdefault("batch_size", 4096)
@@ -117,7 +119,10 @@ def main():
         dist_ctx=DistContext.from_default_group(),
     )
     if args.model == "dlrm":
-        model = DLRM(num_sparse=n_slots, num_dense=args.num_dense, dim=dim).to(device)
+        model = DLRM(
+            num_sparse=n_slots, num_dense=args.num_dense, dim=dim,
+            fused=bool(args.fused_dense) and use_gpu,
+        ).to(device)
     else:
         model = DCNv2(num_sparse=n_slots, num_dense=args.num_dense, dim=dim).to(device)
     if world > 1:

@@ -41,15 +41,23 @@ class DLRM(nn.Module):
         dim: int = 128,
         bottom_mlp: List[int] = (512, 256, 128),
         top_mlp: List[int] = (1024, 1024, 512, 256),
+        fused: bool = False,
     ):
         super().__init__()
         self.dim = dim
         self.num_sparse = num_sparse
-        self.bottom = _mlp([num_dense] + list(bottom_mlp) + [dim], last_relu=True)
-        self.interaction = DotInteraction()
         n_f = num_sparse + 1
         inter_out = n_f * (n_f - 1) // 2
-        self.top = _mlp([dim + inter_out] + list(top_mlp) + [1])
+        if fused:
+            # hand-written CDNA4 MFMA GEMM+bias+ReLU layers (GPU only)
+            from persia_amd.ops.dense import FusedMLP
+
+            self.bottom = FusedMLP([num_dense] + list(bottom_mlp) + [dim], last_relu=True)
+            self.top = FusedMLP([dim + inter_out] + list(top_mlp) + [1])
+        else:
+            self.bottom = _mlp([num_dense] + list(bottom_mlp) + [dim], last_relu=True)
+            self.top = _mlp([dim + inter_out] + list(top_mlp) + [1])
+        self.interaction = DotInteraction()
 
     def forward(
         self, non_id_tensors, embedding_tensors
