@@ -32,8 +32,7 @@ class FusedLinearFn(torch.autograd.Function):
         C = native()
         x_bf = _pad32(x.to(torch.bfloat16), 1).contiguous()
         w_bf = _pad32(weight.to(torch.bfloat16), 1).contiguous()  # [N, Kp]
-        w_t = w_bf.t().contiguous()  # [Kp, N]
-        out = C.gemm_bias_act(x_bf, w_t, bias.float(), act, 0)
+        out = C.gemm_nt_bias_act(x_bf, w_bf, bias.float(), act, 0)
         ctx.save_for_backward(x_bf, w_bf, out)
         ctx.act = act
         ctx.k = x.shape[1]
@@ -49,8 +48,10 @@ class FusedLinearFn(torch.autograd.Function):
         if ctx.act == 1:
             g = C.relu_bwd(g, out)
         db = C.bias_grad(g)
-        # dX = g @ W  (W [N, Kp] row-major IS the [N->Kp] operand)
-        dx = C.gemm_bias_act(g, w_bf, torch.empty(0, device=g.device), 0, 0)
+        # dX = g @ W = g @ (W^T)^T — NT kernel with the [Kp, N] transpose
+        dx = C.gemm_nt_bias_act(
+            g, w_bf.t().contiguous(), torch.empty(0, device=g.device), 0, 0
+        )
         if dx.shape[1] != ctx.k:
             dx = dx[:, : ctx.k]
         dw = C.wgrad(g, x_bf)

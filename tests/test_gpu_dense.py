@@ -12,7 +12,7 @@ def _dev():
 
 @pytest.mark.parametrize("shape", [(256, 64, 128), (512, 96, 256), (4096, 480, 1024)])
 @pytest.mark.parametrize("act", [0, 1])
-def test_gemm_bias_act_matches_torch(shape, act):
+def test_gemm_nt_bias_act_matches_torch(shape, act):
     from persia_amd.ops import native
 
     C = native()
@@ -20,10 +20,10 @@ def test_gemm_bias_act_matches_torch(shape, act):
     torch.manual_seed(0)
     # asymmetric operands (transpose-detecting — guide §5.4 rule 16)
     A = (torch.randn(M, K, device=_dev()) * 0.5).to(torch.bfloat16)
-    B = (torch.randn(K, N, device=_dev()) * 0.5).to(torch.bfloat16)
+    B = (torch.randn(N, K, device=_dev()) * 0.5).to(torch.bfloat16)
     bias = torch.randn(N, device=_dev())
-    out = C.gemm_bias_act(A.contiguous(), B.contiguous(), bias, act, 0)
-    ref = A.float() @ B.float() + bias
+    out = C.gemm_nt_bias_act(A.contiguous(), B.contiguous(), bias, act, 0)
+    ref = A.float() @ B.float().t() + bias
     if act == 1:
         ref = torch.relu(ref)
     assert torch.allclose(out.float(), ref, atol=0.1 + 0.02 * np.sqrt(K), rtol=0.02), (
