@@ -213,7 +213,10 @@ def main():
             with torch.cuda.graph(graph):
                 static["loss"] = iteration()
         except Exception as e:  # pragma: no cover - fall back to eager
-            print(f"# hipGraph capture failed ({e}); running eager", flush=True)
+            import sys as _sys
+
+            print(f"# hipGraph capture failed ({e}); running eager",
+                  file=_sys.stderr, flush=True)
             graph = None
 
     def train_step(tb):
