@@ -233,14 +233,18 @@ __global__ void relu_bwd_kernel(const short* __restrict__ g,
   *(bf16x8*)(g_eff + i) = r;
 }
 
+// db[n] = sum_m dC[m][n]: each block reduces a chunk of rows over ALL cols
+// (threads stride the row -> fully coalesced); one atomicAdd per (block,col).
 __global__ void bias_grad_kernel(const short* __restrict__ dC,
-                                 float* __restrict__ db, int M, int N) {
-  const int col = blockIdx.x * 128 + (threadIdx.x % 128);
-  if (col >= N) return;
-  const int part = threadIdx.x / 128;  // 2 row partitions
-  float acc = 0.0f;
-  for (int m = part; m < M; m += 2) acc += bf2f(dC[(int64_t)m * N + col]);
-  atomicAdd(&db[col], acc);
+                                 float* __restrict__ db, int M, int N,
+                                 int rows_per_block) {
+  const int m_begin = blockIdx.x * rows_per_block;
+  const int m_end = min(M, m_begin + rows_per_block);
+  for (int col = threadIdx.x; col < N; col += blockDim.x) {
+    float acc = 0.0f;
+    for (int m = m_begin; m < m_end; ++m) acc += bf2f(dC[(int64_t)m * N + col]);
+    atomicAdd(&db[col], acc);
+  }
 }
 
 }  // namespace
@@ -295,9 +299,11 @@ torch::Tensor bias_grad(torch::Tensor dC) {
   const int M = (int)dC.size(0), N = (int)dC.size(1);
   auto db = torch::zeros(
       {N}, torch::TensorOptions().dtype(torch::kFloat32).device(dC.device()));
-  hipLaunchKernelGGL(bias_grad_kernel, dim3((N + 127) / 128), dim3(256), 0,
+  const int rows_per_block = std::max(8, (M + 511) / 512);
+  const int blocks = (M + rows_per_block - 1) / rows_per_block;
+  hipLaunchKernelGGL(bias_grad_kernel, dim3(blocks), dim3(256), 0,
                      dcur_stream(), (const short*)dC.data_ptr(),
-                     db.data_ptr<float>(), M, N);
+                     db.data_ptr<float>(), M, N, rows_per_block);
   return db;
 }
 
