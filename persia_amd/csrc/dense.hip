@@ -47,30 +47,39 @@ __device__ __forceinline__ short f2bf(float f) {
 //   C frag: lane holds C[row = (lane>>4)*4 + r][col = lane&15]
 // Both fragments read 8 contiguous shorts from a [row][k] LDS image.
 // ---------------------------------------------------------------------------
-template <int ACT, bool STORE_F32>
+// TBN: output tile N-width (128, or 64 for skinny layers so the grid fills
+// 256 CUs).  TRANS_B: second operand given as [Kdim, N] row-major (the dgrad
+// case dX = g @ W with W [N,K]: K here is the reduction N of the forward) —
+// staged through a transposed LDS image (scalar writes) so fragment reads
+// stay contiguous.
+template <int ACT, bool STORE_F32, bool TRANS_B, int TBN>
 __global__ __launch_bounds__(256) void gemm_nt_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     const float* __restrict__ bias, void* __restrict__ C, int M, int N,
     int K) {
   __shared__ short lds_a[2][BM * LDK];
-  __shared__ short lds_b[2][BN * LDK];
+  __shared__ short lds_b[2][TBN * LDK];
+  constexpr int WN = TBN / 2;  // per-wave N extent
+  constexpr int NFRAG = WN / 16;
 
-  const int n_tiles_n = (N + BN - 1) / BN;
+  const int n_tiles_n = (N + TBN - 1) / TBN;
   const int m0 = (blockIdx.x / n_tiles_n) * BM;
-  const int n0 = (blockIdx.x % n_tiles_n) * BN;
+  const int n0 = (blockIdx.x % n_tiles_n) * TBN;
 
   const int tid = threadIdx.x;
   const int lane = tid % 64;
   const int wave = tid / 64;
-  const int wr = wave / 2, wc = wave % 2;  // 64x64 wave tile position
+  const int wr = wave / 2, wc = wave % 2;  // 64 x WN wave tile position
   const int fi = lane & 15;
   const int fk8 = (lane >> 4) * 8;
 
-  f32x4 acc[4][4] = {};
+  f32x4 acc[4][NFRAG] = {};
 
-  // staging: 256 threads, each copies one 8-short piece per row-half.
-  // A rows: 128 of BK=32 shorts -> 4 pieces/row -> 512 pieces -> 2 rounds.
+  // A staging: 256 threads x 8-short pieces; 128 rows x BK/8 pieces.
   const int s_r = tid / 4, s_c8 = (tid % 4) * 8;  // 64 rows per round
+  // B staging (TRANS_B): B mem rows are the k-reduction axis: B[k0+r][n0+c];
+  // write transposed into lds_b[n][k].
+  const int t_r = tid / 8, t_c8 = (tid % 8) * 8;  // 32 k-rows x 64 cols
 
   auto stage = [&](int buf, int k0) {
 #pragma unroll
@@ -80,10 +89,28 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
       if (m0 + r < M)
         v = *(const bf16x8*)(A + (int64_t)(m0 + r) * K + k0 + s_c8);
       *(bf16x8*)&lds_a[buf][r * LDK + s_c8] = v;
-      bf16x8 w{};
-      if (n0 + r < N)
-        w = *(const bf16x8*)(B + (int64_t)(n0 + r) * K + k0 + s_c8);
-      *(bf16x8*)&lds_b[buf][r * LDK + s_c8] = w;
+    }
+    if (!TRANS_B) {
+#pragma unroll
+      for (int half = 0; half < TBN / 64; ++half) {
+        const int r = s_r + half * 64;  // B row (n)
+        bf16x8 w{};
+        if (n0 + r < N)
+          w = *(const bf16x8*)(B + (int64_t)(n0 + r) * K + k0 + s_c8);
+        *(bf16x8*)&lds_b[buf][r * LDK + s_c8] = w;
+      }
+    } else {
+      // rows = k, cols = n; 32 x TBN tile, transposed write
+#pragma unroll
+      for (int half = 0; half < TBN / 64; ++half) {
+        const int c = t_c8 + half * 64;
+        bf16x8 w{};
+        if (n0 + c < N && t_c8 < 64)
+          w = *(const bf16x8*)(B + (int64_t)(k0 + t_r) * N + n0 + c);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          lds_b[buf][(c + j) * LDK + t_r] = w[j];
+      }
     }
   };
 
@@ -98,9 +125,9 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
       const bf16x8 a_frag =
           *(const bf16x8*)&lds_a[buf][(wr * 64 + am * 16 + fi) * LDK + fk8];
 #pragma unroll
-      for (int bn = 0; bn < 4; ++bn) {
+      for (int bn = 0; bn < NFRAG; ++bn) {
         const bf16x8 b_frag =
-            *(const bf16x8*)&lds_b[buf][(wc * 64 + bn * 16 + fi) * LDK + fk8];
+            *(const bf16x8*)&lds_b[buf][(wc * WN + bn * 16 + fi) * LDK + fk8];
         acc[am][bn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a_frag, b_frag, acc[am][bn], 0, 0, 0);
       }
@@ -112,8 +139,8 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
 #pragma unroll
   for (int am = 0; am < 4; ++am) {
 #pragma unroll
-    for (int bn = 0; bn < 4; ++bn) {
-      const int col = n0 + wc * 64 + bn * 16 + fi;
+    for (int bn = 0; bn < NFRAG; ++bn) {
+      const int col = n0 + wc * WN + bn * 16 + fi;
       if (col >= N) continue;
       const float bval = bias ? bias[col] : 0.0f;
 #pragma unroll
@@ -255,30 +282,47 @@ static hipStream_t dcur_stream() {
   return at::hip::getCurrentHIPStream().stream();
 }
 
-// C = act(A @ B^T + bias); A [M,K] bf16, B [N,K] bf16 (torch weight layout)
+// C = act(A @ op(B) + bias); A [M,K] bf16.
+// trans_b == 0: B is [N, K] (torch weight layout), C = A @ B^T.
+// trans_b == 1: B is [K, N] row-major, C = A @ B (the dgrad case: pass the
+//               [N_layer, K_layer] weight directly as the [K,N] operand).
 torch::Tensor gemm_nt_bias_act(torch::Tensor A, torch::Tensor B,
                                torch::Tensor bias, int64_t act,
-                               int64_t out_f32) {
+                               int64_t out_f32, int64_t trans_b) {
   TORCH_CHECK(A.scalar_type() == torch::kBFloat16 &&
               B.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(A.is_contiguous() && B.is_contiguous());
-  const int M = (int)A.size(0), K = (int)A.size(1), N = (int)B.size(0);
-  TORCH_CHECK(B.size(1) == K, "gemm_nt shape mismatch");
+  const int M = (int)A.size(0), K = (int)A.size(1);
+  const int N = (int)(trans_b ? B.size(1) : B.size(0));
+  TORCH_CHECK((trans_b ? B.size(0) : B.size(1)) == K, "gemm_nt shape mismatch");
   TORCH_CHECK(K % BK == 0, "K must be a multiple of 32 (pad)");
+  TORCH_CHECK(N % 8 == 0, "N must be a multiple of 8");
   auto C = torch::empty(
       {M, N},
       torch::TensorOptions()
           .dtype(out_f32 ? torch::kFloat32 : torch::kBFloat16)
           .device(A.device()));
-  const int grid = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
+  const int m_tiles = (M + BM - 1) / BM;
+  const bool narrow = (int64_t)m_tiles * ((N + 127) / 128) < 192 && N % 64 == 0;
+  const int tbn = narrow ? 64 : 128;
+  const int grid = m_tiles * ((N + tbn - 1) / tbn);
   const float* bias_ptr = bias.numel() ? bias.data_ptr<float>() : nullptr;
   hipStream_t st = dcur_stream();
-#define PA_GEMM(ACTV, F32V)                                                   \
-  hipLaunchKernelGGL((gemm_nt_kernel<ACTV, F32V>), dim3(grid), dim3(256), 0,  \
-                     st, (const bf16*)A.data_ptr(), (const bf16*)B.data_ptr(),\
-                     bias_ptr, C.data_ptr(), M, N, K)
-  if (act == 1) { if (out_f32) PA_GEMM(1, true); else PA_GEMM(1, false); }
-  else          { if (out_f32) PA_GEMM(0, true); else PA_GEMM(0, false); }
+#define PA_GEMM(ACTV, F32V, TBV, TBNV)                                        \
+  hipLaunchKernelGGL((gemm_nt_kernel<ACTV, F32V, TBV, TBNV>), dim3(grid),     \
+                     dim3(256), 0, st, (const bf16*)A.data_ptr(),             \
+                     (const bf16*)B.data_ptr(), bias_ptr, C.data_ptr(), M, N, \
+                     K)
+#define PA_GEMM_T(ACTV, F32V)                                                 \
+  do {                                                                        \
+    if (trans_b) { if (narrow) PA_GEMM(ACTV, F32V, true, 64);                 \
+                   else PA_GEMM(ACTV, F32V, true, 128); }                     \
+    else         { if (narrow) PA_GEMM(ACTV, F32V, false, 64);                \
+                   else PA_GEMM(ACTV, F32V, false, 128); }                    \
+  } while (0)
+  if (act == 1) { if (out_f32) PA_GEMM_T(1, true); else PA_GEMM_T(1, false); }
+  else          { if (out_f32) PA_GEMM_T(0, true); else PA_GEMM_T(0, false); }
+#undef PA_GEMM_T
 #undef PA_GEMM
   return C;
 }

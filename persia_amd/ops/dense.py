@@ -32,7 +32,7 @@ class FusedLinearFn(torch.autograd.Function):
         C = native()
         x_bf = _pad32(x.to(torch.bfloat16), 1).contiguous()
         w_bf = _pad32(weight.to(torch.bfloat16), 1).contiguous()  # [N, Kp]
-        out = C.gemm_nt_bias_act(x_bf, w_bf, bias.float(), act, 0)
+        out = C.gemm_nt_bias_act(x_bf, w_bf, bias.float(), act, 0, 0)
         ctx.save_for_backward(x_bf, w_bf, out)
         ctx.act = act
         ctx.k = x.shape[1]
@@ -48,9 +48,9 @@ class FusedLinearFn(torch.autograd.Function):
         if ctx.act == 1:
             g = C.relu_bwd(g, out)
         db = C.bias_grad(g)
-        # dX = g @ W = g @ (W^T)^T — NT kernel with the [Kp, N] transpose
+        # dX = g @ W: trans_b path consumes the [N, Kp] weight directly
         dx = C.gemm_nt_bias_act(
-            g, w_bf.t().contiguous(), torch.empty(0, device=g.device), 0, 0
+            g, w_bf, torch.empty(0, device=g.device), 0, 0, 1
         )
         if dx.shape[1] != ctx.k:
             dx = dx[:, : ctx.k]

@@ -22,12 +22,18 @@ def test_gemm_nt_bias_act_matches_torch(shape, act):
     A = (torch.randn(M, K, device=_dev()) * 0.5).to(torch.bfloat16)
     B = (torch.randn(N, K, device=_dev()) * 0.5).to(torch.bfloat16)
     bias = torch.randn(N, device=_dev())
-    out = C.gemm_nt_bias_act(A.contiguous(), B.contiguous(), bias, act, 0)
+    out = C.gemm_nt_bias_act(A.contiguous(), B.contiguous(), bias, act, 0, 0)
     ref = A.float() @ B.float().t() + bias
     if act == 1:
         ref = torch.relu(ref)
     assert torch.allclose(out.float(), ref, atol=0.1 + 0.02 * np.sqrt(K), rtol=0.02), (
         f"max err {(out.float() - ref).abs().max()}"
+    )
+    # trans_b path: same product with B given as [K, N]
+    Bt = B.t().contiguous()
+    out2 = C.gemm_nt_bias_act(A.contiguous(), Bt, bias, act, 0, 1)
+    assert torch.allclose(out2.float(), ref, atol=0.1 + 0.02 * np.sqrt(K), rtol=0.02), (
+        f"trans_b max err {(out2.float() - ref).abs().max()}"
     )
 
 
