@@ -54,8 +54,9 @@ def parse_args():
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--graph", type=int, default=1,
                    help="capture dense fwd+bwd in a hipGraph (1 GPU only)")
-    p.add_argument("--fused-dense", type=int, default=1,
-                   help="use the hand-written MFMA GEMM+bias+ReLU layers")
+    p.add_argument("--fused-dense", type=int, default=0,
+                   help="use the hand-written MFMA GEMM+bias+ReLU layers "
+                        "(currently ~0.95x the graphed hipBLASLt path)")
     args = p.parse_args()
     preset = dict(PRESETS[args.preset])
     preset.setdefault("batch_size", 4096)
@@ -116,7 +117,7 @@ def main():
             capacity=capacity, spill_capacity=int(args.spill_capacity or 0)
         ),
         device=device,
-        dist_ctx=DistContext.from_default_group(),
+        dist_ctx=DistContext.new_sparse_group(),
     )
     if args.model == "dlrm":
         model = DLRM(

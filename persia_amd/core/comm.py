@@ -32,6 +32,22 @@ class DistContext:
             return DistContext(dist.get_world_size(), dist.get_rank(), None)
         return DistContext(1, 0, None)
 
+    @staticmethod
+    def new_sparse_group() -> "DistContext":
+        """Dedicated communicator for the embedding all-to-all.
+
+        The sparse exchange is issued from the lookup pipeline THREAD while
+        DDP's all-reduce runs on the main thread; on one shared NCCL
+        communicator their cross-rank issue order is not guaranteed and can
+        deadlock.  A separate process group gives each side its own
+        independently-ordered communicator (collectives within each are
+        issued in the same order on every rank).  Must be called by ALL
+        ranks."""
+        if dist.is_available() and dist.is_initialized():
+            group = dist.new_group(list(range(dist.get_world_size())))
+            return DistContext(dist.get_world_size(), dist.get_rank(), group)
+        return DistContext(1, 0, None)
+
     @property
     def distributed(self) -> bool:
         return self.world_size > 1

@@ -172,7 +172,9 @@ class EmbeddingCtx(BaseCtx):
                 optimizer=embedding_optimizer,
                 gconf=gconf,
                 device=self.device,
-                dist_ctx=DistContext.from_default_group(),
+                # own communicator: the sparse all-to-all must not interleave
+                # with DDP's all-reduce on one NCCL comm (deadlock hazard)
+                dist_ctx=DistContext.new_sparse_group(),
             )
         _LAST_ENGINE = self.engine
         self.current_batch: Optional[PersiaTrainingBatch] = None

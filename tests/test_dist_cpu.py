@@ -89,7 +89,7 @@ def _worker(rank, port, result_dir):
     dist.init_process_group("gloo", rank=rank, world_size=WORLD)
     from persia_amd.core.comm import DistContext
 
-    eng = _make_engine(DistContext.from_default_group())
+    eng = _make_engine(DistContext.new_sparse_group())
     # both ranks see identical batches/grads (worst-case total key overlap)
     outs = _run_steps(eng, grad_mult=1.0)
     if rank == 0:
