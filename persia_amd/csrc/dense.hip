@@ -81,45 +81,68 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
   // write transposed into lds_b[n][k].
   const int t_r = tid / 8, t_c8 = (tid % 8) * 8;  // 32 k-rows x 64 cols
 
-  auto stage = [&](int buf, int k0) {
+  // T14 async-STAGE split (guide §6 G15): issue the next tile's global
+  // loads into REGISTERS before the MFMA block (HBM latency hides under the
+  // compute), write them to LDS after it — never a synchronous
+  // load->ds_write->barrier chain per K-step.
+  constexpr int NB = TBN / 64;
+  auto stage_load = [&](int k0, bf16x8 (&ra)[2], bf16x8 (&rb)[2]) {
 #pragma unroll
     for (int half = 0; half < 2; ++half) {
       const int r = s_r + half * 64;  // A row
-      bf16x8 v{};
+      ra[half] = bf16x8{};
       if (m0 + r < M)
-        v = *(const bf16x8*)(A + (int64_t)(m0 + r) * K + k0 + s_c8);
-      *(bf16x8*)&lds_a[buf][r * LDK + s_c8] = v;
+        ra[half] = *(const bf16x8*)(A + (int64_t)(m0 + r) * K + k0 + s_c8);
     }
     if (!TRANS_B) {
 #pragma unroll
-      for (int half = 0; half < TBN / 64; ++half) {
+      for (int half = 0; half < NB; ++half) {
         const int r = s_r + half * 64;  // B row (n)
-        bf16x8 w{};
+        rb[half] = bf16x8{};
         if (n0 + r < N)
-          w = *(const bf16x8*)(B + (int64_t)(n0 + r) * K + k0 + s_c8);
-        *(bf16x8*)&lds_b[buf][r * LDK + s_c8] = w;
+          rb[half] = *(const bf16x8*)(B + (int64_t)(n0 + r) * K + k0 + s_c8);
       }
     } else {
-      // rows = k, cols = n; 32 x TBN tile, transposed write
 #pragma unroll
-      for (int half = 0; half < TBN / 64; ++half) {
+      for (int half = 0; half < NB; ++half) {
         const int c = t_c8 + half * 64;
-        bf16x8 w{};
-        if (n0 + c < N && t_c8 < 64)
-          w = *(const bf16x8*)(B + (int64_t)(k0 + t_r) * N + n0 + c);
+        rb[half] = bf16x8{};
+        if (n0 + c < N)
+          rb[half] = *(const bf16x8*)(B + (int64_t)(k0 + t_r) * N + n0 + c);
+      }
+    }
+  };
+  auto stage_write = [&](int buf, bf16x8 (&ra)[2], bf16x8 (&rb)[2]) {
+#pragma unroll
+    for (int half = 0; half < 2; ++half)
+      *(bf16x8*)&lds_a[buf][(s_r + half * 64) * LDK + s_c8] = ra[half];
+    if (!TRANS_B) {
+#pragma unroll
+      for (int half = 0; half < NB; ++half)
+        *(bf16x8*)&lds_b[buf][(s_r + half * 64) * LDK + s_c8] = rb[half];
+    } else {
+#pragma unroll
+      for (int half = 0; half < NB; ++half) {
+        const int c = t_c8 + half * 64;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          lds_b[buf][(c + j) * LDK + t_r] = w[j];
+          lds_b[buf][(c + j) * LDK + t_r] = rb[half][j];
       }
     }
   };
 
   int buf = 0;
-  stage(0, 0);
+  {
+    bf16x8 ra[2], rb[2];
+    stage_load(0, ra, rb);
+    stage_write(0, ra, rb);
+  }
   __syncthreads();
 
   for (int k0 = 0; k0 < K; k0 += BK) {
-    if (k0 + BK < K) stage(buf ^ 1, k0 + BK);
+    bf16x8 ra[2], rb[2];
+    const bool prefetch = k0 + BK < K;
+    if (prefetch) stage_load(k0 + BK, ra, rb);
 #pragma unroll
     for (int am = 0; am < 4; ++am) {
       const bf16x8 a_frag =
@@ -132,6 +155,7 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
             a_frag, b_frag, acc[am][bn], 0, 0, 0);
       }
     }
+    if (prefetch) stage_write(buf ^ 1, ra, rb);
     buf ^= 1;
     __syncthreads();
   }
@@ -172,8 +196,8 @@ constexpr int WLD = WTM + PAD;
 __global__ __launch_bounds__(256) void wgrad_kernel(
     const short* __restrict__ dC, const short* __restrict__ A,
     float* __restrict__ dW, int M, int N, int K, int splitm) {
-  __shared__ short lds_dct[64 * WLD];  // [n][m]
-  __shared__ short lds_at[64 * WLD];   // [k][m]
+  __shared__ short lds_dct[2][64 * WLD];  // [n][m]
+  __shared__ short lds_at[2][64 * WLD];   // [k][m]
 
   const int n_tiles_k = (K + 63) / 64;
   const int tile_id = blockIdx.x / splitm;
@@ -194,36 +218,52 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
   f32x4 acc[2][2] = {};
 
   // staging: thread loads 8 contiguous cols of one m-row, writes transposed
+  // (T14 split: loads issued before the MFMA block, writes after)
   const int s_m = tid / 8, s_c8 = (tid % 8) * 8;  // 32 m-rows x 64 cols
 
+  auto wload = [&](int m0, bf16x8& v, bf16x8& w) {
+    const int m = m0 + s_m;
+    v = bf16x8{};
+    if (m < M && n0 + s_c8 < N)
+      v = *(const bf16x8*)(dC + (int64_t)m * N + n0 + s_c8);
+    w = bf16x8{};
+    if (m < M && k0 + s_c8 < K)
+      w = *(const bf16x8*)(A + (int64_t)m * K + k0 + s_c8);
+  };
+  auto wwrite = [&](int buf, bf16x8& v, bf16x8& w) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) lds_dct[buf][(s_c8 + j) * WLD + s_m] = v[j];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) lds_at[buf][(s_c8 + j) * WLD + s_m] = w[j];
+  };
+
+  int buf = 0;
+  {
+    bf16x8 v, w;
+    wload(m_begin, v, w);
+    wwrite(0, v, w);
+  }
+  __syncthreads();
+
   for (int m0 = m_begin; m0 < m_end; m0 += WTM) {
-    {
-      const int m = m0 + s_m;
-      bf16x8 v{};
-      if (m < M && n0 + s_c8 < N)
-        v = *(const bf16x8*)(dC + (int64_t)m * N + n0 + s_c8);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) lds_dct[(s_c8 + j) * WLD + s_m] = v[j];
-      bf16x8 w{};
-      if (m < M && k0 + s_c8 < K)
-        w = *(const bf16x8*)(A + (int64_t)m * K + k0 + s_c8);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) lds_at[(s_c8 + j) * WLD + s_m] = w[j];
-    }
-    __syncthreads();
+    bf16x8 v, w;
+    const bool prefetch = m0 + WTM < m_end;
+    if (prefetch) wload(m0 + WTM, v, w);
     // one MFMA K-step (WTM == 32)
 #pragma unroll
     for (int an = 0; an < 2; ++an) {
       const bf16x8 a_frag =
-          *(const bf16x8*)&lds_dct[(wr * 32 + an * 16 + fi) * WLD + fk8];
+          *(const bf16x8*)&lds_dct[buf][(wr * 32 + an * 16 + fi) * WLD + fk8];
 #pragma unroll
       for (int bk = 0; bk < 2; ++bk) {
         const bf16x8 b_frag =
-            *(const bf16x8*)&lds_at[(wc * 32 + bk * 16 + fi) * WLD + fk8];
+            *(const bf16x8*)&lds_at[buf][(wc * 32 + bk * 16 + fi) * WLD + fk8];
         acc[an][bk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a_frag, b_frag, acc[an][bk], 0, 0, 0);
       }
     }
+    if (prefetch) wwrite(buf ^ 1, v, w);
+    buf ^= 1;
     __syncthreads();
   }
 

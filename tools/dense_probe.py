@@ -20,7 +20,7 @@ for M,N,K in shapes:
     A = (torch.randn(M,K,device=dev)*0.1).to(torch.bfloat16).contiguous()
     B = (torch.randn(N,K,device=dev)*0.1).to(torch.bfloat16).contiguous()
     bias = torch.randn(N, device=dev)
-    t_ours = bench(lambda: C.gemm_nt_bias_act(A,B,bias,1,0))
+    t_ours = bench(lambda: C.gemm_nt_bias_act(A,B,bias,1,0,0))
     t_ref = bench(lambda: torch.relu(A @ B.t() + bias.to(torch.bfloat16)))
     tf = 2*M*N*K/t_ours/1e6
     print(f"{M:>6}{N:>6}{K:>6} {t_ours:9.1f} {t_ref:10.1f} {tf:9.1f}")
