@@ -48,19 +48,26 @@ __device__ __forceinline__ short f2bf(float f) {
 // Both fragments read 8 contiguous shorts from a [row][k] LDS image.
 // ---------------------------------------------------------------------------
 // TBN: output tile N-width (128, or 64 for skinny layers so the grid fills
-// 256 CUs).  TRANS_B: second operand given as [Kdim, N] row-major (the dgrad
-// case dX = g @ W with W [N,K]: K here is the reduction N of the forward) —
-// staged through a transposed LDS image (scalar writes) so fragment reads
-// stay contiguous.
-template <int ACT, bool STORE_F32, bool TRANS_B, int TBN>
+// 256 CUs).  TBK: K-step (32 or 64 — 64 halves the barrier count).
+// TRANS_B: second operand given as [Kdim, N] row-major (the dgrad case
+// dX = g @ W with W [N,K]) — staged through a transposed LDS image (scalar
+// writes) so fragment reads stay contiguous.
+// T14 async-STAGE split (guide §6 G15): the next tile's global loads are
+// issued into REGISTERS before the MFMA block (HBM latency hides under the
+// compute), LDS writes happen after it.
+template <int ACT, bool STORE_F32, bool TRANS_B, int TBN, int TBK>
 __global__ __launch_bounds__(256) void gemm_nt_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     const float* __restrict__ bias, void* __restrict__ C, int M, int N,
     int K) {
-  __shared__ short lds_a[2][BM * LDK];
-  __shared__ short lds_b[2][TBN * LDK];
+  constexpr int LK = TBK + PAD;
+  __shared__ short lds_a[2][BM * LK];
+  __shared__ short lds_b[2][TBN * LK];
   constexpr int WN = TBN / 2;  // per-wave N extent
   constexpr int NFRAG = WN / 16;
+  constexpr int PA = BM * TBK / (256 * 8);   // A pieces per thread
+  constexpr int PB = TBN * TBK / (256 * 8);  // B pieces per thread
+  constexpr int ACPR = TBK / 8;              // A pieces per row
 
   const int n_tiles_n = (N + TBN - 1) / TBN;
   const int m0 = (blockIdx.x / n_tiles_n) * BM;
@@ -75,84 +82,77 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
 
   f32x4 acc[4][NFRAG] = {};
 
-  // A staging: 256 threads x 8-short pieces; 128 rows x BK/8 pieces.
-  const int s_r = tid / 4, s_c8 = (tid % 4) * 8;  // 64 rows per round
-  // B staging (TRANS_B): B mem rows are the k-reduction axis: B[k0+r][n0+c];
-  // write transposed into lds_b[n][k].
-  const int t_r = tid / 8, t_c8 = (tid % 8) * 8;  // 32 k-rows x 64 cols
-
-  // T14 async-STAGE split (guide §6 G15): issue the next tile's global
-  // loads into REGISTERS before the MFMA block (HBM latency hides under the
-  // compute), write them to LDS after it — never a synchronous
-  // load->ds_write->barrier chain per K-step.
-  constexpr int NB = TBN / 64;
-  auto stage_load = [&](int k0, bf16x8 (&ra)[2], bf16x8 (&rb)[2]) {
+  auto stage_load = [&](int k0, bf16x8 (&ra)[PA], bf16x8 (&rb)[PB]) {
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
-      const int r = s_r + half * 64;  // A row
-      ra[half] = bf16x8{};
+    for (int i = 0; i < PA; ++i) {
+      const int p = tid + i * 256;
+      const int r = p / ACPR, c8 = (p % ACPR) * 8;
+      ra[i] = bf16x8{};
       if (m0 + r < M)
-        ra[half] = *(const bf16x8*)(A + (int64_t)(m0 + r) * K + k0 + s_c8);
+        ra[i] = *(const bf16x8*)(A + (int64_t)(m0 + r) * K + k0 + c8);
     }
-    if (!TRANS_B) {
 #pragma unroll
-      for (int half = 0; half < NB; ++half) {
-        const int r = s_r + half * 64;  // B row (n)
-        rb[half] = bf16x8{};
+    for (int i = 0; i < PB; ++i) {
+      const int p = tid + i * 256;
+      rb[i] = bf16x8{};
+      if (!TRANS_B) {
+        const int r = p / ACPR, c8 = (p % ACPR) * 8;  // image [TBN][TBK]
         if (n0 + r < N)
-          rb[half] = *(const bf16x8*)(B + (int64_t)(n0 + r) * K + k0 + s_c8);
-      }
-    } else {
-#pragma unroll
-      for (int half = 0; half < NB; ++half) {
-        const int c = t_c8 + half * 64;
-        rb[half] = bf16x8{};
-        if (n0 + c < N)
-          rb[half] = *(const bf16x8*)(B + (int64_t)(k0 + t_r) * N + n0 + c);
+          rb[i] = *(const bf16x8*)(B + (int64_t)(n0 + r) * K + k0 + c8);
+      } else {
+        const int r = p / (TBN / 8), c8 = (p % (TBN / 8)) * 8;  // mem [TBK][N]
+        if (n0 + c8 < N)
+          rb[i] = *(const bf16x8*)(B + (int64_t)(k0 + r) * N + n0 + c8);
       }
     }
   };
-  auto stage_write = [&](int buf, bf16x8 (&ra)[2], bf16x8 (&rb)[2]) {
+  auto stage_write = [&](int buf, bf16x8 (&ra)[PA], bf16x8 (&rb)[PB]) {
 #pragma unroll
-    for (int half = 0; half < 2; ++half)
-      *(bf16x8*)&lds_a[buf][(s_r + half * 64) * LDK + s_c8] = ra[half];
-    if (!TRANS_B) {
+    for (int i = 0; i < PA; ++i) {
+      const int p = tid + i * 256;
+      const int r = p / ACPR, c8 = (p % ACPR) * 8;
+      *(bf16x8*)&lds_a[buf][r * LK + c8] = ra[i];
+    }
 #pragma unroll
-      for (int half = 0; half < NB; ++half)
-        *(bf16x8*)&lds_b[buf][(s_r + half * 64) * LDK + s_c8] = rb[half];
-    } else {
-#pragma unroll
-      for (int half = 0; half < NB; ++half) {
-        const int c = t_c8 + half * 64;
+    for (int i = 0; i < PB; ++i) {
+      const int p = tid + i * 256;
+      if (!TRANS_B) {
+        const int r = p / ACPR, c8 = (p % ACPR) * 8;
+        *(bf16x8*)&lds_b[buf][r * LK + c8] = rb[i];
+      } else {
+        const int r = p / (TBN / 8), c8 = (p % (TBN / 8)) * 8;  // k-row, n-col
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          lds_b[buf][(c + j) * LDK + t_r] = rb[half][j];
+          lds_b[buf][(c8 + j) * LK + r] = rb[i][j];
       }
     }
   };
 
   int buf = 0;
   {
-    bf16x8 ra[2], rb[2];
+    bf16x8 ra[PA], rb[PB];
     stage_load(0, ra, rb);
     stage_write(0, ra, rb);
   }
   __syncthreads();
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
-    bf16x8 ra[2], rb[2];
-    const bool prefetch = k0 + BK < K;
-    if (prefetch) stage_load(k0 + BK, ra, rb);
+  for (int k0 = 0; k0 < K; k0 += TBK) {
+    bf16x8 ra[PA], rb[PB];
+    const bool prefetch = k0 + TBK < K;
+    if (prefetch) stage_load(k0 + TBK, ra, rb);
 #pragma unroll
-    for (int am = 0; am < 4; ++am) {
-      const bf16x8 a_frag =
-          *(const bf16x8*)&lds_a[buf][(wr * 64 + am * 16 + fi) * LDK + fk8];
+    for (int kk = 0; kk < TBK; kk += 32) {
 #pragma unroll
-      for (int bn = 0; bn < NFRAG; ++bn) {
-        const bf16x8 b_frag =
-            *(const bf16x8*)&lds_b[buf][(wc * WN + bn * 16 + fi) * LDK + fk8];
-        acc[am][bn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a_frag, b_frag, acc[am][bn], 0, 0, 0);
+      for (int am = 0; am < 4; ++am) {
+        const bf16x8 a_frag = *(const bf16x8*)
+            &lds_a[buf][(wr * 64 + am * 16 + fi) * LK + kk + fk8];
+#pragma unroll
+        for (int bn = 0; bn < NFRAG; ++bn) {
+          const bf16x8 b_frag = *(const bf16x8*)
+              &lds_b[buf][(wc * WN + bn * 16 + fi) * LK + kk + fk8];
+          acc[am][bn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag, b_frag, acc[am][bn], 0, 0, 0);
+        }
       }
     }
     if (prefetch) stage_write(buf ^ 1, ra, rb);
@@ -343,26 +343,35 @@ torch::Tensor gemm_nt_bias_act(torch::Tensor A, torch::Tensor B,
           .dtype(out_f32 ? torch::kFloat32 : torch::kBFloat16)
           .device(A.device()));
   const int m_tiles = (M + BM - 1) / BM;
-  const bool narrow = (int64_t)m_tiles * ((N + 127) / 128) < 192 && N % 64 == 0;
+  // tile config: BK=64 halves barriers (needs K%64); BN=64 doubles the grid
+  // (skinny layers underfill 256 CUs at BN=128); (128,64) exceeds 64KB LDS.
+  const bool k64 = K % 64 == 0 && N % 64 == 0;
+  const bool narrow =
+      ((int64_t)m_tiles * ((N + 127) / 128) < 192 && N % 64 == 0) || k64;
   const int tbn = narrow ? 64 : 128;
   const int grid = m_tiles * ((N + tbn - 1) / tbn);
   const float* bias_ptr = bias.numel() ? bias.data_ptr<float>() : nullptr;
   hipStream_t st = dcur_stream();
-#define PA_GEMM(ACTV, F32V, TBV, TBNV)                                        \
-  hipLaunchKernelGGL((gemm_nt_kernel<ACTV, F32V, TBV, TBNV>), dim3(grid),     \
-                     dim3(256), 0, st, (const bf16*)A.data_ptr(),             \
+#define PA_GEMM(ACTV, F32V, TBV, TBNV, TBKV)                                  \
+  hipLaunchKernelGGL((gemm_nt_kernel<ACTV, F32V, TBV, TBNV, TBKV>),           \
+                     dim3(grid), dim3(256), 0, st, (const bf16*)A.data_ptr(), \
                      (const bf16*)B.data_ptr(), bias_ptr, C.data_ptr(), M, N, \
                      K)
+#define PA_GEMM_CFG(ACTV, F32V, TBV)                                          \
+  do {                                                                        \
+    if (k64) PA_GEMM(ACTV, F32V, TBV, 64, 64);                                \
+    else if (narrow) PA_GEMM(ACTV, F32V, TBV, 64, 32);                        \
+    else PA_GEMM(ACTV, F32V, TBV, 128, 32);                                   \
+  } while (0)
 #define PA_GEMM_T(ACTV, F32V)                                                 \
   do {                                                                        \
-    if (trans_b) { if (narrow) PA_GEMM(ACTV, F32V, true, 64);                 \
-                   else PA_GEMM(ACTV, F32V, true, 128); }                     \
-    else         { if (narrow) PA_GEMM(ACTV, F32V, false, 64);                \
-                   else PA_GEMM(ACTV, F32V, false, 128); }                    \
+    if (trans_b) PA_GEMM_CFG(ACTV, F32V, true);                               \
+    else PA_GEMM_CFG(ACTV, F32V, false);                                      \
   } while (0)
   if (act == 1) { if (out_f32) PA_GEMM_T(1, true); else PA_GEMM_T(1, false); }
   else          { if (out_f32) PA_GEMM_T(0, true); else PA_GEMM_T(0, false); }
 #undef PA_GEMM_T
+#undef PA_GEMM_CFG
 #undef PA_GEMM
   return C;
 }
