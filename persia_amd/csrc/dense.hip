@@ -190,25 +190,24 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
 // 64x64 (n,k) tiles, 4 waves as 2x2 of 32x32; M split across blocks with
 // atomicAdd (small-K layers underfill the chip otherwise).
 // ---------------------------------------------------------------------------
-constexpr int WTM = 64;       // m chunk per stage (2 MFMA k-steps)
+constexpr int WTM = 32;       // m chunk per step
 constexpr int WLD = WTM + PAD;
-constexpr int WTN = 128, WTK = 64;  // output tile (n x k)
 
 __global__ __launch_bounds__(256) void wgrad_kernel(
     const short* __restrict__ dC, const short* __restrict__ A,
     float* __restrict__ dW, int M, int N, int K, int splitm) {
-  __shared__ short lds_dct[2][WTN * WLD];  // [n][m]
-  __shared__ short lds_at[2][WTK * WLD];   // [k][m]
+  __shared__ short lds_dct[2][64 * WLD];  // [n][m]
+  __shared__ short lds_at[2][64 * WLD];   // [k][m]
 
-  const int n_tiles_k = (K + WTK - 1) / WTK;
+  const int n_tiles_k = (K + 63) / 64;
   const int tile_id = blockIdx.x / splitm;
   const int m_part = blockIdx.x % splitm;
-  const int n0 = (tile_id / n_tiles_k) * WTN;
-  const int k0 = (tile_id % n_tiles_k) * WTK;
+  const int n0 = (tile_id / n_tiles_k) * 64;
+  const int k0 = (tile_id % n_tiles_k) * 64;
 
   const int tid = threadIdx.x;
   const int wave = tid / 64, lane = tid % 64;
-  const int wr = wave / 2, wc = wave % 2;  // wave tile: 64(n) x 32(k)
+  const int wr = wave / 2, wc = wave % 2;
   const int fi = lane & 15;
   const int fk8 = (lane >> 4) * 8;
 
@@ -216,71 +215,51 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
   const int m_begin = m_part * m_chunk;
   const int m_end = min(M, m_begin + m_chunk);
 
-  f32x4 acc[4][2] = {};
+  f32x4 acc[2][2] = {};
 
-  // staging (T14 split): vector loads from [m][col] memory, transposed
-  // scalar writes into [col][m] LDS images.
-  // dC tile: WTM x WTN = 64x128 -> 4 pieces/thread; A tile: 64x64 -> 2.
-  auto wload = [&](int m0, bf16x8 (&v)[4], bf16x8 (&w)[2]) {
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      const int p = tid + i * 256;
-      const int m = m0 + p / (WTN / 8), c8 = (p % (WTN / 8)) * 8;
-      v[i] = bf16x8{};
-      if (m < M && n0 + c8 < N)
-        v[i] = *(const bf16x8*)(dC + (int64_t)m * N + n0 + c8);
-    }
-#pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      const int p = tid + i * 256;
-      const int m = m0 + p / (WTK / 8), c8 = (p % (WTK / 8)) * 8;
-      w[i] = bf16x8{};
-      if (m < M && k0 + c8 < K)
-        w[i] = *(const bf16x8*)(A + (int64_t)m * K + k0 + c8);
-    }
+  // staging: thread loads 8 contiguous cols of one m-row, writes transposed
+  // (T14 split: loads issued before the MFMA block, writes after)
+  const int s_m = tid / 8, s_c8 = (tid % 8) * 8;  // 32 m-rows x 64 cols
+
+  auto wload = [&](int m0, bf16x8& v, bf16x8& w) {
+    const int m = m0 + s_m;
+    v = bf16x8{};
+    if (m < M && n0 + s_c8 < N)
+      v = *(const bf16x8*)(dC + (int64_t)m * N + n0 + s_c8);
+    w = bf16x8{};
+    if (m < M && k0 + s_c8 < K)
+      w = *(const bf16x8*)(A + (int64_t)m * K + k0 + s_c8);
   };
-  auto wwrite = [&](int buf, bf16x8 (&v)[4], bf16x8 (&w)[2]) {
+  auto wwrite = [&](int buf, bf16x8& v, bf16x8& w) {
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      const int p = tid + i * 256;
-      const int m = p / (WTN / 8), c8 = (p % (WTN / 8)) * 8;
+    for (int j = 0; j < 8; ++j) lds_dct[buf][(s_c8 + j) * WLD + s_m] = v[j];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) lds_dct[buf][(c8 + j) * WLD + m] = v[i][j];
-    }
-#pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      const int p = tid + i * 256;
-      const int m = p / (WTK / 8), c8 = (p % (WTK / 8)) * 8;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) lds_at[buf][(c8 + j) * WLD + m] = w[i][j];
-    }
+    for (int j = 0; j < 8; ++j) lds_at[buf][(s_c8 + j) * WLD + s_m] = w[j];
   };
 
   int buf = 0;
   {
-    bf16x8 v[4], w[2];
+    bf16x8 v, w;
     wload(m_begin, v, w);
     wwrite(0, v, w);
   }
   __syncthreads();
 
   for (int m0 = m_begin; m0 < m_end; m0 += WTM) {
-    bf16x8 v[4], w[2];
+    bf16x8 v, w;
     const bool prefetch = m0 + WTM < m_end;
     if (prefetch) wload(m0 + WTM, v, w);
+    // one MFMA K-step (WTM == 32)
 #pragma unroll
-    for (int ms = 0; ms < WTM; ms += 32) {
+    for (int an = 0; an < 2; ++an) {
+      const bf16x8 a_frag =
+          *(const bf16x8*)&lds_dct[buf][(wr * 32 + an * 16 + fi) * WLD + fk8];
 #pragma unroll
-      for (int an = 0; an < 4; ++an) {
-        const bf16x8 a_frag = *(const bf16x8*)
-            &lds_dct[buf][(wr * 64 + an * 16 + fi) * WLD + ms + fk8];
-#pragma unroll
-        for (int bk = 0; bk < 2; ++bk) {
-          const bf16x8 b_frag = *(const bf16x8*)
-              &lds_at[buf][(wc * 32 + bk * 16 + fi) * WLD + ms + fk8];
-          acc[an][bk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag, b_frag, acc[an][bk], 0, 0, 0);
-        }
+      for (int bk = 0; bk < 2; ++bk) {
+        const bf16x8 b_frag =
+            *(const bf16x8*)&lds_at[buf][(wc * 32 + bk * 16 + fi) * WLD + fk8];
+        acc[an][bk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag, b_frag, acc[an][bk], 0, 0, 0);
       }
     }
     if (prefetch) wwrite(buf ^ 1, v, w);
@@ -289,14 +268,14 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
   }
 
 #pragma unroll
-  for (int an = 0; an < 4; ++an) {
+  for (int an = 0; an < 2; ++an) {
 #pragma unroll
     for (int bk = 0; bk < 2; ++bk) {
       const int k = k0 + wc * 32 + bk * 16 + fi;
       if (k >= K) continue;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int n = n0 + wr * 64 + an * 16 + (lane >> 4) * 4 + r;
+        const int n = n0 + wr * 32 + an * 16 + (lane >> 4) * 4 + r;
         if (n >= N) continue;
         if (splitm > 1)
           atomicAdd(&dW[(int64_t)n * K + k], acc[an][bk][r]);
@@ -426,9 +405,9 @@ torch::Tensor wgrad(torch::Tensor dC, torch::Tensor A) {
   const int M = (int)dC.size(0), N = (int)dC.size(1), K = (int)A.size(1);
   auto dW = torch::empty(
       {N, K}, torch::TensorOptions().dtype(torch::kFloat32).device(A.device()));
-  const int tiles = ((N + WTN - 1) / WTN) * ((K + WTK - 1) / WTK);
+  const int tiles = ((N + 63) / 64) * ((K + 63) / 64);
   int splitm = 1;
-  while (tiles * splitm < 512 && splitm < 64 && (M / (splitm * 2)) >= WTM)
+  while (tiles * splitm < 512 && splitm < 64 && (M / (splitm * 2)) >= 32)
     splitm *= 2;
   if (splitm > 1) dW.zero_();
   hipLaunchKernelGGL(wgrad_kernel, dim3(tiles * splitm), dim3(256), 0,
