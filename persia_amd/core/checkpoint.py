@@ -85,7 +85,7 @@ def dump_embedding(engine: "EmbeddingEngine", dst_dir: str) -> None:
             },
             f,
         )
-    engine.dist.barrier()
+    engine.dist_grad.barrier()
     if rank == 0:
         with open(os.path.join(dst_dir, DONE_MARKER), "w", encoding="utf-8") as f:
             yaml.safe_dump(
@@ -96,7 +96,7 @@ def dump_embedding(engine: "EmbeddingEngine", dst_dir: str) -> None:
                 },
                 f,
             )
-    engine.dist.barrier()
+    engine.dist_grad.barrier()
 
 
 def load_embedding(engine: "EmbeddingEngine", src_dir: str) -> None:
@@ -132,4 +132,4 @@ def load_embedding(engine: "EmbeddingEngine", src_dir: str) -> None:
             for fn in sorted(os.listdir(shard_dir)):
                 if fn.endswith(".emb"):
                     import_file(os.path.join(shard_dir, fn), filter_owner=True)
-    engine.dist.barrier()
+    engine.dist_grad.barrier()

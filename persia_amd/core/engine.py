@@ -254,6 +254,13 @@ class EmbeddingEngine:
         self.gconf = gconf
         self.device = device
         self.dist = dist_ctx or DistContext.from_default_group()
+        # Lookups are issued from the pipeline THREAD and gradient pushes from
+        # the main thread; each needs its OWN communicator so cross-thread
+        # interleaving can't reorder collectives across ranks (NCCL requires
+        # identical issue order per communicator on every rank).
+        self.dist_grad = (
+            DistContext.new_sparse_group() if self.dist.distributed else self.dist
+        )
         self.wire_dtype = wire_dtype
         # group slots by dim (slots of one dim share one store + one a2a)
         self.groups: Dict[int, List[SlotConfig]] = {}
@@ -652,13 +659,14 @@ class EmbeddingEngine:
         if not self.dist.distributed:
             self.skipped_grad_signs += store.update_gradients(group.uniq_keys, buf)
             return
+        comm = self.dist_grad  # main-thread communicator (see __init__)
         send_counts, recv_counts = group.send_counts, group.recv_counts
         if send_counts is None:
-            owner = _owner_of_keys(group.uniq_keys, self.dist.world_size)
-            send_counts = torch.bincount(owner, minlength=self.dist.world_size).tolist()
-            recv_counts = self.dist.all_to_all_lengths(send_counts)
-        keys_recv = self.dist.all_to_all(group.uniq_keys, send_counts, recv_counts)
-        grads_recv = self.dist.all_to_all(
+            owner = _owner_of_keys(group.uniq_keys, comm.world_size)
+            send_counts = torch.bincount(owner, minlength=comm.world_size).tolist()
+            recv_counts = comm.all_to_all_lengths(send_counts)
+        keys_recv = comm.all_to_all(group.uniq_keys, send_counts, recv_counts)
+        grads_recv = comm.all_to_all(
             buf.to(self.wire_dtype), send_counts, recv_counts
         )
         # merge duplicate keys across source ranks: pre-aggregate into one

@@ -292,7 +292,7 @@ class EmbeddingCtx(BaseCtx):
         """Resident rows per dim-group shard, summed over ranks."""
         sizes = [len(s) for _d, s in sorted(self.engine.stores.items())]
         if self.engine.dist.distributed:
-            sizes = [int(self.engine.dist.allreduce_scalar(float(s))) for s in sizes]
+            sizes = [int(self.engine.dist_grad.allreduce_scalar(float(s))) for s in sizes]
         return sizes
 
     def clear_embeddings(self) -> None:
