@@ -282,6 +282,8 @@ class EmbeddingEngine:
         self.nan_grad_batches = 0
         self._plans = {}
         self._empty_scale = None
+        self.model_manager_status = "Idle"
+        self.model_manager_progress = 0.0
         from persia_amd.core.metrics import EngineMetrics
 
         self.metrics_enabled = bool(gconf.enable_metrics)
@@ -768,15 +770,72 @@ class EmbeddingEngine:
 
     # ------------------------------------------------------------ checkpoint
 
+    # checkpoint status machine (reference model-manager lib.rs:63-69:
+    # {Dumping(progress) | Loading(progress) | Idle | Failed})
+    def _set_status(self, status: str, progress: float = 0.0):
+        self.model_manager_status = status
+        self.model_manager_progress = progress
+
     def dump(self, dst_dir: str, blocking: bool = True) -> None:
         from persia_amd.core.checkpoint import dump_embedding
 
-        dump_embedding(self, dst_dir)
+        def run():
+            try:
+                self._set_status("Dumping", 0.0)
+                dump_embedding(self, dst_dir)
+                self._set_status("Idle", 100.0)
+            except Exception as e:
+                _logger.error(f"embedding dump failed: {e}")
+                self._set_status("Failed")
+                if blocking:
+                    raise
+
+        if blocking:
+            run()
+        else:
+            threading.Thread(target=run, daemon=True, name="persia-ckpt-dump").start()
 
     def load(self, src_dir: str, blocking: bool = True) -> None:
         from persia_amd.core.checkpoint import load_embedding
 
-        load_embedding(self, src_dir)
+        def run():
+            try:
+                self._set_status("Loading", 0.0)
+                load_embedding(self, src_dir)
+                self._set_status("Idle", 100.0)
+            except Exception as e:
+                _logger.error(f"embedding load failed: {e}")
+                self._set_status("Failed")
+                if blocking:
+                    raise
+
+        if blocking:
+            run()
+        else:
+            threading.Thread(target=run, daemon=True, name="persia-ckpt-load").start()
+
+    def wait_for_emb_dumping(self, timeout: float = 600.0) -> None:
+        """Poll until the dump completes (reference rpc.rs:211-241)."""
+        import time as _time
+
+        t0 = _time.time()
+        while getattr(self, "model_manager_status", "Idle") == "Dumping":
+            if _time.time() - t0 > timeout:
+                raise TimeoutError("embedding dump did not finish")
+            _time.sleep(0.05)
+        if getattr(self, "model_manager_status", "Idle") == "Failed":
+            raise RuntimeError("embedding dump failed")
+
+    def wait_for_emb_loading(self, timeout: float = 600.0) -> None:
+        import time as _time
+
+        t0 = _time.time()
+        while getattr(self, "model_manager_status", "Idle") == "Loading":
+            if _time.time() - t0 > timeout:
+                raise TimeoutError("embedding load did not finish")
+            _time.sleep(0.05)
+        if getattr(self, "model_manager_status", "Idle") == "Failed":
+            raise RuntimeError("embedding load failed")
 
     def num_resident_rows(self) -> int:
         return sum(len(s) for s in self.stores.values())

@@ -48,6 +48,21 @@ def test_emb_file_roundtrip(tmp_path):
     assert d == 8
 
 
+def test_async_dump_status_machine(tmp_path):
+    eng = _engine()
+    eng.process_batch(_batch())
+    eng.dump(str(tmp_path / "a"), blocking=False)
+    eng.wait_for_emb_dumping()
+    assert eng.model_manager_status == "Idle"
+    import os
+
+    assert os.path.exists(str(tmp_path / "a" / DONE_MARKER))
+    eng2 = _engine()
+    eng2.load(str(tmp_path / "a"), blocking=False)
+    eng2.wait_for_emb_loading()
+    assert eng2.num_resident_rows() == eng.num_resident_rows()
+
+
 def test_dump_load_roundtrip(tmp_path):
     eng = _engine()
     tb = eng.process_batch(_batch())
