@@ -220,8 +220,6 @@ def main():
                   file=_sys.stderr, flush=True)
             graph = None
 
-    update_stream = torch.cuda.Stream() if use_gpu else None
-
     def train_step(tb):
         if graph is not None:
             with torch.no_grad():
@@ -229,19 +227,7 @@ def main():
                 static["base"].copy_(tb._groups[0].sum_base, non_blocking=True)
                 static["label"].copy_(tb.label_tensors[0], non_blocking=True)
             graph.replay()
-            # sparse update on a side stream: clone the grad (the only tensor
-            # the next replay mutates), then the scatter + optimizer kernels
-            # overlap the next iteration's dense replay (async contract)
-            ev_replay = torch.cuda.Event()
-            ev_replay.record()
-            update_stream.wait_event(ev_replay)
-            with torch.cuda.stream(update_stream):
-                g = static["base"].grad.clone()
-                ev_cloned = torch.cuda.Event()
-                ev_cloned.record(update_stream)
-                tb.record_stream(update_stream)
-                engine.apply_gradients_base(tb, sum_base_grads=[g])
-            torch.cuda.current_stream().wait_event(ev_cloned)
+            engine.apply_gradients_base(tb, sum_base_grads=[static["base"].grad])
             pipeline.release_permit()
             return static["loss"]
         with amp_ctx:
