@@ -12,8 +12,8 @@ def bench(fn, n=50):
     torch.cuda.synchronize()
     return (time.perf_counter() - t0) / n * 1e6  # us
 
-shapes = [(4096,512,13+19),(4096,256,512),(4096,128,256),(4096,1024,480),
-          (4096,1024,1024),(4096,512,1024),(4096,256,512)]
+shapes = [(8192,512,13+19),(8192,256,512),(8192,128,256),(8192,1024,480),
+          (8192,1024,1024),(8192,512,1024),(8192,256,512)]
 print(f"{'M':>6}{'N':>6}{'K':>6} {'ours_us':>9} {'blaslt_us':>10} {'ours_TF':>9}")
 for M,N,K in shapes:
     K = (K+31)//32*32
@@ -26,7 +26,7 @@ for M,N,K in shapes:
     print(f"{M:>6}{N:>6}{K:>6} {t_ours:9.1f} {t_ref:10.1f} {tf:9.1f}")
 
 # wgrad
-for M,N,K in [(4096,512,32),(4096,1024,1024),(4096,256,512)]:
+for M,N,K in [(8192,512,32),(8192,1024,1024),(8192,256,512)]:
     dC = (torch.randn(M,N,device=dev)*0.1).to(torch.bfloat16).contiguous()
     A = (torch.randn(M,K,device=dev)*0.1).to(torch.bfloat16).contiguous()
     t_ours = bench(lambda: C.wgrad(dC,A))
