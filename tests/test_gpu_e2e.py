@@ -78,6 +78,38 @@ def test_dlrm_train_loop_learns():
     assert engine.num_resident_rows() > 0
 
 
+def test_adult_income_gpu_deterministic():
+    """The reference's GPU correctness anchor (train.py:24 GPU_TEST_AUC):
+    deterministic mode reproduces bitwise on the GPU engine too."""
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+    def run_once():
+        env = dict(os.environ)
+        env.update(REPRODUCIBLE="1", EMBEDDING_STALENESS="1")
+        code = (
+            "import sys; sys.path.insert(0, r'%s'); sys.path.insert(0, r'%s');"
+            "import train; a, b = train.main(epochs=1); print('AUCS', repr(a), repr(b))"
+            % (os.path.join(repo, "examples", "adult_income"), repo)
+        )
+        out = subprocess.run(
+            [sys.executable, "-c", code],
+            cwd=os.path.join(repo, "examples", "adult_income"),
+            env=env, capture_output=True, text=True, timeout=900,
+        )
+        assert out.returncode == 0, out.stderr[-2000:]
+        return [l for l in out.stdout.splitlines() if l.startswith("AUCS")][0]
+
+    a1 = run_once()
+    a2 = run_once()
+    assert a1 == a2, f"GPU deterministic mode must reproduce: {a1} != {a2}"
+    train_auc = float(a1.split()[1].split("(")[-1].rstrip(")"))
+    assert train_auc > 0.80, a1
+
+
 def test_bench_default_config_one_step():
     """bench.py's engine wiring at the flagship shape (tiny step count)."""
     import subprocess
