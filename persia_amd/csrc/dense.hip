@@ -190,7 +190,7 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
 // 64x64 (n,k) tiles, 4 waves as 2x2 of 32x32; M split across blocks with
 // atomicAdd (small-K layers underfill the chip otherwise).
 // ---------------------------------------------------------------------------
-constexpr int WTM = 32;       // m chunk per step
+constexpr int WTM = 64;       // m chunk per stage (2 MFMA k-steps per barrier)
 constexpr int WLD = WTM + PAD;
 
 __global__ __launch_bounds__(256) void wgrad_kernel(
@@ -198,6 +198,7 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
     float* __restrict__ dW, int M, int N, int K, int splitm) {
   __shared__ short lds_dct[2][64 * WLD];  // [n][m]
   __shared__ short lds_at[2][64 * WLD];   // [k][m]
+  // staging pieces: [WTM rows x 64 cols] / (256 threads x 8 shorts) = 2 each
 
   const int n_tiles_k = (K + 63) / 64;
   const int tile_id = blockIdx.x / splitm;
@@ -219,47 +220,56 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
 
   // staging: thread loads 8 contiguous cols of one m-row, writes transposed
   // (T14 split: loads issued before the MFMA block, writes after)
-  const int s_m = tid / 8, s_c8 = (tid % 8) * 8;  // 32 m-rows x 64 cols
+  const int s_m = tid / 8, s_c8 = (tid % 8) * 8;  // 2x 32 m-rows x 64 cols
 
-  auto wload = [&](int m0, bf16x8& v, bf16x8& w) {
-    const int m = m0 + s_m;
-    v = bf16x8{};
-    if (m < M && n0 + s_c8 < N)
-      v = *(const bf16x8*)(dC + (int64_t)m * N + n0 + s_c8);
-    w = bf16x8{};
-    if (m < M && k0 + s_c8 < K)
-      w = *(const bf16x8*)(A + (int64_t)m * K + k0 + s_c8);
+  auto wload = [&](int m0, bf16x8 (&v)[2], bf16x8 (&w)[2]) {
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int m = m0 + s_m + h * 32;
+      v[h] = bf16x8{};
+      if (m < M && n0 + s_c8 < N)
+        v[h] = *(const bf16x8*)(dC + (int64_t)m * N + n0 + s_c8);
+      w[h] = bf16x8{};
+      if (m < M && k0 + s_c8 < K)
+        w[h] = *(const bf16x8*)(A + (int64_t)m * K + k0 + s_c8);
+    }
   };
-  auto wwrite = [&](int buf, bf16x8& v, bf16x8& w) {
+  auto wwrite = [&](int buf, bf16x8 (&v)[2], bf16x8 (&w)[2]) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) lds_dct[buf][(s_c8 + j) * WLD + s_m] = v[j];
+    for (int h = 0; h < 2; ++h) {
+      const int m = s_m + h * 32;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) lds_at[buf][(s_c8 + j) * WLD + s_m] = w[j];
+      for (int j = 0; j < 8; ++j) lds_dct[buf][(s_c8 + j) * WLD + m] = v[h][j];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) lds_at[buf][(s_c8 + j) * WLD + m] = w[h][j];
+    }
   };
 
   int buf = 0;
   {
-    bf16x8 v, w;
+    bf16x8 v[2], w[2];
     wload(m_begin, v, w);
     wwrite(0, v, w);
   }
   __syncthreads();
 
   for (int m0 = m_begin; m0 < m_end; m0 += WTM) {
-    bf16x8 v, w;
+    bf16x8 v[2], w[2];
     const bool prefetch = m0 + WTM < m_end;
     if (prefetch) wload(m0 + WTM, v, w);
-    // one MFMA K-step (WTM == 32)
 #pragma unroll
-    for (int an = 0; an < 2; ++an) {
-      const bf16x8 a_frag =
-          *(const bf16x8*)&lds_dct[buf][(wr * 32 + an * 16 + fi) * WLD + fk8];
+    for (int ms = 0; ms < WTM; ms += 32) {
 #pragma unroll
-      for (int bk = 0; bk < 2; ++bk) {
-        const bf16x8 b_frag =
-            *(const bf16x8*)&lds_at[buf][(wc * 32 + bk * 16 + fi) * WLD + fk8];
-        acc[an][bk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a_frag, b_frag, acc[an][bk], 0, 0, 0);
+      for (int an = 0; an < 2; ++an) {
+        const bf16x8 a_frag = *(const bf16x8*)
+            &lds_dct[buf][(wr * 32 + an * 16 + fi) * WLD + ms + fk8];
+#pragma unroll
+        for (int bk = 0; bk < 2; ++bk) {
+          const bf16x8 b_frag = *(const bf16x8*)
+              &lds_at[buf][(wc * 32 + bk * 16 + fi) * WLD + ms + fk8];
+          acc[an][bk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag, b_frag, acc[an][bk], 0, 0, 0);
+        }
       }
     }
     if (prefetch) wwrite(buf ^ 1, v, w);
