@@ -847,9 +847,16 @@ class ForwardPipeline:
     Semaphore → output channel; the permit is released when the batch's
     gradients have been pushed — backward.rs:341-343)."""
 
-    def __init__(self, engine: EmbeddingEngine, staleness: int = 8, out_buffer: int = 8):
+    def __init__(self, engine: EmbeddingEngine, staleness: int = 8, out_buffer: int = 8,
+                 reorder: bool = False):
         self.engine = engine
         self.staleness = max(1, staleness)
+        # reproducible mode with several data loaders: process batches in
+        # batch_id order via a min-heap (reference PerisaDataOrderManager,
+        # forward.rs:396-468)
+        self.reorder = reorder
+        self._heap: list = []
+        self._next_id = 0
         self._sem = threading.Semaphore(self.staleness)
         self._in: "queue.Queue" = queue.Queue(maxsize=max(2, out_buffer))
         self._out: "queue.Queue" = queue.Queue(maxsize=max(2, out_buffer))
@@ -869,14 +876,34 @@ class ForwardPipeline:
         stream = None
         if self.engine.device.type == "cuda":
             stream = torch.cuda.Stream(device=self.engine.device)
+        import heapq
+
+        pending_eof = False
         while not self._stop.is_set():
-            try:
-                item = self._in.get(timeout=0.1)
-            except queue.Empty:
-                continue
-            if item is None:
-                self._out.put(None)
-                break
+            item = None
+            if self.reorder and self._heap and self._heap[0][0] == self._next_id:
+                item = heapq.heappop(self._heap)[1]
+                self._next_id += 1
+            else:
+                try:
+                    got = self._in.get(timeout=0.1)
+                except queue.Empty:
+                    if pending_eof and not self._heap:
+                        self._out.put(None)
+                        break
+                    continue
+                if got is None:
+                    if self.reorder and self._heap:
+                        pending_eof = True  # drain the heap first
+                        continue
+                    self._out.put(None)
+                    break
+                if self.reorder and got.batch_id is not None:
+                    if got.batch_id != self._next_id:
+                        heapq.heappush(self._heap, (got.batch_id, got))
+                        continue
+                    self._next_id += 1
+                item = got
             self._sem.acquire()
             self.inflight += 1
             if self.engine.metrics_enabled:

@@ -89,7 +89,8 @@ class DataLoader:
             )
             staleness = 1 if self.reproducible else self.embedding_staleness
             self._pipeline = ForwardPipeline(
-                ctx.engine, staleness=staleness, out_buffer=self.forward_buffer_size
+                ctx.engine, staleness=staleness, out_buffer=self.forward_buffer_size,
+                reorder=self.reproducible,
             )
             if hasattr(ctx, "_pipeline"):
                 ctx._pipeline = self._pipeline
