@@ -395,11 +395,7 @@ class EmbeddingEngine:
         values_np = np.concatenate([f.values for f in feats])
         vals_t = torch.from_numpy(values_np.view(np.int64)).to(dev)
         spacing = self.schema.feature_spacing
-        keys_t = C.sign_prep(
-            vals_t, plan.slot_starts, plan.prefixes,
-            spacing if spacing < (1 << 63) else -1,
-        )
-        uniq_keys, inverse, perm, ustarts = _dedup(keys_t)
+        spacing_arg = spacing if spacing < (1 << 63) else -1
         slot_ctxs = [
             _SlotCtx(
                 name=f.name, cfg=self.schema.get_slot(f.name),
@@ -408,14 +404,38 @@ class EmbeddingEngine:
             )
             for i, f in enumerate(feats)
         ]
-        group = _GroupCtx(
-            dim=dim, uniq_keys=uniq_keys, inverse=inverse, perm=perm,
-            ustarts=ustarts, slots=slot_ctxs, cat_offsets=plan.cat_offsets,
-            seg_id=plan.seg_id, n_sum_slots=plan.S,
-        )
-        rows = self._exchange_rows(group, train)
-        sums = C.segment_sum(rows.contiguous(), inverse, plan.cat_offsets, plan.empty_scale)
-        group.sum_base = sums
+        store = self.stores[dim]
+        if not self.dist.distributed and store.spill is None:
+            # whole lookup in ONE native call (C++ drives sign prep, dedup,
+            # probe/insert, gather and the fused segment-sum)
+            lo, hi = self.hyper.emb_initialization
+            sums, uniq_keys, inverse, perm, ustarts = C.lookup_local(
+                vals_t, plan.slot_starts, plan.prefixes, spacing_arg,
+                plan.cat_offsets, plan.empty_scale,
+                store.keys, store.ticks, store.arena, dim,
+                1 if train else 0, store.next_tick(), float(lo), float(hi),
+                float(self.hyper.admit_probability),
+                float(self.optimizer.state_init(dim)), store.opt_space,
+            )
+            group = _GroupCtx(
+                dim=dim, uniq_keys=uniq_keys, inverse=inverse, perm=perm,
+                ustarts=ustarts, slots=slot_ctxs, cat_offsets=plan.cat_offsets,
+                seg_id=plan.seg_id, n_sum_slots=plan.S,
+            )
+            group.sum_base = sums
+        else:
+            keys_t = C.sign_prep(vals_t, plan.slot_starts, plan.prefixes, spacing_arg)
+            uniq_keys, inverse, perm, ustarts = _dedup(keys_t)
+            group = _GroupCtx(
+                dim=dim, uniq_keys=uniq_keys, inverse=inverse, perm=perm,
+                ustarts=ustarts, slots=slot_ctxs, cat_offsets=plan.cat_offsets,
+                seg_id=plan.seg_id, n_sum_slots=plan.S,
+            )
+            rows = self._exchange_rows(group, train)
+            sums = C.segment_sum(
+                rows.contiguous(), inverse, plan.cat_offsets, plan.empty_scale
+            )
+            group.sum_base = sums
         for i, sc in enumerate(slot_ctxs):
             out.payloads.append(
                 SlotPayload(name=sc.name, cfg=sc.cfg, sum_tensor=sums[i * B : (i + 1) * B])
@@ -742,6 +762,24 @@ class EmbeddingEngine:
                             group.seg_lens.clamp(min=1.0).rsqrt(),
                             torch.ones_like(group.seg_lens),
                         )
+                store = self.stores[group.dim]
+                if (
+                    not self.dist.distributed
+                    and not raw_grads
+                    and hasattr(store, "_opt_code")  # HipEmbeddingStore
+                ):
+                    # fused native backward: scatter + optimizer in one call
+                    powers = store._adam_step_powers()
+                    b1p, b2p = powers if powers else (0.0, 0.0)
+                    C.update_local(
+                        gbase.contiguous(), group.perm, group.ustarts,
+                        group.seg_id, seg_scale, group.uniq_keys,
+                        store.keys, store.ticks, store.arena, group.dim,
+                        store._opt_code, store._opt_params(), float(b1p),
+                        float(b2p), float(self.hyper.weight_bound),
+                        store._skipped,
+                    )
+                    continue
                 C.grad_scatter(
                     gbase.contiguous(), group.perm, group.ustarts, group.seg_id,
                     seg_scale, buf, 0,

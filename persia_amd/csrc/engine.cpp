@@ -1,0 +1,110 @@
+// Native lookup/update drivers: the per-batch hot path of the embedding
+// engine as single C++ calls (the reference's forward/backward engines are
+// native Rust; here the orchestration of sign-prep -> dedup -> probe ->
+// init/gather -> fused segment-sum (and scatter -> fused optimizer on the
+// way back) is C++ driving the HIP kernels + rocPRIM sort via ATen, so the
+// pipeline thread does one extension call instead of ~25 Python dispatches).
+//
+// The distributed path keeps its all_to_all hops in Python (they go through
+// torch.distributed process groups); it reuses the pieces exposed here.
+#include <torch/extension.h>
+
+// kernels.hip entry points
+void store_lookup(torch::Tensor table_keys, torch::Tensor ticks,
+                  torch::Tensor arena, torch::Tensor query, torch::Tensor out,
+                  int64_t dim, int64_t train, int64_t tick, double lo,
+                  double hi, double admit_prob, double state_init,
+                  int64_t opt_space, torch::Tensor evict_keys,
+                  torch::Tensor evict_count, torch::Tensor evict_rows);
+void store_update(torch::Tensor table_keys, torch::Tensor ticks,
+                  torch::Tensor arena, torch::Tensor query,
+                  torch::Tensor grads, int64_t dim, int64_t opt,
+                  std::vector<double> params, double b1_power,
+                  double b2_power, double weight_bound, torch::Tensor skipped);
+torch::Tensor segment_sum(torch::Tensor rows, torch::Tensor inverse,
+                          torch::Tensor seg_offsets, torch::Tensor seg_scale);
+void grad_scatter(torch::Tensor grads, torch::Tensor perm,
+                  torch::Tensor ustarts, torch::Tensor seg_id,
+                  torch::Tensor seg_scale, torch::Tensor out,
+                  int64_t accumulate);
+torch::Tensor sign_prep(torch::Tensor values, torch::Tensor slot_starts,
+                        torch::Tensor prefixes, int64_t spacing);
+
+static constexpr int64_t kFlip = std::numeric_limits<int64_t>::min();
+
+// sort-based dedup (torch.sort on GPU = rocPRIM radix/merge sort).
+// -> (uniq ascending u64-order, inverse, perm, ustarts)
+std::vector<torch::Tensor> dedup_keys(torch::Tensor keys) {
+  auto dev = keys.device();
+  const int64_t nnz = keys.numel();
+  auto opts = torch::TensorOptions().dtype(torch::kInt64).device(dev);
+  if (nnz == 0) {
+    auto z = torch::zeros({0}, opts);
+    return {z, z, z, torch::zeros({1}, opts)};
+  }
+  auto flipped = keys.bitwise_xor(kFlip);
+  auto sorted = flipped.sort();
+  auto svals = std::get<0>(sorted);
+  auto perm = std::get<1>(sorted);
+  auto neq = torch::ones({nnz}, opts.dtype(torch::kBool));
+  neq.slice(0, 1, nnz) =
+      svals.slice(0, 1, nnz).ne(svals.slice(0, 0, nnz - 1));
+  auto inv_sorted = neq.cumsum(0) - 1;
+  auto inverse = torch::empty({nnz}, opts);
+  inverse.index_put_({perm}, inv_sorted);
+  auto uniq = svals.masked_select(neq).bitwise_xor(kFlip);
+  auto starts = neq.nonzero().view(-1);
+  auto ustarts = torch::cat({starts, torch::full({1}, nnz, opts)});
+  return {uniq, inverse, perm, ustarts};
+}
+
+// Single-GPU fused lookup: raw values -> per-sample summed embeddings.
+// Returns {sums f16 [n_segs, dim], uniq_keys, inverse, perm, ustarts}.
+std::vector<torch::Tensor> lookup_local(
+    torch::Tensor values, torch::Tensor slot_starts, torch::Tensor prefixes,
+    int64_t spacing, torch::Tensor cat_offsets, torch::Tensor seg_scale,
+    torch::Tensor table_keys, torch::Tensor ticks, torch::Tensor arena,
+    int64_t dim, int64_t train, int64_t tick, double lo, double hi,
+    double admit_prob, double state_init, int64_t opt_space) {
+  auto keys = sign_prep(values, slot_starts, prefixes, spacing);
+  auto d = dedup_keys(keys);
+  auto& uniq = d[0];
+  auto rows = torch::empty(
+      {uniq.numel(), dim},
+      torch::TensorOptions().dtype(torch::kFloat32).device(values.device()));
+  auto none = torch::empty(
+      {0}, torch::TensorOptions().dtype(torch::kInt64).device(values.device()));
+  auto none_i32 = torch::empty(
+      {0}, torch::TensorOptions().dtype(torch::kInt32).device(values.device()));
+  auto none_f32 = torch::empty(
+      {0}, torch::TensorOptions().dtype(torch::kFloat32).device(values.device()));
+  store_lookup(table_keys, ticks, arena, uniq, rows, dim, train, tick, lo, hi,
+               admit_prob, state_init, opt_space, none, none_i32, none_f32);
+  auto sums = segment_sum(rows, d[1], cat_offsets, seg_scale);
+  return {sums, uniq, d[1], d[2], d[3]};
+}
+
+// Single-GPU fused backward: per-segment grads -> per-sign scatter -> fused
+// optimizer update on the shard.
+void update_local(torch::Tensor grads, torch::Tensor perm,
+                  torch::Tensor ustarts, torch::Tensor seg_id,
+                  torch::Tensor seg_scale, torch::Tensor uniq_keys,
+                  torch::Tensor table_keys, torch::Tensor ticks,
+                  torch::Tensor arena, int64_t dim, int64_t opt,
+                  std::vector<double> params, double b1_power, double b2_power,
+                  double weight_bound, torch::Tensor skipped) {
+  auto buf = torch::empty(
+      {uniq_keys.numel(), dim},
+      torch::TensorOptions().dtype(torch::kFloat32).device(grads.device()));
+  grad_scatter(grads, perm, ustarts, seg_id, seg_scale, buf, /*accumulate=*/0);
+  store_update(table_keys, ticks, arena, uniq_keys, buf, dim, opt, params,
+               b1_power, b2_power, weight_bound, skipped);
+}
+
+void init_engine(pybind11::module_& m) {
+  m.def("dedup_keys", &dedup_keys, "sort-based dedup of u64 keys");
+  m.def("lookup_local", &lookup_local,
+        "fused single-GPU lookup (sign prep + dedup + probe + gather + sum)");
+  m.def("update_local", &update_local,
+        "fused single-GPU backward (scatter + optimizer update)");
+}

@@ -15,7 +15,11 @@ this_dir = os.path.dirname(os.path.abspath(__file__))
 
 ext = CUDAExtension(
     name="persia_amd._C",
-    sources=["persia_amd/csrc/kernels.hip", "persia_amd/csrc/dense.hip"],
+    sources=[
+        "persia_amd/csrc/kernels.hip",
+        "persia_amd/csrc/dense.hip",
+        "persia_amd/csrc/engine.cpp",
+    ],
     include_dirs=[os.path.join(this_dir, "persia_amd", "csrc")],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
