@@ -151,6 +151,8 @@ class EmbeddingCtx(BaseCtx):
         super().__init__(**kwargs)
         self.preprocess_mode = preprocess_mode
         self.model = model
+        if model is not None and self.device.type == "cuda":
+            model.to(self.device)  # reference TrainCtx moves the dense model
         self.embedding_config = embedding_config or EmbeddingConfig()
         global _LAST_ENGINE
         if engine is not None:
