@@ -604,13 +604,15 @@ torch::Tensor sign_prep(torch::Tensor values, torch::Tensor slot_starts,
   return out;
 }
 
-void init_dense(pybind11::module_& m);   // csrc/dense.hip
-void init_engine(pybind11::module_& m);  // csrc/engine.cpp
+void init_dense(pybind11::module_& m);     // csrc/dense.hip
+void init_engine(pybind11::module_& m);    // csrc/engine.cpp
+void init_interact(pybind11::module_& m);  // csrc/interact.hip
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "persia_amd HIP kernels (gfx950)";
   init_dense(m);
   init_engine(m);
+  init_interact(m);
   m.def("store_lookup", &store_lookup, "hash-table lookup/insert + gather");
   m.def("store_probe", &store_probe, "probe-only (spill-tier miss detection)");
   m.def("store_update", &store_update, "fused sparse optimizer update");

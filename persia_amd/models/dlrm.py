@@ -20,13 +20,38 @@ def _mlp(sizes: List[int], last_relu: bool = False) -> nn.Sequential:
     return nn.Sequential(*layers)
 
 
+class _InteractFn(torch.autograd.Function):
+    """Fused pairwise-dot interaction (csrc/interact.hip): hipBLASLt runs
+    this batched tiny GEMM at <10 TF; the fused wave-per-sample kernel is
+    memory-bound on one read of V."""
+
+    @staticmethod
+    def forward(ctx, v):
+        from persia_amd.ops import native
+
+        v = v.to(torch.bfloat16).contiguous()
+        ctx.save_for_backward(v)
+        return native().interact_fwd(v)
+
+    @staticmethod
+    def backward(ctx, g):
+        from persia_amd.ops import native
+
+        (v,) = ctx.saved_tensors
+        return native().interact_bwd(g.to(torch.bfloat16).contiguous(), v)
+
+
 class DotInteraction(nn.Module):
-    """Pairwise dot products of the (num_slots+1) feature vectors, lower
-    triangle (the DLRM interaction op; fused HIP kernel planned —
-    the matmul runs on MFMA via hipBLASLt through torch.bmm)."""
+    """Pairwise dot products of the (num_slots+1) feature vectors, strict
+    lower triangle in torch.tril_indices order."""
 
     def forward(self, vectors: torch.Tensor) -> torch.Tensor:
         # vectors: [B, F, D]
+        if vectors.is_cuda and vectors.shape[2] % 8 == 0:
+            from persia_amd.ops import native_available
+
+            if native_available():
+                return _InteractFn.apply(vectors)
         B, F, _D = vectors.shape
         prod = torch.bmm(vectors, vectors.transpose(1, 2))  # [B, F, F]
         li, lj = torch.tril_indices(F, F, offset=-1, device=vectors.device)

@@ -18,6 +18,7 @@ ext = CUDAExtension(
     sources=[
         "persia_amd/csrc/kernels.hip",
         "persia_amd/csrc/dense.hip",
+        "persia_amd/csrc/interact.hip",
         "persia_amd/csrc/engine.cpp",
     ],
     include_dirs=[os.path.join(this_dir, "persia_amd", "csrc")],

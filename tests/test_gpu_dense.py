@@ -104,6 +104,46 @@ def test_fused_linear_autograd_matches_torch(act):
         assert torch.allclose(x.grad.float(), x2.grad, atol=0.3, rtol=0.05)
 
 
+def test_interaction_matches_torch():
+    from persia_amd.ops import native
+
+    C = native()
+    torch.manual_seed(5)
+    for B, F, D in [(64, 27, 128), (257, 9, 64), (1024, 27, 128)]:
+        V = (torch.randn(B, F, D, device=_dev()) * 0.5).to(torch.bfloat16).contiguous()
+        out = C.interact_fwd(V)
+        prod = torch.bmm(V.float(), V.float().transpose(1, 2))
+        li, lj = torch.tril_indices(F, F, offset=-1, device=_dev())
+        ref = prod[:, li, lj]
+        assert torch.allclose(out.float(), ref, atol=0.2, rtol=0.02), (
+            f"{B}x{F}x{D} fwd max err {(out.float() - ref).abs().max()}"
+        )
+        g = torch.randn_like(ref).to(torch.bfloat16)
+        dV = C.interact_bwd(g.contiguous(), V)
+        # reference dV via autograd
+        V2 = V.float().detach().requires_grad_(True)
+        prod2 = torch.bmm(V2, V2.transpose(1, 2))
+        prod2[:, li, lj].backward(g.float())
+        assert torch.allclose(dV.float(), V2.grad, atol=0.3, rtol=0.02), (
+            f"{B}x{F}x{D} bwd max err {(dV.float() - V2.grad).abs().max()}"
+        )
+
+
+def test_interaction_autograd_in_dlrm_path():
+    from persia_amd.models.dlrm import DotInteraction
+
+    torch.manual_seed(6)
+    B, F, D = 128, 9, 64
+    v = torch.randn(B, F, D, device=_dev(), dtype=torch.bfloat16, requires_grad=True)
+    out = DotInteraction()(v)
+    out.sum().backward()
+    v2 = v.detach().float().requires_grad_(True)
+    prod = torch.bmm(v2, v2.transpose(1, 2))
+    li, lj = torch.tril_indices(F, F, offset=-1, device=_dev())
+    prod[:, li, lj].sum().backward()
+    assert torch.allclose(v.grad.float(), v2.grad, atol=0.3, rtol=0.02)
+
+
 def test_fused_mlp_matches_torch_training():
     """Train FusedMLP and an identically-initialized torch MLP on the same
     data: the loss trajectories must track each other (bf16-level agreement)."""
