@@ -37,7 +37,8 @@ __global__ __launch_bounds__(256) void interact_fwd_kernel(
     int P) {
   extern __shared__ short lds[];
   const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
-  short* v = lds + wave * F * D;
+  const int LD = D + 8;  // +8-short row pad: conflict-free same-column reads
+  short* v = lds + wave * F * LD;
   // uniform trip count across all waves of the block (barriers inside)
   const int64_t stride = (int64_t)gridDim.x * 4;
   const int64_t rounds = ((int64_t)B + stride - 1) / stride;
@@ -45,9 +46,11 @@ __global__ __launch_bounds__(256) void interact_fwd_kernel(
     const int64_t b = r * stride + (int64_t)blockIdx.x * 4 + wave;
     if (b < B) {
       const short* src = V + b * F * D;
-      const int total8 = F * D / 8;
-      for (int t = lane; t < total8; t += 64)
-        *(bf16x8i*)&v[t * 8] = *(const bf16x8i*)&src[t * 8];
+      const int d8 = D / 8;
+      for (int t = lane; t < F * d8; t += 64) {
+        const int fr = t / d8, c8 = (t % d8) * 8;
+        *(bf16x8i*)&v[fr * LD + c8] = *(const bf16x8i*)&src[fr * D + c8];
+      }
     }
     __syncthreads();
     if (b < B) {
@@ -57,8 +60,8 @@ __global__ __launch_bounds__(256) void interact_fwd_kernel(
         while (i * (i - 1) / 2 > p) --i;
         while ((i + 1) * i / 2 <= p) ++i;
         const int j = p - i * (i - 1) / 2;
-        const short* vi = &v[i * D];
-        const short* vj = &v[j * D];
+        const short* vi = &v[i * LD];
+        const short* vj = &v[j * LD];
         float acc = 0.0f;
         for (int d = 0; d < D; d += 8) {
           bf16x8i a = *(const bf16x8i*)&vi[d];
@@ -81,17 +84,20 @@ __global__ __launch_bounds__(256) void interact_bwd_kernel(
   const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
   // per-wave carve: V (F*D shorts) + G (F*F floats stored as 2-short = use
   // float region) — keep G as floats for accuracy
-  short* v = lds + wave * (F * D + 2 * F * F);
-  float* G = (float*)(v + F * D);
+  const int LD = D + 8;
+  short* v = lds + wave * (F * LD + 2 * F * F);
+  float* G = (float*)(v + F * LD);
   const int64_t stride = (int64_t)gridDim.x * 4;
   const int64_t rounds = ((int64_t)B + stride - 1) / stride;
   for (int64_t r = 0; r < rounds; ++r) {
     const int64_t b = r * stride + (int64_t)blockIdx.x * 4 + wave;
     if (b < B) {
       const short* src = V + b * F * D;
-      const int total8 = F * D / 8;
-      for (int t = lane; t < total8; t += 64)
-        *(bf16x8i*)&v[t * 8] = *(const bf16x8i*)&src[t * 8];
+      const int d8l = D / 8;
+      for (int t = lane; t < F * d8l; t += 64) {
+        const int fr = t / d8l, c8 = (t % d8l) * 8;
+        *(bf16x8i*)&v[fr * LD + c8] = *(const bf16x8i*)&src[fr * D + c8];
+      }
       for (int t = lane; t < F * F; t += 64) G[t] = 0.0f;
     }
     __syncthreads();
@@ -118,7 +124,7 @@ __global__ __launch_bounds__(256) void interact_bwd_kernel(
         for (int j = 0; j < F; ++j) {
           const float gij = G[i * F + j];
           if (gij != 0.0f) {
-            const short* vj = &v[j * D + dd];
+            const short* vj = &v[j * LD + dd];
 #pragma unroll
             for (int k = 0; k < 8; ++k) acc[k] += gij * ibf2f(vj[k]);
           }
@@ -144,7 +150,7 @@ torch::Tensor interact_fwd(torch::Tensor V) {
   const int P = F * (F - 1) / 2;
   auto out = torch::empty(
       {B, P}, torch::TensorOptions().dtype(torch::kBFloat16).device(V.device()));
-  const int lds = 4 * F * D * 2;
+  const int lds = 4 * F * (D + 8) * 2;
   const int grid = std::min((B + 3) / 4, 8192);
   hipLaunchKernelGGL(interact_fwd_kernel, dim3(grid), dim3(256), lds,
                      icur_stream(), (const short*)V.data_ptr(),
@@ -156,7 +162,7 @@ torch::Tensor interact_bwd(torch::Tensor g, torch::Tensor V) {
   const int B = (int)V.size(0), F = (int)V.size(1), D = (int)V.size(2);
   const int P = F * (F - 1) / 2;
   auto dV = torch::empty_like(V);
-  const int lds = 4 * (F * D * 2 + 4 * F * F);
+  const int lds = 4 * (F * (D + 8) * 2 + 4 * F * F);
   TORCH_CHECK(lds <= 65536, "interaction tile exceeds LDS");
   const int grid = std::min((B + 3) / 4, 8192);
   hipLaunchKernelGGL(interact_bwd_kernel, dim3(grid), dim3(256), lds,
