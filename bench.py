@@ -198,8 +198,8 @@ def main():
                     loss = loss_fn(logits.float(), static["label"])
                 loss.backward()
                 opt.step()
-                for p in model.parameters():
-                    p.grad.zero_()
+                grads = [p.grad for p in model.parameters() if p.grad is not None]
+                torch._foreach_zero_(grads)  # one multi-tensor launch
                 return loss
 
             # warmup on a side stream (allocator state, autotuned GEMMs)
