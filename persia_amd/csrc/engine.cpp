@@ -29,6 +29,14 @@ void grad_scatter(torch::Tensor grads, torch::Tensor perm,
                   int64_t accumulate);
 torch::Tensor sign_prep(torch::Tensor values, torch::Tensor slot_starts,
                         torch::Tensor prefixes, int64_t spacing);
+void scatter_update(torch::Tensor table_keys, torch::Tensor ticks,
+                    torch::Tensor arena, torch::Tensor uniq,
+                    torch::Tensor grads, torch::Tensor perm,
+                    torch::Tensor ustarts, torch::Tensor seg_id,
+                    torch::Tensor seg_scale, int64_t dim, int64_t opt,
+                    std::vector<double> params, double b1_power,
+                    double b2_power, double weight_bound,
+                    torch::Tensor skipped);
 
 static constexpr int64_t kFlip = std::numeric_limits<int64_t>::min();
 
@@ -93,6 +101,13 @@ void update_local(torch::Tensor grads, torch::Tensor perm,
                   torch::Tensor arena, int64_t dim, int64_t opt,
                   std::vector<double> params, double b1_power, double b2_power,
                   double weight_bound, torch::Tensor skipped) {
+  if (grads.scalar_type() == torch::kFloat16 && dim <= 512) {
+    // single fused kernel: ordered scatter + optimizer, no [U,dim] buffer
+    scatter_update(table_keys, ticks, arena, uniq_keys, grads, perm, ustarts,
+                   seg_id, seg_scale, dim, opt, params, b1_power, b2_power,
+                   weight_bound, skipped);
+    return;
+  }
   auto buf = torch::empty(
       {uniq_keys.numel(), dim},
       torch::TensorOptions().dtype(torch::kFloat32).device(grads.device()));

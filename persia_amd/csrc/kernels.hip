@@ -418,6 +418,111 @@ __global__ void grad_scatter_kernel(const GradT* __restrict__ grads,
   }
 }
 
+// ------------------------------------------- fused scatter + optimizer update
+
+// One wave per unique sign: reduce its gradient from the per-segment grads
+// (ordered, via the sort permutation) straight into registers, probe the
+// table, and apply the optimizer — no intermediate [U, dim] buffer.
+// Supports dim <= 8*PA_WAVE (register-resident grad row).
+__global__ void scatter_update_kernel(
+    ull* __restrict__ table_keys, unsigned* __restrict__ ticks,
+    float* __restrict__ arena, const ull* __restrict__ uniq,
+    const __half* __restrict__ grads, const int64_t* __restrict__ perm,
+    const int64_t* __restrict__ ustarts, const int64_t* __restrict__ seg_id,
+    const float* __restrict__ seg_scale, int64_t n, int dim, int row_width,
+    int64_t n_buckets, int opt, float p0, float p1, float p2, float p3,
+    float b1_power, float b2_power, float weight_bound,
+    int* __restrict__ skipped) {
+  const int wave = threadIdx.x / PA_WAVE;
+  const int lane = threadIdx.x % PA_WAVE;
+  const int waves_per_block = blockDim.x / PA_WAVE;
+  const int64_t mask = n_buckets - 1;
+  const int nchunk = (dim + PA_WAVE - 1) / PA_WAVE;
+  for (int64_t u = (int64_t)blockIdx.x * waves_per_block + wave; u < n;
+       u += (int64_t)gridDim.x * waves_per_block) {
+    const ull k = uniq[u];
+    // lane-parallel probe over the 32-slot window
+    long long slot = -1;
+    const int64_t b = (int64_t)(k & (ull)mask);
+    if (lane < PA_PROBE_BUCKETS * PA_BUCKET_SIZE) {
+      const int p = lane / PA_BUCKET_SIZE;
+      const int s = lane % PA_BUCKET_SIZE;
+      const int64_t j = ((b + p) & mask) * PA_BUCKET_SIZE + s;
+      if (table_keys[j] == k) slot = j;
+    }
+    const unsigned long long found = __ballot(slot >= 0);
+    if (found == 0) {
+      if (lane == 0) atomicAdd(&skipped[0], 1);
+      continue;
+    }
+    slot = __shfl(slot, __ffsll((long long)found) - 1);
+    // ordered gradient reduction into registers
+    float acc[8];
+#pragma unroll
+    for (int t = 0; t < 8; ++t) acc[t] = 0.0f;
+    const int64_t lo = ustarts[u], hi = ustarts[u + 1];
+    for (int64_t p = lo; p < hi; ++p) {
+      const int64_t s = seg_id[perm[p]];
+      if (s < 0) continue;
+      const float sc = seg_scale ? seg_scale[s] : 1.0f;
+      if (sc == 0.0f) continue;
+      const __half* g = grads + s * dim;
+      for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t)
+        acc[t] += __half2float(g[c]) * sc;
+    }
+    bool has_nan = false;
+    for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t)
+      has_nan |= isnan(acc[t]);
+    if (__ballot(has_nan) != 0) {
+      if (lane == 0) atomicAdd(&skipped[1], 1);
+      continue;
+    }
+    float* row = arena + (int64_t)slot * row_width;
+    if (opt == 0) {  // SGD
+      for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
+        float w = row[c] - p0 * (acc[t] + p1 * row[c]);
+        if (weight_bound > 0.0f) w = fminf(fmaxf(w, -weight_bound), weight_bound);
+        row[c] = w;
+      }
+    } else if (opt == 1) {  // Adagrad
+      if (p3 > 0.5f) {
+        const float a0 = row[dim];
+        float gsq = 0.0f;
+        for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
+          float w = row[c] - p0 * acc[t] * rsqrtf(a0 + p2);
+          if (weight_bound > 0.0f) w = fminf(fmaxf(w, -weight_bound), weight_bound);
+          row[c] = w;
+          gsq += acc[t] * acc[t];
+        }
+#pragma unroll
+        for (int off = PA_WAVE / 2; off > 0; off >>= 1)
+          gsq += __shfl_down(gsq, off);
+        if (lane == 0) row[dim] = a0 * p1 + gsq / (float)dim;
+      } else {
+        for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
+          const float a0 = row[dim + c];
+          float w = row[c] - p0 * acc[t] * rsqrtf(a0 + p2);
+          if (weight_bound > 0.0f) w = fminf(fmaxf(w, -weight_bound), weight_bound);
+          row[c] = w;
+          row[dim + c] = a0 * p1 + acc[t] * acc[t];
+        }
+      }
+    } else {  // Adam
+      const float om1 = 1.0f - p1, om2 = 1.0f - p2;
+      const float c1 = 1.0f / (1.0f - b1_power), c2 = 1.0f / (1.0f - b2_power);
+      for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
+        const float m = p1 * row[dim + c] + om1 * acc[t];
+        const float v = p2 * row[2 * dim + c] + om2 * acc[t] * acc[t];
+        float w = row[c] - p0 * (m * c1) / (p3 + sqrtf(v * c2));
+        if (weight_bound > 0.0f) w = fminf(fmaxf(w, -weight_bound), weight_bound);
+        row[c] = w;
+        row[dim + c] = m;
+        row[2 * dim + c] = v;
+      }
+    }
+  }
+}
+
 inline int n_blocks_for(int64_t work_items, int per_block) {
   int64_t b = (work_items + per_block - 1) / per_block;
   // >> 256 CUs needed to fill the chip; cap and grid-stride beyond
@@ -590,6 +695,37 @@ void grad_scatter(torch::Tensor grads, torch::Tensor perm,
 #undef PA_GS
 }
 
+void scatter_update(torch::Tensor table_keys, torch::Tensor ticks,
+                    torch::Tensor arena, torch::Tensor uniq,
+                    torch::Tensor grads, torch::Tensor perm,
+                    torch::Tensor ustarts, torch::Tensor seg_id,
+                    torch::Tensor seg_scale, int64_t dim, int64_t opt,
+                    std::vector<double> params, double b1_power,
+                    double b2_power, double weight_bound,
+                    torch::Tensor skipped) {
+  const int64_t n = uniq.numel();
+  if (n == 0) return;
+  TORCH_CHECK(grads.scalar_type() == torch::kFloat16, "scatter_update: f16 grads");
+  TORCH_CHECK(dim <= 8 * PA_WAVE, "scatter_update: dim too large");
+  const int64_t n_buckets = table_keys.numel() / PA_BUCKET_SIZE;
+  const int row_width = (int)arena.size(1);
+  const float* scale_ptr =
+      seg_scale.numel() ? seg_scale.data_ptr<float>() : nullptr;
+  hipLaunchKernelGGL(scatter_update_kernel, dim3(n_blocks_for(n, 4)),
+                     dim3(256), 0, cur_stream(),
+                     (ull*)table_keys.data_ptr<int64_t>(),
+                     (unsigned*)ticks.data_ptr<int32_t>(),
+                     arena.data_ptr<float>(),
+                     (const ull*)uniq.data_ptr<int64_t>(),
+                     (const __half*)grads.data_ptr<at::Half>(),
+                     perm.data_ptr<int64_t>(), ustarts.data_ptr<int64_t>(),
+                     seg_id.data_ptr<int64_t>(), scale_ptr, n, (int)dim,
+                     row_width, n_buckets, (int)opt, (float)params[0],
+                     (float)params[1], (float)params[2], (float)params[3],
+                     (float)b1_power, (float)b2_power, (float)weight_bound,
+                     skipped.data_ptr<int32_t>());
+}
+
 torch::Tensor sign_prep(torch::Tensor values, torch::Tensor slot_starts,
                         torch::Tensor prefixes, int64_t spacing) {
   const int64_t n = values.numel();
@@ -620,4 +756,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("segment_sum", &segment_sum, "fused gather + per-sample summation");
   m.def("grad_scatter", &grad_scatter, "ordered per-sign gradient scatter");
   m.def("sign_prep", &sign_prep, "prefix-fold + splitmix64 key mixing");
+  m.def("scatter_update", &scatter_update,
+        "fused ordered grad scatter + optimizer update (no [U,dim] buffer)");
 }
