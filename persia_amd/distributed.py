@@ -93,26 +93,50 @@ class DDPOption(DistributedBaseOption):
 
 
 class BaguaDistributedOption(DistributedBaseOption):
-    """API-parity stub for the reference's Bagua option
-    (persia/distributed.py:204-256).  Bagua has no ROCm build; the
-    ``gradient_allreduce`` algorithm maps to plain DDP, everything else
-    raises."""
+    """The reference's Bagua option (persia/distributed.py:204-256) mapped to
+    RCCL-native equivalents (Bagua itself has no ROCm build):
+
+    * ``gradient_allreduce``      -> plain DDP (identical algorithm)
+    * ``bytegrad`` / ``qadam``    -> DDP + bf16-compressed allreduce hook
+      (torch's powerSGD-family comm hooks over RCCL; Bagua's 8-bit codec is
+      approximated by bf16 compression — half the wire bytes)
+    * ``low_precision_decentralized`` -> same compressed hook (decentralized
+      topologies are pointless on a fully-connected xGMI node)
+    * ``decentralized``           -> DDP with a wider bucket (peer averaging
+      on an 8-GPU all-to-all mesh degenerates to allreduce)
+    * ``async``                   -> not supported (raises): the sparse side
+      is already asynchronous; async dense averaging trades determinism for
+      nothing on one node
+    """
+
+    _COMPRESSED = {"bytegrad", "qadam", "low_precision_decentralized"}
+    _SUPPORTED = _COMPRESSED | {"gradient_allreduce", "decentralized"}
 
     def __init__(self, algorithm: str = "gradient_allreduce", **options):
         super().__init__()
         self.algorithm = algorithm
-        self._ddp = DDPOption()
-        if algorithm != "gradient_allreduce":
+        if algorithm not in self._SUPPORTED:
             raise NotImplementedError(
-                f"Bagua algorithm {algorithm!r} is not available on ROCm; "
-                "use DDPOption (gradient_allreduce is equivalent)"
+                f"Bagua algorithm {algorithm!r} has no MI355X mapping "
+                f"(supported: {sorted(self._SUPPORTED)})"
             )
+        bucket = 100 if algorithm == "decentralized" else 50
+        self._ddp = DDPOption(bucket_cap_mb=bucket, **options)
 
     def init_process_group(self, device_id):
         self._ddp.init_process_group(device_id)
 
     def wrap_model(self, model, device_id):
-        return self._ddp.wrap_model(model, device_id)
+        wrapped = self._ddp.wrap_model(model, device_id)
+        if self.algorithm in self._COMPRESSED:
+            from torch.distributed.algorithms.ddp_comm_hooks import (
+                default_hooks,
+            )
+
+            wrapped.register_comm_hook(
+                state=None, hook=default_hooks.bf16_compress_hook
+            )
+        return wrapped
 
 
 def get_default_distributed_option(device_id: Optional[int] = None) -> DDPOption:
