@@ -220,15 +220,30 @@ def main():
                   file=_sys.stderr, flush=True)
             graph = None
 
+    timing = os.environ.get("PA_BENCH_TIMING", "0") == "1"
+    tstats = {"get": 0.0, "copy": 0.0, "replay": 0.0, "apply": 0.0, "n": 0}
+
     def train_step(tb):
         if graph is not None:
+            if timing:
+                t0 = time.perf_counter()
             with torch.no_grad():
                 static["dense"].copy_(tb.non_id_type_tensors[0], non_blocking=True)
                 static["base"].copy_(tb._groups[0].sum_base, non_blocking=True)
                 static["label"].copy_(tb.label_tensors[0], non_blocking=True)
+            if timing:
+                t1 = time.perf_counter()
             graph.replay()
+            if timing:
+                t2 = time.perf_counter()
             engine.apply_gradients_base(tb, sum_base_grads=[static["base"].grad])
             pipeline.release_permit()
+            if timing:
+                t3 = time.perf_counter()
+                tstats["copy"] += t1 - t0
+                tstats["replay"] += t2 - t1
+                tstats["apply"] += t3 - t2
+                tstats["n"] += 1
             return static["loss"]
         with amp_ctx:
             embs = tb.training_embeddings()
@@ -264,7 +279,11 @@ def main():
     feeder = threading.Thread(target=feed, args=(args.warmup, args.steps), daemon=True)
     feeder.start()
     for _ in range(args.steps):
-        train_step(pipeline.get())
+        tg0 = time.perf_counter() if timing else 0.0
+        tb = pipeline.get()
+        if timing:
+            tstats["get"] += time.perf_counter() - tg0
+        train_step(tb)
     if world > 1:
         dist.barrier()
     if use_gpu:
@@ -278,6 +297,16 @@ def main():
         elapsed = float(t.item())
 
     pipeline.stop()
+    if timing and tstats["n"]:
+        import sys as _s
+
+        n = tstats["n"]
+        print(
+            f"# phase ms/step: get={tstats['get']/n*1e3:.3f} "
+            f"copy={tstats['copy']/n*1e3:.3f} replay={tstats['replay']/n*1e3:.3f} "
+            f"apply={tstats['apply']/n*1e3:.3f}",
+            file=_s.stderr, flush=True,
+        )
     samples = args.steps * B * world
     if rank == 0:
         result = {
