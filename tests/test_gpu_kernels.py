@@ -248,6 +248,8 @@ def test_engine_gpu_matches_cpu_engine():
     from persia_amd.embedding.data import IDTypeFeature, Label, PersiaBatch
     from persia_amd.embedding.optim import Adagrad
 
+    from persia_amd.core.schema import HashStackConfig
+
     def mk_engine(device):
         return EmbeddingEngine(
             schema=EmbeddingSchema(
@@ -256,6 +258,12 @@ def test_engine_gpu_matches_cpu_engine():
                     "b": SlotConfig(name="b", dim=16, sqrt_scaling=True),
                     "r": SlotConfig(name="r", dim=16, embedding_summation=False,
                                     sample_fixed_size=4),
+                    "h": SlotConfig(
+                        name="h", dim=16,
+                        hash_stack_config=HashStackConfig(
+                            hash_stack_rounds=2, embedding_size=64
+                        ),
+                    ),
                 }
             ),
             hyper=EmbeddingConfig(emb_initialization=(-0.5, 0.5)),
@@ -273,7 +281,7 @@ def test_engine_gpu_matches_cpu_engine():
                 n, [rng.integers(0, 300, size=rng.integers(0, 6), dtype=np.uint64)
                     for _ in range(B)]
             )
-            for n in ("a", "b", "r")
+            for n in ("a", "b", "r", "h")
         ]
         return PersiaBatch(feats, labels=[Label(np.ones((B, 1), np.float32))],
                            requires_grad=True)
@@ -298,6 +306,7 @@ def test_engine_gpu_matches_cpu_engine():
         g = {
             "a": torch.full((32, 16), 0.25, dtype=torch.float16),
             "b": torch.full((32, 16), -0.5, dtype=torch.float16),
+            "h": torch.full((32, 16), 0.125, dtype=torch.float16),
             "r": torch.full((tb_c.payloads[2].raw_distinct.shape[0] - 1, 16), 0.1),
         }
         gg = {k: v.to(_dev()) for k, v in g.items()}
