@@ -47,7 +47,9 @@ class DotInteraction(nn.Module):
 
     def forward(self, vectors: torch.Tensor) -> torch.Tensor:
         # vectors: [B, F, D]
-        if vectors.is_cuda and vectors.shape[2] % 8 == 0:
+        F, D = vectors.shape[1], vectors.shape[2]
+        bwd_lds = 4 * (F * (D + 8) * 2 + 4 * F * F)  # must fit 64KB (4 waves)
+        if vectors.is_cuda and D % 8 == 0 and bwd_lds <= 65536:
             from persia_amd.ops import native_available
 
             if native_available():
