@@ -52,7 +52,8 @@ class DistContext:
     def distributed(self) -> bool:
         return self.world_size > 1
 
-    def all_to_all_lengths(self, send_counts: List[int]) -> List[int]:
+    def all_to_all_lengths(self, send_counts: List[int],
+                           device: Optional[torch.device] = None) -> List[int]:
         """Exchange per-destination element counts (the two-phase counts/payload
         exchange the reference's RPC shape implies — SURVEY §7 hard part 3)."""
         inp = torch.tensor(send_counts, dtype=torch.int64)
@@ -60,7 +61,7 @@ class DistContext:
         if self._gloo_like():
             dist.all_to_all_single(out, inp, group=self.group)
         else:
-            dev = torch.device("cuda", torch.cuda.current_device())
+            dev = device or torch.device("cuda", torch.cuda.current_device())
             inp_d = inp.to(dev, non_blocking=True)
             out_d = torch.empty_like(inp_d)
             dist.all_to_all_single(out_d, inp_d, group=self.group)

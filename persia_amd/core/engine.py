@@ -338,7 +338,7 @@ class EmbeddingEngine:
             return rows
         owner = _owner_of_keys(group.uniq_keys, self.dist.world_size)
         send_counts = torch.bincount(owner, minlength=self.dist.world_size).tolist()
-        recv_counts = self.dist.all_to_all_lengths(send_counts)
+        recv_counts = self.dist.all_to_all_lengths(send_counts, self.device)
         group.send_counts, group.recv_counts = send_counts, recv_counts
         recv_keys = self.dist.all_to_all(group.uniq_keys, send_counts, recv_counts)
         rows_local = store.lookup(recv_keys, train).to(self.wire_dtype)
@@ -725,7 +725,7 @@ class EmbeddingEngine:
         if send_counts is None:
             owner = _owner_of_keys(group.uniq_keys, comm.world_size)
             send_counts = torch.bincount(owner, minlength=comm.world_size).tolist()
-            recv_counts = comm.all_to_all_lengths(send_counts)
+            recv_counts = comm.all_to_all_lengths(send_counts, self.device)
         keys_recv = comm.all_to_all(group.uniq_keys, send_counts, recv_counts)
         grads_recv = comm.all_to_all(
             buf.to(self.wire_dtype), send_counts, recv_counts
@@ -949,9 +949,11 @@ class ForwardPipeline:
 
     def _run(self):
         # On GPU, lookups run on a dedicated HIP stream so the sparse pipeline
-        # overlaps the dense fwd/bwd on the default stream.
+        # overlaps the dense fwd/bwd on the default stream.  A fresh thread
+        # defaults to device 0 — bind it to this rank's GPU first.
         stream = None
         if self.engine.device.type == "cuda":
+            torch.cuda.set_device(self.engine.device)
             stream = torch.cuda.Stream(device=self.engine.device)
         import heapq
 
