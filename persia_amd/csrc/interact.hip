@@ -82,11 +82,12 @@ __global__ __launch_bounds__(256) void interact_bwd_kernel(
     short* __restrict__ dV, int B, int F, int D, int P) {
   extern __shared__ short lds[];
   const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
-  // per-wave carve: V (F*D shorts) + G (F*F floats stored as 2-short = use
-  // float region) — keep G as floats for accuracy
+  // per-wave carve: V (F*LD shorts) + G (F*F bf16 — the incoming pair grads
+  // are bf16, so storing G as bf16 loses nothing; halves LDS so F=65/dim-8
+  // configs fit)
   const int LD = D + 8;
-  short* v = lds + wave * (F * LD + 2 * F * F);
-  float* G = (float*)(v + F * LD);
+  short* v = lds + wave * (F * LD + F * F);
+  short* G = v + F * LD;
   const int64_t stride = (int64_t)gridDim.x * 4;
   const int64_t rounds = ((int64_t)B + stride - 1) / stride;
   for (int64_t r = 0; r < rounds; ++r) {
@@ -98,7 +99,7 @@ __global__ __launch_bounds__(256) void interact_bwd_kernel(
         const int fr = t / d8l, c8 = (t % d8l) * 8;
         *(bf16x8i*)&v[fr * LD + c8] = *(const bf16x8i*)&src[fr * D + c8];
       }
-      for (int t = lane; t < F * F; t += 64) G[t] = 0.0f;
+      for (int t = lane; t < F * F; t += 64) G[t] = 0;
     }
     __syncthreads();
     if (b < B) {
@@ -108,7 +109,7 @@ __global__ __launch_bounds__(256) void interact_bwd_kernel(
         while (i * (i - 1) / 2 > p) --i;
         while ((i + 1) * i / 2 <= p) ++i;
         const int j = p - i * (i - 1) / 2;
-        const float gv = ibf2f(gp[p]);
+        const short gv = gp[p];
         G[i * F + j] = gv;
         G[j * F + i] = gv;
       }
@@ -122,7 +123,7 @@ __global__ __launch_bounds__(256) void interact_bwd_kernel(
         const int i = t / d8, dd = (t % d8) * 8;
         float acc[8] = {};
         for (int j = 0; j < F; ++j) {
-          const float gij = G[i * F + j];
+          const float gij = ibf2f(G[i * F + j]);
           if (gij != 0.0f) {
             const short* vj = &v[j * LD + dd];
 #pragma unroll
@@ -162,7 +163,7 @@ torch::Tensor interact_bwd(torch::Tensor g, torch::Tensor V) {
   const int B = (int)V.size(0), F = (int)V.size(1), D = (int)V.size(2);
   const int P = F * (F - 1) / 2;
   auto dV = torch::empty_like(V);
-  const int lds = 4 * (F * (D + 8) * 2 + 4 * F * F);
+  const int lds = 4 * (F * (D + 8) * 2 + 2 * F * F);
   TORCH_CHECK(lds <= 65536, "interaction tile exceeds LDS");
   const int grid = std::min((B + 3) / 4, 8192);
   hipLaunchKernelGGL(interact_bwd_kernel, dim3(grid), dim3(256), lds,

@@ -48,7 +48,7 @@ class DotInteraction(nn.Module):
     def forward(self, vectors: torch.Tensor) -> torch.Tensor:
         # vectors: [B, F, D]
         F, D = vectors.shape[1], vectors.shape[2]
-        bwd_lds = 4 * (F * (D + 8) * 2 + 4 * F * F)  # must fit 64KB (4 waves)
+        bwd_lds = 4 * (F * (D + 8) * 2 + 2 * F * F)  # must fit 64KB (4 waves)
         if vectors.is_cuda and D % 8 == 0 and bwd_lds <= 65536:
             from persia_amd.ops import native_available
 

@@ -109,7 +109,7 @@ def test_interaction_matches_torch():
 
     C = native()
     torch.manual_seed(5)
-    for B, F, D in [(64, 27, 128), (257, 9, 64), (1024, 27, 128)]:
+    for B, F, D in [(64, 27, 128), (257, 9, 64), (1024, 27, 128), (512, 65, 8)]:
         V = (torch.randn(B, F, D, device=_dev()) * 0.5).to(torch.bfloat16).contiguous()
         out = C.interact_fwd(V)
         prod = torch.bmm(V.float(), V.float().transpose(1, 2))
