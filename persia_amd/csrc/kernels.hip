@@ -172,29 +172,36 @@ __global__ void init_gather_kernel(float* __restrict__ arena,
   const int wave = threadIdx.x / PA_WAVE;
   const int lane = threadIdx.x % PA_WAVE;
   const int waves_per_block = blockDim.x / PA_WAVE;
-  for (int64_t i = (int64_t)blockIdx.x * waves_per_block + wave; i < n;
-       i += (int64_t)gridDim.x * waves_per_block) {
+  const int G = (dim < PA_WAVE && (PA_WAVE % dim) == 0) ? PA_WAVE / dim : 1;
+  const int sub = (G > 1) ? lane / dim : 0;
+  const int c0 = (G > 1) ? lane % dim : lane;
+  const int64_t n_waveitems = (n + G - 1) / G;
+  for (int64_t w = (int64_t)blockIdx.x * waves_per_block + wave;
+       w < n_waveitems; w += (int64_t)gridDim.x * waves_per_block) {
+    const int64_t i = w * G + sub;
+    if (i >= n) continue;
     const long long slot = slots[i];
     float* dst = out + i * dim;
     if (slot < 0) {
-      for (int c = lane; c < dim; c += PA_WAVE) dst[c] = 0.0f;
+      for (int c = c0; c < dim; c += PA_WAVE) dst[c] = 0.0f;
       continue;
     }
     float* row = arena + (int64_t)slot * row_width;
+    const int st = (G > 1) ? dim : PA_WAVE;  // subgroup stride
     if (evict_rows && is_new[i] && evict_idx[i] >= 0) {
       // spill: save the victim's full row before overwriting it
       float* spill = evict_rows + evict_idx[i] * row_width;
-      for (int c = lane; c < row_width; c += PA_WAVE) spill[c] = row[c];
+      for (int c = c0; c < row_width; c += st) spill[c] = row[c];
     }
     if (is_new[i]) {
       const uint64_t sign = pa_splitmix64_inv((uint64_t)query[i]);
       const uint64_t seed = pa_init_seed(sign);
-      for (int c = lane; c < dim; c += PA_WAVE)
+      for (int c = c0; c < dim; c += st)
         row[c] = pa_init_val(seed, c, lo, hi);
-      for (int c = dim + lane; c < row_width; c += PA_WAVE)
+      for (int c = dim + c0; c < row_width; c += st)
         row[c] = state_init;
     }
-    for (int c = lane; c < dim; c += PA_WAVE) dst[c] = row[c];
+    for (int c = c0; c < dim; c += st) dst[c] = row[c];
   }
 }
 
@@ -354,6 +361,8 @@ __global__ void import_kernel(ull* __restrict__ table_keys,
 // out[s, :] = seg_scale[s] * sum_{k in [off[s], off[s+1])} rows[inverse[k], :]
 // One wave per segment (all sum-slots of a dim-group in ONE launch);
 // f32 accumulate; f16 store (reference wire dtype, persia-common lib.rs:88-99).
+// Small dims pack 64/dim segments per wave (sub-wave groups) so dim-8
+// tables still use all lanes; large dims stride as before.
 template <typename RowT>
 __global__ void segment_sum_kernel(const RowT* __restrict__ rows,
                                    const int64_t* __restrict__ inverse,
@@ -364,12 +373,18 @@ __global__ void segment_sum_kernel(const RowT* __restrict__ rows,
   const int wave = threadIdx.x / PA_WAVE;
   const int lane = threadIdx.x % PA_WAVE;
   const int waves_per_block = blockDim.x / PA_WAVE;
-  for (int64_t s = (int64_t)blockIdx.x * waves_per_block + wave; s < n_seg;
-       s += (int64_t)gridDim.x * waves_per_block) {
+  const int G = (dim < PA_WAVE && (PA_WAVE % dim) == 0) ? PA_WAVE / dim : 1;
+  const int sub = (G > 1) ? lane / dim : 0;
+  const int c0 = (G > 1) ? lane % dim : lane;
+  const int64_t n_waveitems = (n_seg + G - 1) / G;
+  for (int64_t w = (int64_t)blockIdx.x * waves_per_block + wave;
+       w < n_waveitems; w += (int64_t)gridDim.x * waves_per_block) {
+    const int64_t s = w * G + sub;
+    if (s >= n_seg) continue;
     const int64_t lo = seg_offsets[s], hi = seg_offsets[s + 1];
     const float scale = seg_scale ? seg_scale[s] : 1.0f;
     __half* dst = out + s * dim;
-    for (int c = lane; c < dim; c += PA_WAVE) {
+    for (int c = c0; c < dim; c += PA_WAVE) {
       float acc = 0.0f;
       for (int64_t k = lo; k < hi; ++k) {
         acc += pa_to_float(rows[inverse[k] * dim + c]);
@@ -437,85 +452,104 @@ __global__ void scatter_update_kernel(
   const int lane = threadIdx.x % PA_WAVE;
   const int waves_per_block = blockDim.x / PA_WAVE;
   const int64_t mask = n_buckets - 1;
-  const int nchunk = (dim + PA_WAVE - 1) / PA_WAVE;
-  for (int64_t u = (int64_t)blockIdx.x * waves_per_block + wave; u < n;
-       u += (int64_t)gridDim.x * waves_per_block) {
-    const ull k = uniq[u];
-    // lane-parallel probe over the 32-slot window
+  // small dims pack 64/dim keys per wave (sub-wave groups)
+  const int G = (dim < PA_WAVE && (PA_WAVE % dim) == 0) ? PA_WAVE / dim : 1;
+  const int sub = (G > 1) ? lane / dim : 0;
+  const int c0 = (G > 1) ? lane % dim : lane;
+  const int st = (G > 1) ? dim : PA_WAVE;
+  const int sg = (G > 1) ? dim : PA_WAVE;  // subgroup width
+  const ull sg_mask = (sg >= 64) ? ~0ull : ((1ull << sg) - 1ull);
+  const int64_t n_waveitems = (n + G - 1) / G;
+  constexpr int WINDOW = PA_PROBE_BUCKETS * PA_BUCKET_SIZE;
+  for (int64_t w = (int64_t)blockIdx.x * waves_per_block + wave;
+       w < n_waveitems; w += (int64_t)gridDim.x * waves_per_block) {
+    const int64_t u = w * G + sub;
+    const bool active = u < n;
+    const ull k = active ? uniq[u] : 0;
+    // subgroup-parallel probe over the 32-slot window
     long long slot = -1;
-    const int64_t b = (int64_t)(k & (ull)mask);
-    if (lane < PA_PROBE_BUCKETS * PA_BUCKET_SIZE) {
-      const int p = lane / PA_BUCKET_SIZE;
-      const int s = lane % PA_BUCKET_SIZE;
-      const int64_t j = ((b + p) & mask) * PA_BUCKET_SIZE + s;
-      if (table_keys[j] == k) slot = j;
+    if (active) {
+      const int64_t b = (int64_t)(k & (ull)mask);
+      for (int it = 0; it * sg + c0 < WINDOW; ++it) {
+        const int widx = it * sg + c0;
+        const int p = widx / PA_BUCKET_SIZE;
+        const int s = widx % PA_BUCKET_SIZE;
+        const int64_t j = ((b + p) & mask) * PA_BUCKET_SIZE + s;
+        if (table_keys[j] == k) slot = j;
+      }
     }
     const unsigned long long found = __ballot(slot >= 0);
-    if (found == 0) {
-      if (lane == 0) atomicAdd(&skipped[0], 1);
-      continue;
+    const unsigned long long sub_found = (found >> (sub * sg)) & sg_mask;
+    bool ok = active;
+    if (ok && sub_found == 0) {
+      if (c0 == 0) atomicAdd(&skipped[0], 1);
+      ok = false;
     }
-    slot = __shfl(slot, __ffsll((long long)found) - 1);
+    if (ok)
+      slot = __shfl(slot, sub * sg + __ffsll((long long)sub_found) - 1);
     // ordered gradient reduction into registers
     float acc[8];
 #pragma unroll
     for (int t = 0; t < 8; ++t) acc[t] = 0.0f;
-    const int64_t lo = ustarts[u], hi = ustarts[u + 1];
-    for (int64_t p = lo; p < hi; ++p) {
-      const int64_t s = seg_id[perm[p]];
-      if (s < 0) continue;
-      const float sc = seg_scale ? seg_scale[s] : 1.0f;
-      if (sc == 0.0f) continue;
-      const __half* g = grads + s * dim;
-      for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t)
-        acc[t] += __half2float(g[c]) * sc;
+    if (ok) {
+      const int64_t lo = ustarts[u], hi = ustarts[u + 1];
+      for (int64_t p = lo; p < hi; ++p) {
+        const int64_t s = seg_id[perm[p]];
+        if (s < 0) continue;
+        const float sc = seg_scale ? seg_scale[s] : 1.0f;
+        if (sc == 0.0f) continue;
+        const __half* g = grads + s * dim;
+        for (int t = 0, c = c0; c < dim; c += st, ++t)
+          acc[t] += __half2float(g[c]) * sc;
+      }
     }
     bool has_nan = false;
-    for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t)
-      has_nan |= isnan(acc[t]);
-    if (__ballot(has_nan) != 0) {
-      if (lane == 0) atomicAdd(&skipped[1], 1);
-      continue;
+    if (ok)
+      for (int t = 0, c = c0; c < dim; c += st, ++t) has_nan |= isnan(acc[t]);
+    const unsigned long long nanb = __ballot(has_nan);
+    if (ok && ((nanb >> (sub * sg)) & sg_mask) != 0) {
+      if (c0 == 0) atomicAdd(&skipped[1], 1);
+      ok = false;
     }
+    if (!ok) continue;
     float* row = arena + (int64_t)slot * row_width;
     if (opt == 0) {  // SGD
-      for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
-        float w = row[c] - p0 * (acc[t] + p1 * row[c]);
-        if (weight_bound > 0.0f) w = fminf(fmaxf(w, -weight_bound), weight_bound);
-        row[c] = w;
+      for (int t = 0, c = c0; c < dim; c += st, ++t) {
+        float w2 = row[c] - p0 * (acc[t] + p1 * row[c]);
+        if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
+        row[c] = w2;
       }
     } else if (opt == 1) {  // Adagrad
       if (p3 > 0.5f) {
         const float a0 = row[dim];
         float gsq = 0.0f;
-        for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
-          float w = row[c] - p0 * acc[t] * rsqrtf(a0 + p2);
-          if (weight_bound > 0.0f) w = fminf(fmaxf(w, -weight_bound), weight_bound);
-          row[c] = w;
+        for (int t = 0, c = c0; c < dim; c += st, ++t) {
+          float w2 = row[c] - p0 * acc[t] * rsqrtf(a0 + p2);
+          if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
+          row[c] = w2;
           gsq += acc[t] * acc[t];
         }
-#pragma unroll
-        for (int off = PA_WAVE / 2; off > 0; off >>= 1)
-          gsq += __shfl_down(gsq, off);
-        if (lane == 0) row[dim] = a0 * p1 + gsq / (float)dim;
+        for (int off = sg / 2; off > 0; off >>= 1)
+          gsq += __shfl_down(gsq, off, sg);
+        if (c0 == 0) row[dim] = a0 * p1 + gsq / (float)dim;
       } else {
-        for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
+        for (int t = 0, c = c0; c < dim; c += st, ++t) {
           const float a0 = row[dim + c];
-          float w = row[c] - p0 * acc[t] * rsqrtf(a0 + p2);
-          if (weight_bound > 0.0f) w = fminf(fmaxf(w, -weight_bound), weight_bound);
-          row[c] = w;
+          float w2 = row[c] - p0 * acc[t] * rsqrtf(a0 + p2);
+          if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
+          row[c] = w2;
           row[dim + c] = a0 * p1 + acc[t] * acc[t];
         }
       }
     } else {  // Adam
       const float om1 = 1.0f - p1, om2 = 1.0f - p2;
       const float c1 = 1.0f / (1.0f - b1_power), c2 = 1.0f / (1.0f - b2_power);
-      for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
+      for (int t = 0, c = c0; c < dim; c += st, ++t) {
         const float m = p1 * row[dim + c] + om1 * acc[t];
         const float v = p2 * row[2 * dim + c] + om2 * acc[t] * acc[t];
-        float w = row[c] - p0 * (m * c1) / (p3 + sqrtf(v * c2));
-        if (weight_bound > 0.0f) w = fminf(fmaxf(w, -weight_bound), weight_bound);
-        row[c] = w;
+        float w2 = row[c] - p0 * (m * c1) / (p3 + sqrtf(v * c2));
+        if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
+        row[c] = w2;
         row[dim + c] = m;
         row[2 * dim + c] = v;
       }
