@@ -6,14 +6,22 @@ import torch.nn as nn
 
 
 class CrossLayerV2(nn.Module):
-    """x_{l+1} = x0 * (W x_l + b) + x_l (DCN-v2 full-rank cross)."""
+    """x_{l+1} = x0 * (U(V x_l) + b) + x_l — DCN-v2 cross layer in its
+    low-rank ("mixture of experts" degenerate) form, the production variant
+    of the DCN-v2 paper; rank=0 selects the full-rank W."""
 
-    def __init__(self, dim: int):
+    def __init__(self, dim: int, rank: int = 256):
         super().__init__()
-        self.w = nn.Linear(dim, dim)
+        if rank and rank < dim:
+            self.v = nn.Linear(dim, rank, bias=False)
+            self.u = nn.Linear(rank, dim)
+            self.w = None
+        else:
+            self.w = nn.Linear(dim, dim)
 
     def forward(self, x0: torch.Tensor, xl: torch.Tensor) -> torch.Tensor:
-        return x0 * self.w(xl) + xl
+        proj = self.w(xl) if self.w is not None else self.u(self.v(xl))
+        return x0 * proj + xl
 
 
 class DCNv2(nn.Module):
@@ -24,10 +32,13 @@ class DCNv2(nn.Module):
         dim: int = 64,
         num_cross: int = 3,
         deep: List[int] = (512, 256, 128),
+        cross_rank: int = 256,
     ):
         super().__init__()
         in_dim = num_dense + num_sparse * dim
-        self.cross = nn.ModuleList([CrossLayerV2(in_dim) for _ in range(num_cross)])
+        self.cross = nn.ModuleList(
+            [CrossLayerV2(in_dim, rank=cross_rank) for _ in range(num_cross)]
+        )
         layers: List[nn.Module] = []
         sizes = [in_dim] + list(deep)
         for i in range(len(sizes) - 1):
