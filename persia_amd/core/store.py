@@ -158,6 +158,15 @@ class EmbeddingStoreBase:
         self.tick += 1
         return t
 
+    def _spill_export(self):
+        """Host-tier rows for checkpointing (spilled rows are table state)."""
+        if self.spill is None or len(self.spill) == 0:
+            return None
+        keys = np.fromiter(self.spill._map.keys(), dtype=np.uint64,
+                           count=len(self.spill._map))
+        rows = np.stack(list(self.spill._map.values()))
+        return hashing.splitmix64_inv(keys), rows.astype(np.float32)
+
     def _adam_step_powers(self):
         o = self.optimizer
         if isinstance(o, Adam):
@@ -320,6 +329,10 @@ class CpuEmbeddingStore(EmbeddingStoreBase):
         occ = np.nonzero(self.keys != EMPTY_KEY)[0]
         signs = hashing.splitmix64_inv(self.keys[occ])
         inner = self.arena[torch.from_numpy(occ)].numpy()
+        sp = self._spill_export()
+        if sp is not None:
+            signs = np.concatenate([signs, sp[0]])
+            inner = np.concatenate([inner, sp[1]])
         return signs, inner
 
     def import_rows(self, signs: np.ndarray, inner: np.ndarray) -> None:
@@ -379,15 +392,38 @@ class HipEmbeddingStore(EmbeddingStoreBase):
         )
 
     def _drain_evictions(self, ev):
+        """Deferred: stash this batch's eviction log with an async count
+        readback; the NEXT lookup drains it (avoids a device sync per batch —
+        an evicted key re-looked-up within that one-batch window re-inits,
+        which bounded-window LRU makes a cold-key non-event)."""
         keys_t, count_t, rows_t = ev
         if keys_t.numel() == 0:
             return
-        cnt = int(count_t.item())  # sync (spill path only)
+        self._drain_pending()
+        cnt_pin = torch.zeros(1, dtype=torch.int32, pin_memory=True)
+        cnt_pin.copy_(count_t, non_blocking=True)
+        done = torch.cuda.Event()
+        done.record()
+        self._pending_evict = (keys_t, rows_t, cnt_pin, done)
+
+    def _drain_pending(self):
+        pend = getattr(self, "_pending_evict", None)
+        if pend is None:
+            return
+        self._pending_evict = None
+        keys_t, rows_t, cnt_pin, done = pend
+        done.synchronize()
+        cnt = int(cnt_pin.item())
         if cnt == 0:
             return
         keys_np = keys_t[:cnt].cpu().numpy().view(np.uint64)
         rows_np = rows_t[:cnt].cpu().numpy()
         self.spill.insert(keys_np, rows_np)
+
+    def flush_spill(self):
+        """Drain any stashed eviction log (checkpoint/export paths)."""
+        if self.spill is not None:
+            self._drain_pending()
 
     def _opt_params(self):
         o = self.optimizer
@@ -414,6 +450,9 @@ class HipEmbeddingStore(EmbeddingStoreBase):
         n = keys.numel()
         ev = self._no_evict
         if self.spill is not None and train and n:
+            self._drain_pending()
+            ev = self._evict_buffers(n)
+        if self.spill is not None and train and n and len(self.spill):
             # spill phase 1: restore missing-but-spilled rows into HBM first
             slots = self._C.store_probe(self.keys, self.ticks, keys, tick)
             miss_keys = keys[slots < 0]
@@ -421,6 +460,7 @@ class HipEmbeddingStore(EmbeddingStoreBase):
                 miss_np = miss_keys.cpu().numpy().view(np.uint64)
                 rows, found = self.spill.fetch(miss_np)
                 if found.any():
+                    # re-size the log: imports can evict too
                     ev = self._evict_buffers(n + int(found.sum()))
                     found_keys = torch.from_numpy(
                         miss_np[found].view(np.int64).copy()
@@ -430,8 +470,6 @@ class HipEmbeddingStore(EmbeddingStoreBase):
                         self.keys, self.ticks, self.arena, found_keys, rows_t,
                         tick, *ev,
                     )
-            if ev is self._no_evict:
-                ev = self._evict_buffers(n)
         out = torch.empty(n, self.dim, dtype=torch.float32, device=self.device)
         lo, hi = self.hyper.emb_initialization
         self._C.store_lookup(
@@ -479,10 +517,15 @@ class HipEmbeddingStore(EmbeddingStoreBase):
         return int(self._skipped[1].item())
 
     def export_rows(self) -> Tuple[np.ndarray, np.ndarray]:
+        self.flush_spill()
         occ = torch.nonzero(self.keys != 0, as_tuple=False).view(-1)
         keys = self.keys[occ].cpu().numpy().view(np.uint64)
         signs = hashing.splitmix64_inv(keys)
         inner = self.arena[occ].cpu().numpy()
+        sp = self._spill_export()
+        if sp is not None:
+            signs = np.concatenate([signs, sp[0]])
+            inner = np.concatenate([inner, sp[1]])
         return signs, inner
 
     def import_rows(self, signs: np.ndarray, inner: np.ndarray) -> None:

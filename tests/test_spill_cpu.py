@@ -53,6 +53,19 @@ def test_evicted_row_survives_in_host_tier():
     )
 
 
+def test_export_includes_spilled_rows():
+    store = CpuEmbeddingStore(
+        2, BUCKET_SIZE * PROBE_BUCKETS, Adagrad(lr=0.1), EmbeddingConfig(),
+        spill_capacity=1 << 14,
+    )
+    for s in range(0, 400, 8):
+        store.lookup(_keys(list(range(s, s + 8))), train=True)
+    assert len(store.spill) > 0
+    signs, inner = store.export_rows()
+    assert len(signs) == len(store) + len(store.spill)
+    assert inner.shape[1] == store.row_width
+
+
 def test_spill_capacity_bounded():
     store = CpuEmbeddingStore(
         2, BUCKET_SIZE * PROBE_BUCKETS, Adagrad(lr=0.1), EmbeddingConfig(),
