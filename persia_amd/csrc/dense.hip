@@ -193,25 +193,6 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
 constexpr int WTM = 64;       // m chunk per stage (2 MFMA k-steps per barrier)
 constexpr int WLD = WTM + PAD;
 
-// In-register 8x8 bf16 transpose across the 8-lane-stride partner group
-// (lanes l, l^8, l^16, ... of a wave): butterfly with static register
-// indices — v_new[k] = ((k^a)&s) ? shfl_xor(v[k^s], 8s) : v[k], s=1,2,4,
-// where a = (lane>>3)&7.  Turns the transposed-staging scalar ds_write
-// storm into one b128 write per lane (validated against a numpy model).
-__device__ __forceinline__ void pa_xpose8(bf16x8& x, int lane) {
-  const int a = (lane >> 3) & 7;
-#pragma unroll
-  for (int s = 1; s <= 4; s <<= 1) {
-    bf16x8 nx;
-#pragma unroll
-    for (int k = 0; k < 8; ++k) {
-      const int other = __shfl_xor((int)(unsigned short)x[k ^ s], s * 8);
-      nx[k] = (short)(((k ^ a) & s) ? other : (int)(unsigned short)x[k]);
-    }
-    x = nx;
-  }
-}
-
 __global__ __launch_bounds__(256) void wgrad_kernel(
     const short* __restrict__ dC, const short* __restrict__ A,
     float* __restrict__ dW, int M, int N, int K, int splitm) {
@@ -254,16 +235,13 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
     }
   };
   auto wwrite = [&](int buf, bf16x8 (&v)[2], bf16x8 (&w)[2]) {
-    // transpose in registers (8x8 butterfly over the lane-stride-8 group),
-    // then ONE b128 write per image per half
-    const int c = 8 * (lane & 7) + ((lane >> 3) & 7);
-    const int m_base = wave * 8;
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
-      pa_xpose8(v[h], lane);
-      *(bf16x8*)&lds_dct[buf][c * WLD + m_base + h * 32] = v[h];
-      pa_xpose8(w[h], lane);
-      *(bf16x8*)&lds_at[buf][c * WLD + m_base + h * 32] = w[h];
+      const int m = s_m + h * 32;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) lds_dct[buf][(s_c8 + j) * WLD + m] = v[h][j];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) lds_at[buf][(s_c8 + j) * WLD + m] = w[h][j];
     }
   };
 
