@@ -180,7 +180,15 @@ class PersiaTrainingBatch:
     PersiaTrainingBatch, persia-core/src/forward.rs:256-331)."""
 
     def __init__(self):
-        self.payloads: List[SlotPayload] = []
+        self._payloads: List[SlotPayload] = []
+        # fast-path groups whose SlotPayloads have not been built yet:
+        # (group, slot_names, B).  The flagship loop reads group.sum_base
+        # directly (enable_training_views / bench graph path), so building
+        # ~2*n_slots python objects per batch in the pipeline thread is
+        # wasted work unless someone actually asks for .payloads.
+        self._lazy_sum_groups: List[Tuple["_GroupCtx", List[str], int]] = []
+        self._order: Optional[Dict[str, int]] = None
+        self._sorted: bool = False
         self.non_id_type_tensors: List[torch.Tensor] = []
         self.label_tensors: List[torch.Tensor] = []
         self.batch_size: int = 0
@@ -189,6 +197,26 @@ class PersiaTrainingBatch:
         self.batch_id: Optional[int] = None
         self._groups: List[_GroupCtx] = []
         self._engine: Optional["EmbeddingEngine"] = None
+
+    @property
+    def payloads(self) -> List[SlotPayload]:
+        """Per-slot payloads in the original id_type_features order
+        (materialized on first access)."""
+        if self._lazy_sum_groups:
+            for group, _names, B in self._lazy_sum_groups:
+                sums = group.sum_base
+                for i, sc in enumerate(group.slots):
+                    self._payloads.append(
+                        SlotPayload(name=sc.name, cfg=sc.cfg,
+                                    sum_tensor=sums[i * B : (i + 1) * B])
+                    )
+            self._lazy_sum_groups.clear()
+            self._sorted = False
+        if not self._sorted:
+            if self._order is not None:
+                self._payloads.sort(key=lambda p: self._order[p.name])
+            self._sorted = True
+        return self._payloads
 
     def enable_training_views(self) -> Dict[str, torch.Tensor]:
         """Mark each group's sum base requires_grad and return fresh per-slot
@@ -230,7 +258,10 @@ class PersiaTrainingBatch:
                 for t in (sc.seg_offsets, sc.slot_uniq_global, sc.slot_inverse):
                     if t is not None and t.is_cuda:
                         yield t
-        for p in self.payloads:
+        # lazy groups hold only sum payloads (no per-payload device tensors),
+        # so iterating the eagerly-built list is sufficient and avoids
+        # materializing payloads in the pipeline thread
+        for p in self._payloads:
             for t in (p.raw_distinct, p.raw_index, p.raw_non_empty_index,
                       p.raw_sample_id_num):
                 if t is not None and t.is_cuda:
@@ -372,9 +403,10 @@ class EmbeddingEngine:
         for dim, feats in feats_by_dim.items():
             group = self._process_group(dim, feats, out, train, src_batch=batch)
             out._groups.append(group)
-        # keep payloads in the original id_type_features order
-        order = {f.name: i for i, f in enumerate(batch.id_type_features)}
-        out.payloads.sort(key=lambda p: order[p.name])
+        # keep payloads in the original id_type_features order (applied
+        # lazily when .payloads is first materialized)
+        out._order = {f.name: i for i, f in enumerate(batch.id_type_features)}
+        out._sorted = False
         return out
 
     def _upload_values(self, dim: int, values_np: np.ndarray) -> torch.Tensor:
@@ -461,6 +493,10 @@ class EmbeddingEngine:
                 seg_id=plan.seg_id, n_sum_slots=plan.S,
             )
             group.sum_base = sums
+            # defer SlotPayload construction (consumers on the fused path read
+            # group.sum_base directly; ~2*n_slots python objects per batch)
+            out._lazy_sum_groups.append((group, [sc.name for sc in slot_ctxs], B))
+            return group
         else:
             keys_t = C.sign_prep(vals_t, plan.slot_starts, plan.prefixes, spacing_arg)
             uniq_keys, inverse, perm, ustarts = _dedup(keys_t)
@@ -475,7 +511,7 @@ class EmbeddingEngine:
             )
             group.sum_base = sums
         for i, sc in enumerate(slot_ctxs):
-            out.payloads.append(
+            out._payloads.append(
                 SlotPayload(name=sc.name, cfg=sc.cfg, sum_tensor=sums[i * B : (i + 1) * B])
             )
         return group
@@ -599,7 +635,7 @@ class EmbeddingEngine:
             for sc in sum_slots:
                 nseg = sc.seg_offsets.numel() - 1
                 p = SlotPayload(name=sc.name, cfg=sc.cfg, sum_tensor=sums[b0 : b0 + nseg])
-                out.payloads.append(p)
+                out._payloads.append(p)
                 b0 += nseg
 
         # ---- raw slots (torch path both backends: not in the flagship loop)
@@ -619,7 +655,7 @@ class EmbeddingEngine:
                 rows[slot_uniq], slot_inv, sc.seg_offsets, sc.cfg.sample_fixed_size,
                 scale, torch.float16,
             )
-            out.payloads.append(
+            out._payloads.append(
                 SlotPayload(
                     name=sc.name, cfg=sc.cfg, raw_distinct=distinct, raw_index=index,
                     raw_non_empty_index=non_empty, raw_sample_id_num=num,
