@@ -325,6 +325,62 @@ __global__ void sign_prep_kernel(const ull* __restrict__ values,
   out[i] = k;
 }
 
+// sign prep with hashstack expansion (reference
+// indices_to_hashstack_indices, embedding_worker_service/mod.rs:348-400):
+// each raw id of a slot with hash_stack_rounds=R becomes R bucketed ids
+// (round r: iterate splitmix64, fold into [r*size,(r+1)*size)), laid out
+// position-major per sample ([r0 ids..., r1 ids...] inside each sample's
+// span) to match the CPU path bit-for-bit.  One thread per INPUT id writes
+// its R outputs; slots with R==0 pass through the plain prefix+mix path.
+__global__ void sign_prep_stack_kernel(
+    const ull* __restrict__ values, const int64_t* __restrict__ in_starts,
+    const int64_t* __restrict__ out_starts, const ull* __restrict__ prefixes,
+    const int* __restrict__ rounds, const int64_t* __restrict__ sizes,
+    const int64_t* __restrict__ off_starts,
+    const int64_t* __restrict__ all_offs,  // EXPANDED per-slot seg offsets
+    int n_slots, ull spacing, ull* __restrict__ out, int64_t n_in) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n_in) return;
+  int lo = 0, hi = n_slots;  // find slot: in_starts[s] <= i < in_starts[s+1]
+  while (hi - lo > 1) {
+    const int mid = (lo + hi) >> 1;
+    if (i >= in_starts[mid]) lo = mid; else hi = mid;
+  }
+  const int s = lo;
+  const ull prefix = prefixes[s];
+  const int R = rounds[s];
+  if (R == 0) {
+    ull sign = values[i];
+    if (prefix != 0) sign = sign % spacing + prefix;
+    ull k = pa_splitmix64(sign);
+    if (k == PA_EMPTY_KEY) k = 0xD1B54A32D192ED03ull;
+    out[out_starts[s] + (i - in_starts[s])] = k;
+    return;
+  }
+  const int64_t j = i - in_starts[s];  // slot-local input position
+  // sample b: expanded offsets E[b] = off[b]*R, so E[b] <= j*R < E[b+1]
+  const int64_t* E = all_offs + off_starts[s];
+  const int64_t nE = off_starts[s + 1] - off_starts[s];
+  const int64_t key = j * (int64_t)R;
+  int64_t blo = 0, bhi = nE - 1;
+  while (bhi - blo > 1) {
+    const int64_t mid = (blo + bhi) >> 1;
+    if (key >= E[mid]) blo = mid; else bhi = mid;
+  }
+  const int64_t seg_lo = E[blo] / R, seg_hi = E[blo + 1] / R;
+  const int64_t base = out_starts[s] + E[blo];
+  const int64_t size = sizes[s];
+  ull h = values[i];
+  for (int r = 0; r < R; ++r) {
+    h = pa_splitmix64(h);
+    ull sign = h % (ull)size + (ull)((int64_t)r * size);
+    if (prefix != 0) sign = sign % spacing + prefix;
+    ull k = pa_splitmix64(sign);
+    if (k == PA_EMPTY_KEY) k = 0xD1B54A32D192ED03ull;
+    out[base + (int64_t)r * (seg_hi - seg_lo) + (j - seg_lo)] = k;
+  }
+}
+
 // --------------------------------------------------------------- import rows
 
 __global__ void import_kernel(ull* __restrict__ table_keys,

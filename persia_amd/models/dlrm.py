@@ -89,10 +89,14 @@ class DLRM(nn.Module):
     def forward(
         self, non_id_tensors, embedding_tensors
     ) -> torch.Tensor:
-        if torch.is_tensor(non_id_tensors):
-            dense = non_id_tensors.float()
-        else:
-            dense = non_id_tensors[0].float()
+        dense = (
+            non_id_tensors if torch.is_tensor(non_id_tensors) else non_id_tensors[0]
+        )
+        # follow the tower's weight dtype: f32 normally (autocast inserts the
+        # bf16 casts), bf16 when the whole model runs in bf16 with f32 master
+        # weights (bench graph path — no autocast, no per-layer casts)
+        dt = next(self.bottom.parameters()).dtype
+        dense = dense.to(dt)
         x = self.bottom(dense)  # [B, D]
         if torch.is_tensor(embedding_tensors):
             # packed slot-major [S*B, D] (the engine's fused sum output) —
