@@ -195,26 +195,46 @@ def main():
             # weight on every replay (~40 elementwise kernels/step).  Keeping
             # the weights bf16 gives the IDENTICAL forward compute (autocast
             # rounds f32 weights to bf16 per use anyway) while the f32 master
-            # copy preserves full-precision SGD accumulation — 3 multi-tensor
+            # copy preserves full-precision SGD accumulation — 4 multi-tensor
             # launches replace the per-layer cast storm.
-            model.bfloat16()
-            g_params = list(model.parameters())
-            g_masters = [p.detach().clone().float() for p in g_params]
-            lr = opt.param_groups[0]["lr"]
-            for p in g_params:
-                p.grad = torch.zeros_like(p)
+            # PA_GRAPH_BF16=0 selects the autocast variant (A/B switch).
+            bf16_weights = os.environ.get("PA_GRAPH_BF16", "1") == "1"
+            if bf16_weights:
+                model.bfloat16()
+                g_params = list(model.parameters())
+                g_masters = [p.detach().clone().float() for p in g_params]
+                g_grads32 = [torch.zeros_like(m) for m in g_masters]
+                lr = opt.param_groups[0]["lr"]
+                for p in g_params:
+                    p.grad = torch.zeros_like(p)
 
-            def iteration():
-                static["base"].grad.zero_()
-                logits = model(static["dense"], static["base"])
-                loss = loss_fn(logits.float(), static["label"])
-                loss.backward()
-                grads = [p.grad for p in g_params]
-                torch._foreach_add_(g_masters, grads, alpha=-lr)
-                with torch.no_grad():
-                    torch._foreach_copy_(g_params, g_masters)
-                torch._foreach_zero_(grads)
-                return loss
+                def iteration():
+                    static["base"].grad.zero_()
+                    logits = model(static["dense"], static["base"])
+                    loss = loss_fn(logits.float(), static["label"])
+                    loss.backward()
+                    grads = [p.grad for p in g_params]
+                    # same-dtype foreach lists (mixed-dtype falls off the
+                    # multi-tensor fast path): cast-copy, f32 SGD, write back
+                    torch._foreach_copy_(g_grads32, grads)
+                    torch._foreach_add_(g_masters, g_grads32, alpha=-lr)
+                    with torch.no_grad():
+                        torch._foreach_copy_(g_params, g_masters)
+                    torch._foreach_zero_(grads)
+                    return loss
+
+            else:
+
+                def iteration():
+                    static["base"].grad.zero_()
+                    with torch.autocast("cuda", dtype=amp_dtype, cache_enabled=False):
+                        logits = model(static["dense"], static["base"])
+                        loss = loss_fn(logits.float(), static["label"])
+                    loss.backward()
+                    opt.step()
+                    grads = [p.grad for p in model.parameters() if p.grad is not None]
+                    torch._foreach_zero_(grads)
+                    return loss
 
             # warmup on a side stream (allocator state, autotuned GEMMs)
             static["base"].grad = torch.zeros_like(static["base"], dtype=torch.float16)

@@ -535,25 +535,38 @@ class EmbeddingEngine:
         feats = sorted(
             feats, key=lambda f: 0 if self.schema.get_slot(f.name).embedding_summation else 1
         )
-        gpu_prep = native and not any(
-            self.schema.get_slot(f.name).hash_stack_rounds > 0 for f in feats
-        )
+        gpu_prep = native
         key_arrays = []
         offset_arrays = []
         slot_ctxs: List[_SlotCtx] = []
-        pos = 0
+        in_lens: List[int] = []      # uploaded (pre-expansion) lengths
+        rounds_list: List[int] = []  # hashstack rounds per slot (GPU prep)
+        sizes_list: List[int] = []
+        pos = 0  # OUTPUT (post-hashstack-expansion) position space
         for feat in feats:
             cfg = self.schema.get_slot(feat.name)
+            r = cfg.hash_stack_rounds
             if gpu_prep:
-                keys, offsets = feat.values, feat.offsets  # raw; mixed on-GPU
+                # raw ids; hashstack expansion + prefix + mix all on-GPU.
+                # expanded seg offsets are exactly offsets*r (every segment
+                # grows by the same factor)
+                keys = feat.values
+                offsets = feat.offsets * r if r > 0 else feat.offsets
+                out_len = len(keys) * max(r, 1)
             else:
                 keys, offsets = self._prepare_slot_keys(feat)
+                out_len = len(keys)
             key_arrays.append(keys)
             offset_arrays.append(offsets)
-            sc = _SlotCtx(name=feat.name, cfg=cfg, pos_slice=(pos, pos + len(keys)),
+            in_lens.append(len(keys))
+            rounds_list.append(r if gpu_prep else 0)
+            sizes_list.append(
+                cfg.hash_stack_config.embedding_size if r > 0 else 1
+            )
+            sc = _SlotCtx(name=feat.name, cfg=cfg, pos_slice=(pos, pos + out_len),
                           seg_offsets=None)
             slot_ctxs.append(sc)
-            pos += len(keys)
+            pos += out_len
 
         all_keys = np.concatenate(key_arrays) if key_arrays else np.empty(0, np.uint64)
         # single H2D upload of the whole group's values + all offsets
@@ -567,7 +580,7 @@ class EmbeddingEngine:
             sc.seg_offsets = all_offs_t[o0 : o0 + len(offs)]
             o0 += len(offs)
         if gpu_prep:
-            slot_starts = torch.tensor(
+            out_starts = torch.tensor(
                 [sc.pos_slice[0] for sc in slot_ctxs] + [pos],
                 dtype=torch.int64, device=dev,
             )
@@ -576,12 +589,28 @@ class EmbeddingEngine:
                 .view(np.int64)
             ).to(dev)
             spacing = self.schema.feature_spacing
+            spacing_arg = spacing if spacing < (1 << 63) else -1
             from persia_amd.ops import native as _native2
 
-            keys_t = _native2().sign_prep(
-                keys_t, slot_starts, prefixes,
-                spacing if spacing < (1 << 63) else -1,
-            )
+            if any(rounds_list):
+                in_starts = torch.from_numpy(
+                    np.concatenate([[0], np.cumsum(in_lens)]).astype(np.int64)
+                ).to(dev)
+                off_starts = torch.from_numpy(
+                    np.concatenate(
+                        [[0], np.cumsum([len(o) for o in offset_arrays])]
+                    ).astype(np.int64)
+                ).to(dev)
+                keys_t = _native2().sign_prep_stack(
+                    keys_t, in_starts, out_starts, prefixes,
+                    torch.tensor(rounds_list, dtype=torch.int32, device=dev),
+                    torch.tensor(sizes_list, dtype=torch.int64, device=dev),
+                    off_starts, all_offs_t, spacing_arg, pos,
+                )
+            else:
+                keys_t = _native2().sign_prep(
+                    keys_t, out_starts, prefixes, spacing_arg
+                )
 
         uniq_keys, inverse, perm, ustarts = _dedup(keys_t)
         group = _GroupCtx(

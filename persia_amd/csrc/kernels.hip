@@ -830,6 +830,28 @@ torch::Tensor sign_prep(torch::Tensor values, torch::Tensor slot_starts,
   return out;
 }
 
+torch::Tensor sign_prep_stack(torch::Tensor values, torch::Tensor in_starts,
+                              torch::Tensor out_starts, torch::Tensor prefixes,
+                              torch::Tensor rounds, torch::Tensor sizes,
+                              torch::Tensor off_starts, torch::Tensor all_offs,
+                              int64_t spacing, int64_t n_out) {
+  const int64_t n_in = values.numel();
+  auto out = torch::empty({n_out}, values.options());
+  if (n_in == 0) return out;
+  hipLaunchKernelGGL(sign_prep_stack_kernel, dim3(n_blocks_for(n_in, 256)),
+                     dim3(256), 0, cur_stream(),
+                     (const ull*)values.data_ptr<int64_t>(),
+                     in_starts.data_ptr<int64_t>(),
+                     out_starts.data_ptr<int64_t>(),
+                     (const ull*)prefixes.data_ptr<int64_t>(),
+                     rounds.data_ptr<int32_t>(), sizes.data_ptr<int64_t>(),
+                     off_starts.data_ptr<int64_t>(),
+                     all_offs.data_ptr<int64_t>(),
+                     (int)(in_starts.numel() - 1), (ull)spacing,
+                     (ull*)out.data_ptr<int64_t>(), n_in);
+  return out;
+}
+
 void init_dense(pybind11::module_& m);     // csrc/dense.hip
 void init_engine(pybind11::module_& m);    // csrc/engine.cpp
 void init_interact(pybind11::module_& m);  // csrc/interact.hip
@@ -846,6 +868,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("segment_sum", &segment_sum, "fused gather + per-sample summation");
   m.def("grad_scatter", &grad_scatter, "ordered per-sign gradient scatter");
   m.def("sign_prep", &sign_prep, "prefix-fold + splitmix64 key mixing");
+  m.def("sign_prep_stack", &sign_prep_stack,
+        "hashstack expansion + prefix-fold + splitmix64 key mixing");
   m.def("scatter_update", &scatter_update,
         "fused ordered grad scatter + optimizer update (no [U,dim] buffer)");
 }
