@@ -240,6 +240,60 @@ def test_sign_prep_matches_numpy():
     assert np.array_equal(got, expected)
 
 
+def test_sign_prep_stack_matches_cpu():
+    """GPU hashstack expansion (sign_prep_stack) is bitwise equal to the CPU
+    prep path (hash_stack + apply_prefix + splitmix64, position-major
+    interleave per sample)."""
+    from persia_amd.ops import native
+
+    C = native()
+    rng = np.random.default_rng(11)
+    spacing = (1 << 56) - 1
+    # slot 0: plain (rounds=0); slot 1: rounds=3 stack, ragged segments
+    v0 = rng.integers(0, 2 ** 62, size=40, dtype=np.uint64)
+    lens = rng.integers(0, 5, size=16)
+    v1 = rng.integers(0, 2 ** 62, size=int(lens.sum()), dtype=np.uint64)
+    offs1 = np.zeros(17, dtype=np.int64)
+    np.cumsum(lens, out=offs1[1:])
+    R, SIZE = 3, 97
+    p0, p1 = np.uint64(1 << 56), np.uint64(2 << 56)
+
+    # CPU expectation, exactly as engine._prepare_slot_keys builds it
+    exp0 = hashing.splitmix64(hashing.apply_prefix(v0, int(p0), spacing))
+    exp0[exp0 == 0] = np.uint64(0xD1B54A32D192ED03)
+    stacked = hashing.hash_stack(v1, R, SIZE)  # (R, nnz)
+    new_vals = np.empty(len(v1) * R, dtype=np.uint64)
+    new_offs = offs1 * R
+    for r in range(R):
+        for b in range(16):
+            s, e = offs1[b], offs1[b + 1]
+            dst = new_offs[b] + r * (e - s)
+            new_vals[dst : dst + (e - s)] = stacked[r, s:e]
+    exp1 = hashing.splitmix64(hashing.apply_prefix(new_vals, int(p1), spacing))
+    exp1[exp1 == 0] = np.uint64(0xD1B54A32D192ED03)
+    expected = np.concatenate([exp0, exp1])
+
+    dev = _dev()
+    vals = torch.from_numpy(np.concatenate([v0, v1]).view(np.int64)).to(dev)
+    in_starts = torch.tensor([0, 40, 40 + len(v1)], dtype=torch.int64, device=dev)
+    out_starts = torch.tensor(
+        [0, 40, 40 + len(v1) * R], dtype=torch.int64, device=dev
+    )
+    pref_t = torch.from_numpy(np.array([p0, p1]).view(np.int64)).to(dev)
+    # concatenated per-slot EXPANDED seg offsets (slot 0 has a trivial one)
+    offs0 = np.array([0, 40], dtype=np.int64)
+    all_offs = torch.from_numpy(np.concatenate([offs0, new_offs])).to(dev)
+    off_starts = torch.tensor([0, 2, 2 + 17], dtype=torch.int64, device=dev)
+    out = C.sign_prep_stack(
+        vals, in_starts, out_starts, pref_t,
+        torch.tensor([0, R], dtype=torch.int32, device=dev),
+        torch.tensor([1, SIZE], dtype=torch.int64, device=dev),
+        off_starts, all_offs, spacing, 40 + len(v1) * R,
+    )
+    got = out.cpu().numpy().view(np.uint64)
+    assert np.array_equal(got, expected)
+
+
 def test_engine_gpu_matches_cpu_engine():
     from persia_amd.core.comm import DistContext
     from persia_amd.core.engine import EmbeddingEngine
