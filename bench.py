@@ -335,10 +335,13 @@ def main():
         import sys as _s
 
         n = tstats["n"]
+        prod = (
+            pipeline.prod_s / pipeline.prod_n * 1e3 if pipeline.prod_n else 0.0
+        )
         print(
             f"# phase ms/step: get={tstats['get']/n*1e3:.3f} "
             f"copy={tstats['copy']/n*1e3:.3f} replay={tstats['replay']/n*1e3:.3f} "
-            f"apply={tstats['apply']/n*1e3:.3f}",
+            f"apply={tstats['apply']/n*1e3:.3f} producer={prod:.3f}",
             file=_s.stderr, flush=True,
         )
     samples = args.steps * B * world

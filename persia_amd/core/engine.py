@@ -26,6 +26,7 @@ Per batch (one dim-group):
 """
 import queue
 import threading
+import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Tuple
 
@@ -1006,6 +1007,8 @@ class ForwardPipeline:
         self._stop = threading.Event()
         self._exc: Optional[BaseException] = None
         self.inflight = 0  # lookups ahead of their gradient push (staleness)
+        self.prod_s = 0.0  # cumulative producer-thread process_batch seconds
+        self.prod_n = 0
 
     def start(self):
         if self._thread is None:
@@ -1053,6 +1056,7 @@ class ForwardPipeline:
             if self.engine.metrics_enabled:
                 self.engine.metrics.staleness.set(self.inflight)
             try:
+                t0 = time.perf_counter()
                 if stream is not None:
                     with torch.cuda.stream(stream):
                         tb = self.engine.process_batch(item)
@@ -1061,6 +1065,10 @@ class ForwardPipeline:
                     tb._ready_event = ev
                 else:
                     tb = self.engine.process_batch(item)
+                # host-side producer cost (kernel issue + any CPU work);
+                # excludes GPU completion — see `prod_n`/`prod_s` for the avg
+                self.prod_s += time.perf_counter() - t0
+                self.prod_n += 1
                 self._out.put(tb)
             except BaseException as e:  # propagate to consumer
                 self._exc = e

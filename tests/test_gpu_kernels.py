@@ -244,6 +244,7 @@ def test_sign_prep_stack_matches_cpu():
     """GPU hashstack expansion (sign_prep_stack) is bitwise equal to the CPU
     prep path (hash_stack + apply_prefix + splitmix64, position-major
     interleave per sample)."""
+    from persia_amd.core import hashing
     from persia_amd.ops import native
 
     C = native()
