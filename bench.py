@@ -344,6 +344,14 @@ def main():
             f"apply={tstats['apply']/n*1e3:.3f} producer={prod:.3f}",
             file=_s.stderr, flush=True,
         )
+        pt = engine._pt
+        if pt["n"]:
+            pn = pt["n"]
+            print(
+                f"# producer ms/batch: total={pt['batch']/pn*1e3:.3f} "
+                f"prep={pt['prep']/pn*1e3:.3f} native={pt['native']/pn*1e3:.3f}",
+                file=_s.stderr, flush=True,
+            )
     samples = args.steps * B * world
     if rank == 0:
         result = {

@@ -118,10 +118,12 @@ class _SlotCtx:
 @dataclass
 class _GroupCtx:
     dim: int
-    uniq_keys: torch.Tensor  # [U] int64 (u64 bit pattern)
+    uniq_keys: torch.Tensor  # [U] int64 (u64 bit pattern); fast path: [nnz]
+    #   padded with the true count ONLY in u_count (device) — sync-free dedup
     inverse: torch.Tensor  # [nnz] i64
     perm: torch.Tensor  # [nnz] sort permutation (ordered grad scatter)
     ustarts: torch.Tensor  # [U+1] unique boundaries in sorted order
+    u_count: Optional[torch.Tensor] = None  # [1] i64 device unique count
     slots: List[_SlotCtx] = field(default_factory=list)
     # fused sum-slot segment space (sum slots occupy positions [0, sum_end))
     cat_offsets: Optional[torch.Tensor] = None  # [n_sum_segs+1]
@@ -251,8 +253,9 @@ class PersiaTrainingBatch:
             if t.is_cuda:
                 yield t
         for g in self._groups:
-            for t in (g.uniq_keys, g.inverse, g.perm, g.ustarts, g.sum_base,
-                      g.cat_offsets, g.seg_id, g.seg_lens, g.sqrt_mask):
+            for t in (g.uniq_keys, g.inverse, g.perm, g.ustarts, g.u_count,
+                      g.sum_base, g.cat_offsets, g.seg_id, g.seg_lens,
+                      g.sqrt_mask):
                 if t is not None and t.is_cuda:
                     yield t
             for sc in g.slots:
@@ -314,10 +317,15 @@ class EmbeddingEngine:
         self.nan_grad_batches = 0
         self._plans = {}
         self._empty_scale = None
+        self._empty_count = None
         self._pin_rings: Dict[int, list] = {}
         self._pin_idx: Dict[int, int] = {}
         self.model_manager_status = "Idle"
         self.model_manager_progress = 0.0
+        import os as _os
+
+        self._prod_timing = _os.environ.get("PA_PROD_TIMING", "0") == "1"
+        self._pt = {"prep": 0.0, "native": 0.0, "batch": 0.0, "n": 0}
         from persia_amd.core.metrics import EngineMetrics
 
         self.metrics_enabled = bool(gconf.enable_metrics)
@@ -327,6 +335,26 @@ class EmbeddingEngine:
         if self._empty_scale is None:
             self._empty_scale = torch.empty(0, dtype=torch.float32, device=self.device)
         return self._empty_scale
+
+    def _empty_i64(self) -> torch.Tensor:
+        if self._empty_count is None:
+            self._empty_count = torch.empty(0, dtype=torch.int64, device=self.device)
+        return self._empty_count
+
+    @staticmethod
+    def _materialize_group(group: "_GroupCtx") -> "_GroupCtx":
+        """Resolve a padded sync-free dedup group (fast path: uniq/ustarts are
+        nnz-padded, true count on-device) to exact host shapes.  Costs one
+        stream sync — only rare paths (raw grads, f32/large-dim fallback,
+        skipped-grad groups) need it; the fused kernels read the count from
+        the device."""
+        if group.u_count is None:
+            return group
+        U = int(group.u_count.item())
+        group.uniq_keys = group.uniq_keys.narrow(0, 0, U)
+        group.ustarts = group.ustarts.narrow(0, 0, U + 1)
+        group.u_count = None
+        return group
 
     # ------------------------------------------------------------- sign prep
 
@@ -380,6 +408,8 @@ class EmbeddingEngine:
     # ---------------------------------------------------------- forward path
 
     def process_batch(self, batch: PersiaBatch, train: Optional[bool] = None) -> PersiaTrainingBatch:
+        if self._prod_timing:
+            _tb0 = time.perf_counter()
         train = batch.requires_grad if train is None else train
         if self.gconf.job_type == "infer":
             train = False
@@ -408,6 +438,9 @@ class EmbeddingEngine:
         # lazily when .payloads is first materialized)
         out._order = {f.name: i for i, f in enumerate(batch.id_type_features)}
         out._sorted = False
+        if self._prod_timing:
+            self._pt["batch"] += time.perf_counter() - _tb0
+            self._pt["n"] += 1
         return out
 
     def _upload_values(self, dim: int, values_np: np.ndarray) -> torch.Tensor:
@@ -438,6 +471,8 @@ class EmbeddingEngine:
         upload, ~15 kernel launches, static device-side plan."""
         from persia_amd.ops import native as _native
 
+        if self._prod_timing:
+            _t0 = time.perf_counter()
         C = _native()
         dev = self.device
         B = len(feats[0].values)
@@ -479,8 +514,11 @@ class EmbeddingEngine:
         if not self.dist.distributed and store.spill is None:
             # whole lookup in ONE native call (C++ drives sign prep, dedup,
             # probe/insert, gather and the fused segment-sum)
+            if self._prod_timing:
+                self._pt["prep"] += time.perf_counter() - _t0
+                _t1 = time.perf_counter()
             lo, hi = self.hyper.emb_initialization
-            sums, uniq_keys, inverse, perm, ustarts = C.lookup_local(
+            sums, uniq_keys, inverse, perm, ustarts, u_count = C.lookup_local(
                 vals_t, plan.slot_starts, plan.prefixes, spacing_arg,
                 plan.cat_offsets, plan.empty_scale,
                 store.keys, store.ticks, store.arena, dim,
@@ -490,13 +528,16 @@ class EmbeddingEngine:
             )
             group = _GroupCtx(
                 dim=dim, uniq_keys=uniq_keys, inverse=inverse, perm=perm,
-                ustarts=ustarts, slots=slot_ctxs, cat_offsets=plan.cat_offsets,
-                seg_id=plan.seg_id, n_sum_slots=plan.S,
+                ustarts=ustarts, u_count=u_count, slots=slot_ctxs,
+                cat_offsets=plan.cat_offsets, seg_id=plan.seg_id,
+                n_sum_slots=plan.S,
             )
             group.sum_base = sums
             # defer SlotPayload construction (consumers on the fused path read
             # group.sum_base directly; ~2*n_slots python objects per batch)
             out._lazy_sum_groups.append((group, [sc.name for sc in slot_ctxs], B))
+            if self._prod_timing:
+                self._pt["native"] += time.perf_counter() - _t1
             return group
         else:
             keys_t = C.sign_prep(vals_t, plan.slot_starts, plan.prefixes, spacing_arg)
@@ -710,6 +751,7 @@ class EmbeddingEngine:
 
             C = _native()
         for group in training_batch._groups:
+            self._materialize_group(group)  # generic path needs exact U
             U = group.uniq_keys.numel()
             buf = torch.zeros(U, group.dim, dtype=torch.float32, device=self.device)
             any_grad = False
@@ -844,8 +886,7 @@ class EmbeddingEngine:
 
         C = _native()
         for gi, group in enumerate(training_batch._groups):
-            U = group.uniq_keys.numel()
-            buf = torch.empty(U, group.dim, dtype=torch.float32, device=self.device)
+            buf = None  # allocated lazily: the fused path never needs it
             if sum_base_grads is not None:
                 gbase = sum_base_grads[gi]
             else:
@@ -883,14 +924,26 @@ class EmbeddingEngine:
                         store._opt_code, store._opt_params(), float(b1p),
                         float(b2p), float(self.hyper.weight_bound),
                         store._skipped,
+                        group.u_count
+                        if group.u_count is not None
+                        else self._empty_i64(),
                     )
                     continue
+                self._materialize_group(group)
+                buf = torch.empty(
+                    group.uniq_keys.numel(), group.dim,
+                    dtype=torch.float32, device=self.device,
+                )
                 C.grad_scatter(
-                    gbase.contiguous(), group.perm, group.ustarts, group.seg_id,
-                    seg_scale, buf, 0,
+                    gbase.contiguous(), group.perm, group.ustarts,
+                    group.seg_id, seg_scale, buf, 0,
                 )
             else:
-                buf.zero_()
+                self._materialize_group(group)
+                buf = torch.zeros(
+                    group.uniq_keys.numel(), group.dim,
+                    dtype=torch.float32, device=self.device,
+                )
             # raw slots add on top (buf fully written above)
             if raw_grads:
                 for sc in group.slots:

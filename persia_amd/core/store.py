@@ -487,6 +487,7 @@ class HipEmbeddingStore(EmbeddingStoreBase):
             float(self.optimizer.state_init(self.dim)),
             self.opt_space,
             *ev,
+            torch.empty(0, dtype=torch.int64, device=self.device),
         )
         self._drain_evictions(ev)
         return out

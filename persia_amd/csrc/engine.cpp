@@ -15,7 +15,8 @@ void store_lookup(torch::Tensor table_keys, torch::Tensor ticks,
                   int64_t dim, int64_t train, int64_t tick, double lo,
                   double hi, double admit_prob, double state_init,
                   int64_t opt_space, torch::Tensor evict_keys,
-                  torch::Tensor evict_count, torch::Tensor evict_rows);
+                  torch::Tensor evict_count, torch::Tensor evict_rows,
+                  torch::Tensor u_count);
 void store_update(torch::Tensor table_keys, torch::Tensor ticks,
                   torch::Tensor arena, torch::Tensor query,
                   torch::Tensor grads, int64_t dim, int64_t opt,
@@ -36,7 +37,11 @@ void scatter_update(torch::Tensor table_keys, torch::Tensor ticks,
                     torch::Tensor seg_scale, int64_t dim, int64_t opt,
                     std::vector<double> params, double b1_power,
                     double b2_power, double weight_bound,
-                    torch::Tensor skipped);
+                    torch::Tensor skipped, torch::Tensor u_count);
+void dedup_finalize(torch::Tensor svals_flipped, torch::Tensor perm,
+                    torch::Tensor neq, torch::Tensor rank, int64_t flip,
+                    torch::Tensor inverse, torch::Tensor uniq,
+                    torch::Tensor ustarts, torch::Tensor u_count);
 
 static constexpr int64_t kFlip = std::numeric_limits<int64_t>::min();
 
@@ -75,21 +80,40 @@ std::vector<torch::Tensor> lookup_local(
     int64_t dim, int64_t train, int64_t tick, double lo, double hi,
     double admit_prob, double state_init, int64_t opt_space) {
   auto keys = sign_prep(values, slot_starts, prefixes, spacing);
-  auto d = dedup_keys(keys);
-  auto& uniq = d[0];
+  // fixed-shape dedup: every tensor is nnz-padded and the true unique count
+  // lives ONLY on the device (u_count) — no host synchronization anywhere in
+  // the lookup, so the pipeline thread issues the whole batch and returns
+  // (the old masked_select/nonzero dedup forced two stream syncs per batch,
+  // serializing the producer behind the GPU)
+  const int64_t nnz = keys.numel();
+  auto dev = values.device();
+  auto opts = torch::TensorOptions().dtype(torch::kInt64).device(dev);
+  constexpr int64_t kFlipLocal = std::numeric_limits<int64_t>::min();
+  auto flipped = keys.bitwise_xor(kFlipLocal);
+  auto sorted = flipped.sort();
+  auto svals = std::get<0>(sorted);
+  auto perm = std::get<1>(sorted);
+  auto neq = torch::ones({nnz}, opts.dtype(torch::kBool));
+  neq.slice(0, 1, nnz) =
+      svals.slice(0, 1, nnz).ne(svals.slice(0, 0, nnz - 1));
+  auto rank = neq.cumsum(0);
+  rank.sub_(1);
+  auto inverse = torch::empty({nnz}, opts);
+  auto uniq = torch::empty({nnz}, opts);
+  auto ustarts = torch::empty({nnz + 1}, opts);
+  auto u_count = torch::empty({1}, opts);
+  dedup_finalize(svals, perm, neq, rank, kFlipLocal, inverse, uniq, ustarts,
+                 u_count);
   auto rows = torch::empty(
-      {uniq.numel(), dim},
-      torch::TensorOptions().dtype(torch::kFloat32).device(values.device()));
-  auto none = torch::empty(
-      {0}, torch::TensorOptions().dtype(torch::kInt64).device(values.device()));
-  auto none_i32 = torch::empty(
-      {0}, torch::TensorOptions().dtype(torch::kInt32).device(values.device()));
-  auto none_f32 = torch::empty(
-      {0}, torch::TensorOptions().dtype(torch::kFloat32).device(values.device()));
+      {nnz, dim}, torch::TensorOptions().dtype(torch::kFloat32).device(dev));
+  auto none = torch::empty({0}, opts);
+  auto none_i32 = torch::empty({0}, opts.dtype(torch::kInt32));
+  auto none_f32 = torch::empty({0}, opts.dtype(torch::kFloat32));
   store_lookup(table_keys, ticks, arena, uniq, rows, dim, train, tick, lo, hi,
-               admit_prob, state_init, opt_space, none, none_i32, none_f32);
-  auto sums = segment_sum(rows, d[1], cat_offsets, seg_scale);
-  return {sums, uniq, d[1], d[2], d[3]};
+               admit_prob, state_init, opt_space, none, none_i32, none_f32,
+               u_count);
+  auto sums = segment_sum(rows, inverse, cat_offsets, seg_scale);
+  return {sums, uniq, inverse, perm, ustarts, u_count};
 }
 
 // Single-GPU fused backward: per-segment grads -> per-sign scatter -> fused
@@ -100,13 +124,21 @@ void update_local(torch::Tensor grads, torch::Tensor perm,
                   torch::Tensor table_keys, torch::Tensor ticks,
                   torch::Tensor arena, int64_t dim, int64_t opt,
                   std::vector<double> params, double b1_power, double b2_power,
-                  double weight_bound, torch::Tensor skipped) {
+                  double weight_bound, torch::Tensor skipped,
+                  torch::Tensor u_count) {
   if (grads.scalar_type() == torch::kFloat16 && dim <= 512) {
     // single fused kernel: ordered scatter + optimizer, no [U,dim] buffer
     scatter_update(table_keys, ticks, arena, uniq_keys, grads, perm, ustarts,
                    seg_id, seg_scale, dim, opt, params, b1_power, b2_power,
-                   weight_bound, skipped);
+                   weight_bound, skipped, u_count);
     return;
+  }
+  // rare fallback (f32 grads or dim > 512): needs exact shapes — for a
+  // padded (device-count) dedup, pay the one host sync and slice
+  if (u_count.numel()) {
+    const int64_t U = u_count.item<int64_t>();
+    uniq_keys = uniq_keys.narrow(0, 0, U);
+    ustarts = ustarts.narrow(0, 0, U + 1);
   }
   auto buf = torch::empty(
       {uniq_keys.numel(), dim},

@@ -125,7 +125,12 @@ __global__ void probe_claim_kernel(ull* __restrict__ table_keys,
                                    int* __restrict__ out_new,
                                    ull* __restrict__ evict_keys,
                                    int* __restrict__ evict_count,
-                                   long long* __restrict__ out_evict_idx) {
+                                   long long* __restrict__ out_evict_idx,
+                                   const long long* __restrict__ n_dev) {
+  // n_dev: device-side element count (padded-dedup path — the true unique
+  // count is produced on-GPU so the host never synchronizes); n is the
+  // grid-sizing upper bound
+  if (n_dev) n = *n_dev;
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   const ull k = query[i];
@@ -168,7 +173,9 @@ __global__ void init_gather_kernel(float* __restrict__ arena,
                                    int row_width, double lo, double hi,
                                    float state_init,
                                    const long long* __restrict__ evict_idx,
-                                   float* __restrict__ evict_rows) {
+                                   float* __restrict__ evict_rows,
+                                   const long long* __restrict__ n_dev) {
+  if (n_dev) n = *n_dev;
   const int wave = threadIdx.x / PA_WAVE;
   const int lane = threadIdx.x % PA_WAVE;
   const int waves_per_block = blockDim.x / PA_WAVE;
@@ -381,6 +388,31 @@ __global__ void sign_prep_stack_kernel(
   }
 }
 
+// Fixed-shape dedup epilogue: after sort + neq + rank=cumsum(neq)-1 (all
+// shape-static ATen ops), ONE pass finalizes inverse/uniq/ustarts into
+// nnz-padded buffers and writes the true unique count to a device scalar —
+// the host never learns U, so the whole lookup issues without a single
+// synchronization (the probe/gather/update kernels read U from n_dev).
+__global__ void dedup_finalize_kernel(
+    const ull* __restrict__ svals_flipped, const int64_t* __restrict__ perm,
+    const bool* __restrict__ neq, const int64_t* __restrict__ rank,
+    int64_t n, ull flip, int64_t* __restrict__ inverse,
+    ull* __restrict__ uniq, int64_t* __restrict__ ustarts,
+    long long* __restrict__ u_count) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const int64_t r = rank[i];
+  inverse[perm[i]] = r;
+  if (neq[i]) {
+    uniq[r] = svals_flipped[i] ^ flip;
+    ustarts[r] = i;
+  }
+  if (i == n - 1) {
+    u_count[0] = r + 1;
+    ustarts[r + 1] = n;
+  }
+}
+
 // --------------------------------------------------------------- import rows
 
 __global__ void import_kernel(ull* __restrict__ table_keys,
@@ -503,7 +535,8 @@ __global__ void scatter_update_kernel(
     const float* __restrict__ seg_scale, int64_t n, int dim, int row_width,
     int64_t n_buckets, int opt, float p0, float p1, float p2, float p3,
     float b1_power, float b2_power, float weight_bound,
-    int* __restrict__ skipped) {
+    int* __restrict__ skipped, const long long* __restrict__ n_dev) {
+  if (n_dev) n = *n_dev;
   const int wave = threadIdx.x / PA_WAVE;
   const int lane = threadIdx.x % PA_WAVE;
   const int waves_per_block = blockDim.x / PA_WAVE;
@@ -634,12 +667,15 @@ void store_lookup(torch::Tensor table_keys, torch::Tensor ticks,
                   int64_t dim, int64_t train, int64_t tick, double lo,
                   double hi, double admit_prob, double state_init,
                   int64_t opt_space, torch::Tensor evict_keys,
-                  torch::Tensor evict_count, torch::Tensor evict_rows) {
+                  torch::Tensor evict_count, torch::Tensor evict_rows,
+                  torch::Tensor u_count) {
   const int64_t n = query.numel();
   if (n == 0) return;
   const int64_t n_buckets = table_keys.numel() / PA_BUCKET_SIZE;
   const int row_width = (int)(dim + opt_space);
   const bool spill = evict_keys.numel() > 0;
+  const long long* n_dev =
+      u_count.numel() ? (const long long*)u_count.data_ptr<int64_t>() : nullptr;
   auto opts = torch::TensorOptions().dtype(torch::kInt64).device(query.device());
   auto slots = torch::empty({n}, opts);
   auto is_new = torch::empty({n}, opts.dtype(torch::kInt32));
@@ -655,7 +691,8 @@ void store_lookup(torch::Tensor table_keys, torch::Tensor ticks,
                      is_new.data_ptr<int32_t>(),
                      spill ? (ull*)evict_keys.data_ptr<int64_t>() : nullptr,
                      spill ? evict_count.data_ptr<int32_t>() : nullptr,
-                     spill ? (long long*)evict_idx.data_ptr<int64_t>() : nullptr);
+                     spill ? (long long*)evict_idx.data_ptr<int64_t>() : nullptr,
+                     n_dev);
   hipLaunchKernelGGL(init_gather_kernel, dim3(n_blocks_for(n, 4)), dim3(256),
                      0, st, arena.data_ptr<float>(),
                      (const ull*)query.data_ptr<int64_t>(),
@@ -663,7 +700,7 @@ void store_lookup(torch::Tensor table_keys, torch::Tensor ticks,
                      is_new.data_ptr<int32_t>(), out.data_ptr<float>(), n,
                      (int)dim, row_width, lo, hi, (float)state_init,
                      spill ? (const long long*)evict_idx.data_ptr<int64_t>() : nullptr,
-                     spill ? evict_rows.data_ptr<float>() : nullptr);
+                     spill ? evict_rows.data_ptr<float>() : nullptr, n_dev);
 }
 
 torch::Tensor store_probe(torch::Tensor table_keys, torch::Tensor ticks,
@@ -680,7 +717,8 @@ torch::Tensor store_probe(torch::Tensor table_keys, torch::Tensor ticks,
                      (const ull*)query.data_ptr<int64_t>(), n, n_buckets,
                      /*train=*/0, (unsigned)tick, 1.0f,
                      (long long*)slots.data_ptr<int64_t>(),
-                     is_new.data_ptr<int32_t>(), nullptr, nullptr, nullptr);
+                     is_new.data_ptr<int32_t>(), nullptr, nullptr, nullptr,
+                     nullptr);
   return slots;
 }
 
@@ -792,7 +830,7 @@ void scatter_update(torch::Tensor table_keys, torch::Tensor ticks,
                     torch::Tensor seg_scale, int64_t dim, int64_t opt,
                     std::vector<double> params, double b1_power,
                     double b2_power, double weight_bound,
-                    torch::Tensor skipped) {
+                    torch::Tensor skipped, torch::Tensor u_count) {
   const int64_t n = uniq.numel();
   if (n == 0) return;
   TORCH_CHECK(grads.scalar_type() == torch::kFloat16, "scatter_update: f16 grads");
@@ -813,7 +851,10 @@ void scatter_update(torch::Tensor table_keys, torch::Tensor ticks,
                      row_width, n_buckets, (int)opt, (float)params[0],
                      (float)params[1], (float)params[2], (float)params[3],
                      (float)b1_power, (float)b2_power, (float)weight_bound,
-                     skipped.data_ptr<int32_t>());
+                     skipped.data_ptr<int32_t>(),
+                     u_count.numel()
+                         ? (const long long*)u_count.data_ptr<int64_t>()
+                         : nullptr);
 }
 
 torch::Tensor sign_prep(torch::Tensor values, torch::Tensor slot_starts,
@@ -828,6 +869,23 @@ torch::Tensor sign_prep(torch::Tensor values, torch::Tensor slot_starts,
                      (int)(slot_starts.numel() - 1), (ull)spacing,
                      (ull*)out.data_ptr<int64_t>(), n);
   return out;
+}
+
+void dedup_finalize(torch::Tensor svals_flipped, torch::Tensor perm,
+                    torch::Tensor neq, torch::Tensor rank, int64_t flip,
+                    torch::Tensor inverse, torch::Tensor uniq,
+                    torch::Tensor ustarts, torch::Tensor u_count) {
+  const int64_t n = svals_flipped.numel();
+  if (n == 0) return;
+  hipLaunchKernelGGL(dedup_finalize_kernel, dim3(n_blocks_for(n, 256)),
+                     dim3(256), 0, cur_stream(),
+                     (const ull*)svals_flipped.data_ptr<int64_t>(),
+                     perm.data_ptr<int64_t>(), neq.data_ptr<bool>(),
+                     rank.data_ptr<int64_t>(), n, (ull)flip,
+                     inverse.data_ptr<int64_t>(),
+                     (ull*)uniq.data_ptr<int64_t>(),
+                     ustarts.data_ptr<int64_t>(),
+                     (long long*)u_count.data_ptr<int64_t>());
 }
 
 torch::Tensor sign_prep_stack(torch::Tensor values, torch::Tensor in_starts,
@@ -870,6 +928,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sign_prep", &sign_prep, "prefix-fold + splitmix64 key mixing");
   m.def("sign_prep_stack", &sign_prep_stack,
         "hashstack expansion + prefix-fold + splitmix64 key mixing");
+  m.def("dedup_finalize", &dedup_finalize,
+        "fixed-shape dedup epilogue (padded uniq/ustarts + device count)");
   m.def("scatter_update", &scatter_update,
         "fused ordered grad scatter + optimizer update (no [U,dim] buffer)");
 }
