@@ -98,6 +98,49 @@ class HostTier:
                 found[i] = True
         return rows, found
 
+    def export(self):
+        """-> (keys u64 [n], rows f32 [n, row_width]) for checkpointing."""
+        n = len(self._map)
+        keys = np.fromiter(self._map.keys(), dtype=np.uint64, count=n)
+        rows = (
+            np.stack(list(self._map.values()))
+            if n
+            else np.zeros((0, self.row_width), dtype=np.float32)
+        )
+        return keys, rows.astype(np.float32)
+
+
+class NativeHostTier:
+    """C++ host spill tier (csrc/engine.cpp NativeHostTier) behind the same
+    numpy interface as :class:`HostTier` — the python dict loops were the
+    dominant producer-thread cost of spill-heavy configs."""
+
+    def __init__(self, capacity: int, row_width: int):
+        from persia_amd.ops import native
+
+        self.capacity = capacity
+        self.row_width = row_width
+        self._impl = native().HostTier(capacity, row_width)
+
+    def __len__(self) -> int:
+        return int(self._impl.size())
+
+    def insert(self, keys: np.ndarray, rows: np.ndarray) -> None:
+        self._impl.insert(
+            torch.from_numpy(np.ascontiguousarray(keys).view(np.int64)),
+            torch.from_numpy(np.ascontiguousarray(rows, dtype=np.float32)),
+        )
+
+    def fetch(self, keys: np.ndarray):
+        rows, found = self._impl.fetch(
+            torch.from_numpy(np.ascontiguousarray(keys).view(np.int64))
+        )
+        return rows.numpy(), found.numpy()
+
+    def export(self):
+        keys, rows = self._impl.export_all()
+        return keys.numpy().view(np.uint64), rows.numpy()
+
 
 class EmbeddingStoreBase:
     """One rank's shard for one dim-group of slots."""
@@ -162,10 +205,8 @@ class EmbeddingStoreBase:
         """Host-tier rows for checkpointing (spilled rows are table state)."""
         if self.spill is None or len(self.spill) == 0:
             return None
-        keys = np.fromiter(self.spill._map.keys(), dtype=np.uint64,
-                           count=len(self.spill._map))
-        rows = np.stack(list(self.spill._map.values()))
-        return hashing.splitmix64_inv(keys), rows.astype(np.float32)
+        keys, rows = self.spill.export()
+        return hashing.splitmix64_inv(keys), rows
 
     def _adam_step_powers(self):
         o = self.optimizer
@@ -369,6 +410,9 @@ class HipEmbeddingStore(EmbeddingStoreBase):
 
         self._C = native()
         assert device.type == "cuda"
+        if self.spill is not None:
+            # native spill tier: same LRU semantics, no python dict loops
+            self.spill = NativeHostTier(spill_capacity, self.row_width)
         self.keys = torch.zeros(self.n_slots, dtype=torch.int64, device=device)
         self.ticks = torch.zeros(self.n_slots, dtype=torch.int32, device=device)
         self.arena = torch.zeros(

@@ -148,8 +148,103 @@ void update_local(torch::Tensor grads, torch::Tensor perm,
                b1_power, b2_power, weight_bound, skipped);
 }
 
+
+// ---------------------------------------------------------------- host tier
+// Host-DRAM spill tier (native): rows evicted from the HBM table park here
+// and come back on their next lookup.  Exclusive bounded LRU with
+// insert-order recency, matching persia_amd/core/store.py HostTier (the
+// python oracle) exactly: re-insert refreshes recency, fetch removes, the
+// oldest entry is dropped past capacity.  Replaces per-key python dict
+// loops in the pipeline thread (reference: remote CPU parameter-server
+// shards, embedding_parameter_server/lib.rs).
+#include <list>
+#include <unordered_map>
+
+struct NativeHostTier {
+  int64_t capacity, row_width;
+  std::list<std::pair<uint64_t, std::vector<float>>> lru;  // front = oldest
+  std::unordered_map<uint64_t,
+                     std::list<std::pair<uint64_t, std::vector<float>>>::iterator>
+      map;
+
+  NativeHostTier(int64_t cap, int64_t rw) : capacity(cap), row_width(rw) {}
+
+  void insert(torch::Tensor keys, torch::Tensor rows) {
+    TORCH_CHECK(!keys.is_cuda() && !rows.is_cuda(), "HostTier: cpu tensors");
+    TORCH_CHECK(rows.size(1) == row_width, "HostTier: row width mismatch");
+    auto kc = keys.contiguous();
+    auto rc = rows.contiguous();
+    const int64_t* kp = kc.data_ptr<int64_t>();
+    const float* rp = rc.data_ptr<float>();
+    const int64_t k = kc.numel();
+    for (int64_t i = 0; i < k; ++i) {
+      const uint64_t key = (uint64_t)kp[i];
+      auto it = map.find(key);
+      if (it != map.end()) {
+        lru.erase(it->second);
+        map.erase(it);
+      }
+      lru.emplace_back(key, std::vector<float>(rp + i * row_width,
+                                               rp + (i + 1) * row_width));
+      map[key] = std::prev(lru.end());
+    }
+    while ((int64_t)lru.size() > capacity) {
+      map.erase(lru.front().first);
+      lru.pop_front();
+    }
+  }
+
+  std::tuple<torch::Tensor, torch::Tensor> fetch(torch::Tensor keys) {
+    auto kc = keys.contiguous();
+    const int64_t* kp = kc.data_ptr<int64_t>();
+    const int64_t k = kc.numel();
+    auto rows = torch::zeros({k, row_width}, torch::kFloat32);
+    auto found = torch::zeros({k}, torch::kBool);
+    float* rp = rows.data_ptr<float>();
+    bool* fp = found.data_ptr<bool>();
+    for (int64_t i = 0; i < k; ++i) {
+      auto it = map.find((uint64_t)kp[i]);
+      if (it == map.end()) continue;
+      std::copy(it->second->second.begin(), it->second->second.end(),
+                rp + i * row_width);
+      fp[i] = true;
+      lru.erase(it->second);
+      map.erase(it);
+    }
+    return {rows, found};
+  }
+
+  std::tuple<torch::Tensor, torch::Tensor> export_all() {
+    const int64_t n = (int64_t)lru.size();
+    auto keys = torch::empty({n}, torch::kInt64);
+    auto rows = torch::empty({n, row_width}, torch::kFloat32);
+    int64_t* kp = keys.data_ptr<int64_t>();
+    float* rp = rows.data_ptr<float>();
+    int64_t i = 0;
+    for (auto& e : lru) {
+      kp[i] = (int64_t)e.first;
+      std::copy(e.second.begin(), e.second.end(), rp + i * row_width);
+      ++i;
+    }
+    return {keys, rows};
+  }
+
+  int64_t size() const { return (int64_t)lru.size(); }
+  void clear() {
+    lru.clear();
+    map.clear();
+  }
+};
+
 void init_engine(pybind11::module_& m) {
   m.def("dedup_keys", &dedup_keys, "sort-based dedup of u64 keys");
+  pybind11::class_<NativeHostTier>(m, "HostTier")
+      .def(pybind11::init<int64_t, int64_t>())
+      .def("insert", &NativeHostTier::insert)
+      .def("fetch", &NativeHostTier::fetch)
+      .def("export_all", &NativeHostTier::export_all)
+      .def("size", &NativeHostTier::size)
+      .def("clear", &NativeHostTier::clear);
   m.def("lookup_local", &lookup_local,
         "fused single-GPU lookup (sign prep + dedup + probe + gather + sum)");
   m.def("update_local", &update_local,

@@ -1,6 +1,7 @@
 """Host-DRAM spill tier semantics (CPU model; the HIP path is covered by
 tests/test_gpu_kernels.py::test_spill_roundtrip_gpu)."""
 import numpy as np
+import pytest
 import torch
 
 from persia_amd.core import hashing
@@ -64,6 +65,33 @@ def test_export_includes_spilled_rows():
     signs, inner = store.export_rows()
     assert len(signs) == len(store) + len(store.spill)
     assert inner.shape[1] == store.row_width
+
+
+def test_native_host_tier_matches_python_oracle():
+    """C++ NativeHostTier (csrc/engine.cpp) vs the python HostTier under a
+    random insert/fetch/evict workload: identical membership, rows, sizes,
+    and export contents."""
+    pytest.importorskip("persia_amd._C")
+    from persia_amd.core.store import HostTier, NativeHostTier
+
+    rng = np.random.default_rng(0)
+    py, nat = HostTier(50, 4), NativeHostTier(50, 4)
+    for step in range(300):
+        k = rng.integers(0, 120, size=rng.integers(1, 9), dtype=np.uint64)
+        if step % 3 == 2:
+            rp, fp = py.fetch(k)
+            rn, fn = nat.fetch(k)
+            assert np.array_equal(fp, fn), step
+            assert np.allclose(rp, rn), step
+        else:
+            r = rng.normal(size=(len(k), 4)).astype(np.float32)
+            py.insert(k, r)
+            nat.insert(k, r)
+        assert len(py) == len(nat), step
+    kp, rp = py.export()
+    kn, rn = nat.export()
+    assert np.array_equal(np.sort(kp), np.sort(kn))
+    assert np.allclose(rp[np.argsort(kp)], rn[np.argsort(kn)])
 
 
 def test_spill_capacity_bounded():
