@@ -134,6 +134,9 @@ def main():
             device_ids=[local_rank] if use_gpu else None,
             gradient_as_bucket_view=True,
             bucket_cap_mb=50,
+            # required for capturing the DDP backward (bucket allreduce
+            # order must be frozen) into the per-iteration hipGraph
+            static_graph=bool(args.graph) and use_gpu,
         )
     opt = torch.optim.SGD(model.parameters(), lr=0.01)
     loss_fn = torch.nn.functional.binary_cross_entropy_with_logits

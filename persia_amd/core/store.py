@@ -98,6 +98,9 @@ class HostTier:
                 found[i] = True
         return rows, found
 
+    def __contains__(self, key) -> bool:
+        return int(key) in self._map
+
     def export(self):
         """-> (keys u64 [n], rows f32 [n, row_width]) for checkpointing."""
         n = len(self._map)
@@ -136,6 +139,9 @@ class NativeHostTier:
             torch.from_numpy(np.ascontiguousarray(keys).view(np.int64))
         )
         return rows.numpy(), found.numpy()
+
+    def __contains__(self, key) -> bool:
+        return bool(self._impl.contains(int(np.int64(np.uint64(key)))))
 
     def export(self):
         keys, rows = self._impl.export_all()

@@ -230,6 +230,7 @@ struct NativeHostTier {
   }
 
   int64_t size() const { return (int64_t)lru.size(); }
+  bool contains(int64_t key) const { return map.count((uint64_t)key) != 0; }
   void clear() {
     lru.clear();
     map.clear();
@@ -244,6 +245,7 @@ void init_engine(pybind11::module_& m) {
       .def("fetch", &NativeHostTier::fetch)
       .def("export_all", &NativeHostTier::export_all)
       .def("size", &NativeHostTier::size)
+      .def("contains", &NativeHostTier::contains)
       .def("clear", &NativeHostTier::clear);
   m.def("lookup_local", &lookup_local,
         "fused single-GPU lookup (sign prep + dedup + probe + gather + sum)");

@@ -149,7 +149,7 @@ def test_spill_roundtrip_gpu():
     sign = 1000
     # flood until key 7 lands in the host tier (don't probe it — probing
     # refreshes its tick and protects it from eviction)
-    while key7 not in store.spill._map:
+    while key7 not in store.spill:
         store.lookup(_keys(list(range(sign, sign + 8))).to(_dev()), train=True)
         sign += 8
         assert sign < 100000, "sign 7 never evicted?"
