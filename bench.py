@@ -126,6 +126,15 @@ def main():
         ).to(device)
     else:
         model = DCNv2(num_sparse=n_slots, num_dense=args.num_dense, dim=dim).to(device)
+    # bf16 model weights + f32 master weights (see the graph block below):
+    # must happen BEFORE the DDP wrap so the gradient buckets are built bf16
+    # (halves allreduce bytes over xGMI as a bonus)
+    bf16_weights = (
+        bool(args.graph) and use_gpu
+        and os.environ.get("PA_GRAPH_BF16", "1") == "1"
+    )
+    if bf16_weights:
+        model.bfloat16()
     if world > 1:
         from torch.nn.parallel import DistributedDataParallel
 
@@ -183,7 +192,7 @@ def main():
     # replays (the dense side was launch-bound, ~200 kernels/step).
     graph = None
     static = {}
-    if args.graph and use_gpu and world == 1:
+    if args.graph and use_gpu:
         try:
             static = {
                 "dense": torch.zeros(B, args.num_dense, device=device),
@@ -201,9 +210,8 @@ def main():
             # copy preserves full-precision SGD accumulation — 4 multi-tensor
             # launches replace the per-layer cast storm.
             # PA_GRAPH_BF16=0 selects the autocast variant (A/B switch).
-            bf16_weights = os.environ.get("PA_GRAPH_BF16", "1") == "1"
+            # (model.bfloat16() itself ran before the DDP wrap, above)
             if bf16_weights:
-                model.bfloat16()
                 g_params = list(model.parameters())
                 g_masters = [p.detach().clone().float() for p in g_params]
                 g_grads32 = [torch.zeros_like(m) for m in g_masters]
