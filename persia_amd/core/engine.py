@@ -394,7 +394,7 @@ class EmbeddingEngine:
         """uniq keys -> rows [U, dim] (wire dtype on GPU path, f32 on CPU)."""
         store = self.stores[group.dim]
         if not self.dist.distributed:
-            rows = store.lookup(group.uniq_keys, train)
+            rows = store.lookup(group.uniq_keys, train, u_count=group.u_count)
             return rows
         owner = _owner_of_keys(group.uniq_keys, self.dist.world_size)
         send_counts = torch.bincount(owner, minlength=self.dist.world_size).tolist()
@@ -560,11 +560,19 @@ class EmbeddingEngine:
             return group
         else:
             keys_t = C.sign_prep(vals_t, plan.slot_starts, plan.prefixes, spacing_arg)
-            uniq_keys, inverse, perm, ustarts = _dedup(keys_t)
+            if not self.dist.distributed:
+                # spill path, single GPU: sync-free padded dedup (the spill
+                # restore phase has its own host consult, but the dedup needn't
+                # add two more stream syncs)
+                uniq_keys, inverse, perm, ustarts, u_count = C.dedup_padded(keys_t)
+            else:
+                uniq_keys, inverse, perm, ustarts = _dedup(keys_t)
+                u_count = None
             group = _GroupCtx(
                 dim=dim, uniq_keys=uniq_keys, inverse=inverse, perm=perm,
-                ustarts=ustarts, slots=slot_ctxs, cat_offsets=plan.cat_offsets,
-                seg_id=plan.seg_id, n_sum_slots=plan.S,
+                ustarts=ustarts, u_count=u_count, slots=slot_ctxs,
+                cat_offsets=plan.cat_offsets, seg_id=plan.seg_id,
+                n_sum_slots=plan.S,
             )
             rows = self._exchange_rows(group, train)
             sums = C.segment_sum(

@@ -180,11 +180,12 @@ class EmbeddingStoreBase:
     def __len__(self) -> int:
         raise NotImplementedError
 
-    def lookup(self, keys: torch.Tensor, train: bool) -> torch.Tensor:
+    def lookup(self, keys: torch.Tensor, train: bool,
+               u_count: Optional[torch.Tensor] = None) -> torch.Tensor:
         """keys: int64[n] (mixed u64 bit pattern) -> rows float32[n, dim].
         train=True inserts on miss (admit-gated, seeded init);
         train=False returns zeros on miss (reference parameter
-        mod.rs:231-251)."""
+        mod.rs:231-251).  ``u_count``: HIP-only padded-prefix length."""
         raise NotImplementedError
 
     def update_gradients(self, keys: torch.Tensor, grads: torch.Tensor) -> int:
@@ -306,7 +307,9 @@ class CpuEmbeddingStore(EmbeddingStoreBase):
         return float(_u01_from_u64(u)[0]) < self.hyper.admit_probability
 
     # -- public ops ---------------------------------------------------------
-    def lookup(self, keys: torch.Tensor, train: bool) -> torch.Tensor:
+    def lookup(self, keys: torch.Tensor, train: bool,
+               u_count: Optional[torch.Tensor] = None) -> torch.Tensor:
+        assert u_count is None, "padded lookup is a HIP-path feature"
         tick = self.next_tick()
         k_np = keys.cpu().numpy().view(np.uint64).copy()
         k_np[k_np == EMPTY_KEY] = _ZERO_REMAP
@@ -495,16 +498,27 @@ class HipEmbeddingStore(EmbeddingStoreBase):
     def __len__(self) -> int:
         return int((self.keys != 0).sum().item())
 
-    def lookup(self, keys: torch.Tensor, train: bool) -> torch.Tensor:
+    def lookup(
+        self, keys: torch.Tensor, train: bool,
+        u_count: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        """``u_count``: optional device-side i64[1] valid-prefix length for
+        nnz-padded ``keys`` (sync-free dedup) — the padding tail is the
+        empty-key sentinel and is never probed or claimed."""
         tick = self.next_tick()
         n = keys.numel()
+        uc = (
+            u_count
+            if u_count is not None
+            else torch.empty(0, dtype=torch.int64, device=self.device)
+        )
         ev = self._no_evict
         if self.spill is not None and train and n:
             self._drain_pending()
             ev = self._evict_buffers(n)
         if self.spill is not None and train and n and len(self.spill):
             # spill phase 1: restore missing-but-spilled rows into HBM first
-            slots = self._C.store_probe(self.keys, self.ticks, keys, tick)
+            slots = self._C.store_probe(self.keys, self.ticks, keys, tick, uc)
             miss_keys = keys[slots < 0]
             if miss_keys.numel():
                 miss_np = miss_keys.cpu().numpy().view(np.uint64)
@@ -537,7 +551,7 @@ class HipEmbeddingStore(EmbeddingStoreBase):
             float(self.optimizer.state_init(self.dim)),
             self.opt_space,
             *ev,
-            torch.empty(0, dtype=torch.int64, device=self.device),
+            uc,
         )
         self._drain_evictions(ev)
         return out

@@ -71,23 +71,18 @@ std::vector<torch::Tensor> dedup_keys(torch::Tensor keys) {
   return {uniq, inverse, perm, ustarts};
 }
 
-// Single-GPU fused lookup: raw values -> per-sample summed embeddings.
-// Returns {sums f16 [n_segs, dim], uniq_keys, inverse, perm, ustarts}.
-std::vector<torch::Tensor> lookup_local(
-    torch::Tensor values, torch::Tensor slot_starts, torch::Tensor prefixes,
-    int64_t spacing, torch::Tensor cat_offsets, torch::Tensor seg_scale,
-    torch::Tensor table_keys, torch::Tensor ticks, torch::Tensor arena,
-    int64_t dim, int64_t train, int64_t tick, double lo, double hi,
-    double admit_prob, double state_init, int64_t opt_space) {
-  auto keys = sign_prep(values, slot_starts, prefixes, spacing);
-  // fixed-shape dedup: every tensor is nnz-padded and the true unique count
-  // lives ONLY on the device (u_count) — no host synchronization anywhere in
-  // the lookup, so the pipeline thread issues the whole batch and returns
-  // (the old masked_select/nonzero dedup forced two stream syncs per batch,
-  // serializing the producer behind the GPU)
+// Fixed-shape dedup: every tensor is nnz-padded and the true unique count
+// lives ONLY on the device (u_count) — no host synchronization anywhere, so
+// the pipeline thread issues the whole batch and returns (a
+// masked_select/nonzero dedup forces two stream syncs per batch,
+// serializing the producer behind the GPU).  The uniq tail is ZERO-filled
+// (the empty-key sentinel, which sign prep never produces), so consumers
+// that do scan past u_count probe an impossible key.
+// -> {uniq zeros-padded [nnz], inverse [nnz], perm [nnz],
+//     ustarts padded [nnz+1], u_count device i64[1]}
+std::vector<torch::Tensor> dedup_padded(torch::Tensor keys) {
   const int64_t nnz = keys.numel();
-  auto dev = values.device();
-  auto opts = torch::TensorOptions().dtype(torch::kInt64).device(dev);
+  auto opts = torch::TensorOptions().dtype(torch::kInt64).device(keys.device());
   constexpr int64_t kFlipLocal = std::numeric_limits<int64_t>::min();
   auto flipped = keys.bitwise_xor(kFlipLocal);
   auto sorted = flipped.sort();
@@ -99,11 +94,32 @@ std::vector<torch::Tensor> lookup_local(
   auto rank = neq.cumsum(0);
   rank.sub_(1);
   auto inverse = torch::empty({nnz}, opts);
-  auto uniq = torch::empty({nnz}, opts);
+  auto uniq = torch::zeros({nnz}, opts);
   auto ustarts = torch::empty({nnz + 1}, opts);
   auto u_count = torch::empty({1}, opts);
   dedup_finalize(svals, perm, neq, rank, kFlipLocal, inverse, uniq, ustarts,
                  u_count);
+  return {uniq, inverse, perm, ustarts, u_count};
+}
+
+// Single-GPU fused lookup: raw values -> per-sample summed embeddings.
+// Returns {sums f16 [n_segs, dim], uniq_keys, inverse, perm, ustarts}.
+std::vector<torch::Tensor> lookup_local(
+    torch::Tensor values, torch::Tensor slot_starts, torch::Tensor prefixes,
+    int64_t spacing, torch::Tensor cat_offsets, torch::Tensor seg_scale,
+    torch::Tensor table_keys, torch::Tensor ticks, torch::Tensor arena,
+    int64_t dim, int64_t train, int64_t tick, double lo, double hi,
+    double admit_prob, double state_init, int64_t opt_space) {
+  auto keys = sign_prep(values, slot_starts, prefixes, spacing);
+  auto d = dedup_padded(keys);
+  auto& uniq = d[0];
+  auto& inverse = d[1];
+  auto& perm = d[2];
+  auto& ustarts = d[3];
+  auto& u_count = d[4];
+  const int64_t nnz = keys.numel();
+  auto dev = values.device();
+  auto opts = torch::TensorOptions().dtype(torch::kInt64).device(dev);
   auto rows = torch::empty(
       {nnz, dim}, torch::TensorOptions().dtype(torch::kFloat32).device(dev));
   auto none = torch::empty({0}, opts);
@@ -239,6 +255,8 @@ struct NativeHostTier {
 
 void init_engine(pybind11::module_& m) {
   m.def("dedup_keys", &dedup_keys, "sort-based dedup of u64 keys");
+  m.def("dedup_padded", &dedup_padded,
+        "sync-free dedup: nnz-padded outputs + device unique count");
   pybind11::class_<NativeHostTier>(m, "HostTier")
       .def(pybind11::init<int64_t, int64_t>())
       .def("insert", &NativeHostTier::insert)

@@ -704,10 +704,13 @@ void store_lookup(torch::Tensor table_keys, torch::Tensor ticks,
 }
 
 torch::Tensor store_probe(torch::Tensor table_keys, torch::Tensor ticks,
-                          torch::Tensor query, int64_t tick) {
+                          torch::Tensor query, int64_t tick,
+                          torch::Tensor u_count) {
   const int64_t n = query.numel();
   auto opts = torch::TensorOptions().dtype(torch::kInt64).device(query.device());
-  auto slots = torch::empty({n}, opts);
+  // padded queries: slots defaults to 0 ("found") beyond the device count so
+  // the padding never reads as missing
+  auto slots = u_count.numel() ? torch::zeros({n}, opts) : torch::empty({n}, opts);
   if (n == 0) return slots;
   auto is_new = torch::empty({n}, opts.dtype(torch::kInt32));
   const int64_t n_buckets = table_keys.numel() / PA_BUCKET_SIZE;
@@ -718,7 +721,9 @@ torch::Tensor store_probe(torch::Tensor table_keys, torch::Tensor ticks,
                      /*train=*/0, (unsigned)tick, 1.0f,
                      (long long*)slots.data_ptr<int64_t>(),
                      is_new.data_ptr<int32_t>(), nullptr, nullptr, nullptr,
-                     nullptr);
+                     u_count.numel()
+                         ? (const long long*)u_count.data_ptr<int64_t>()
+                         : nullptr);
   return slots;
 }
 
