@@ -48,11 +48,10 @@ class DotInteraction(nn.Module):
     def forward(self, vectors: torch.Tensor) -> torch.Tensor:
         # vectors: [B, F, D]
         F, D = vectors.shape[1], vectors.shape[2]
-        bwd_lds = 4 * (F * (D + 8) * 2 + 2 * F * F)  # must fit 64KB (4 waves)
-        if vectors.is_cuda and D % 8 == 0 and bwd_lds <= 65536:
-            from persia_amd.ops import native_available
+        if vectors.is_cuda and D % 8 == 0:
+            from persia_amd.ops import native, native_available
 
-            if native_available():
+            if native_available() and native().interact_feasible(F, D):
                 return _InteractFn.apply(vectors)
         B, F, _D = vectors.shape
         prod = torch.bmm(vectors, vectors.transpose(1, 2))  # [B, F, F]
