@@ -212,6 +212,59 @@ __global__ void init_gather_kernel(float* __restrict__ arena,
   }
 }
 
+// init + DIRECT f16 scatter to the per-position sum rows (fast-path variant
+// of init_gather_kernel: the all-single-ID path has exactly one id per
+// segment, so segment p's sum IS the row of the unique key at sorted
+// position p — writing sums[perm[p]] straight from the arena skips the
+// [nnz, dim] f32 rows round-trip that segment_sum would re-read).
+__global__ void init_scatter_kernel(float* __restrict__ arena,
+                                    const ull* __restrict__ query,
+                                    const long long* __restrict__ slots,
+                                    const int* __restrict__ is_new,
+                                    const int64_t* __restrict__ perm,
+                                    const int64_t* __restrict__ ustarts,
+                                    __half* __restrict__ sums, int64_t n,
+                                    int dim, int row_width, double lo,
+                                    double hi, float state_init,
+                                    const long long* __restrict__ n_dev) {
+  if (n_dev) n = *n_dev;
+  const int wave = threadIdx.x / PA_WAVE;
+  const int lane = threadIdx.x % PA_WAVE;
+  const int waves_per_block = blockDim.x / PA_WAVE;
+  const int G = (dim < PA_WAVE && (PA_WAVE % dim) == 0) ? PA_WAVE / dim : 1;
+  const int sub = (G > 1) ? lane / dim : 0;
+  const int c0 = (G > 1) ? lane % dim : lane;
+  const int st = (G > 1) ? dim : PA_WAVE;
+  const int64_t n_waveitems = (n + G - 1) / G;
+  for (int64_t w = (int64_t)blockIdx.x * waves_per_block + wave;
+       w < n_waveitems; w += (int64_t)gridDim.x * waves_per_block) {
+    const int64_t i = w * G + sub;
+    if (i >= n) continue;
+    const long long slot = slots[i];
+    const int64_t lo_p = ustarts[i], hi_p = ustarts[i + 1];
+    if (slot < 0) {
+      for (int64_t p = lo_p; p < hi_p; ++p) {
+        __half* dst = sums + perm[p] * dim;
+        for (int c = c0; c < dim; c += st) dst[c] = __float2half(0.0f);
+      }
+      continue;
+    }
+    float* row = arena + (int64_t)slot * row_width;
+    if (is_new[i]) {
+      const uint64_t sign = pa_splitmix64_inv((uint64_t)query[i]);
+      const uint64_t seed = pa_init_seed(sign);
+      for (int c = c0; c < dim; c += st)
+        row[c] = pa_init_val(seed, c, lo, hi);
+      for (int c = dim + c0; c < row_width; c += st)
+        row[c] = state_init;
+    }
+    for (int64_t p = lo_p; p < hi_p; ++p) {
+      __half* dst = sums + perm[p] * dim;
+      for (int c = c0; c < dim; c += st) dst[c] = __float2half(row[c]);
+    }
+  }
+}
+
 // ------------------------------------------------------------- sparse update
 
 // one wave per key: probe (lane-parallel over the 32-slot window) + fused
@@ -703,6 +756,43 @@ void store_lookup(torch::Tensor table_keys, torch::Tensor ticks,
                      spill ? evict_rows.data_ptr<float>() : nullptr, n_dev);
 }
 
+// fast-path lookup: probe/claim + fused init/f16-scatter to sum rows
+// (single-ID groups only — no spill; see init_scatter_kernel)
+void store_lookup_sums(torch::Tensor table_keys, torch::Tensor ticks,
+                       torch::Tensor arena, torch::Tensor query,
+                       torch::Tensor perm, torch::Tensor ustarts,
+                       torch::Tensor sums, int64_t dim, int64_t train,
+                       int64_t tick, double lo, double hi, double admit_prob,
+                       double state_init, int64_t opt_space,
+                       torch::Tensor u_count) {
+  const int64_t n = query.numel();
+  if (n == 0) return;
+  const int64_t n_buckets = table_keys.numel() / PA_BUCKET_SIZE;
+  const int row_width = (int)(dim + opt_space);
+  const long long* n_dev =
+      u_count.numel() ? (const long long*)u_count.data_ptr<int64_t>() : nullptr;
+  auto opts = torch::TensorOptions().dtype(torch::kInt64).device(query.device());
+  auto slots = torch::empty({n}, opts);
+  auto is_new = torch::empty({n}, opts.dtype(torch::kInt32));
+  hipStream_t st = cur_stream();
+  hipLaunchKernelGGL(probe_claim_kernel, dim3(n_blocks_for(n, 256)), dim3(256),
+                     0, st, (ull*)table_keys.data_ptr<int64_t>(),
+                     (unsigned*)ticks.data_ptr<int32_t>(),
+                     (const ull*)query.data_ptr<int64_t>(), n, n_buckets,
+                     (int)train, (unsigned)tick, (float)admit_prob,
+                     (long long*)slots.data_ptr<int64_t>(),
+                     is_new.data_ptr<int32_t>(), nullptr, nullptr, nullptr,
+                     n_dev);
+  hipLaunchKernelGGL(init_scatter_kernel, dim3(n_blocks_for(n, 4)), dim3(256),
+                     0, st, arena.data_ptr<float>(),
+                     (const ull*)query.data_ptr<int64_t>(),
+                     (const long long*)slots.data_ptr<int64_t>(),
+                     is_new.data_ptr<int32_t>(), perm.data_ptr<int64_t>(),
+                     ustarts.data_ptr<int64_t>(),
+                     (__half*)sums.data_ptr<at::Half>(), n, (int)dim,
+                     row_width, lo, hi, (float)state_init, n_dev);
+}
+
 torch::Tensor store_probe(torch::Tensor table_keys, torch::Tensor ticks,
                           torch::Tensor query, int64_t tick,
                           torch::Tensor u_count) {
@@ -926,6 +1016,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   init_interact(m);
   m.def("store_lookup", &store_lookup, "hash-table lookup/insert + gather");
   m.def("store_probe", &store_probe, "probe-only (spill-tier miss detection)");
+  m.def("store_lookup_sums", &store_lookup_sums,
+        "fast-path lookup: probe + fused init/f16 scatter to sum rows");
   m.def("store_update", &store_update, "fused sparse optimizer update");
   m.def("store_import", &store_import, "bulk insert rows (checkpoint load)");
   m.def("segment_sum", &segment_sum, "fused gather + per-sample summation");
