@@ -49,6 +49,8 @@ void dedup_finalize(torch::Tensor svals_flipped, torch::Tensor perm,
                     torch::Tensor neq, torch::Tensor rank, int64_t flip,
                     torch::Tensor inverse, torch::Tensor uniq,
                     torch::Tensor ustarts, torch::Tensor u_count);
+std::vector<torch::Tensor> sort_pairs_u64(torch::Tensor keys);
+torch::Tensor neq_flags(torch::Tensor sorted);
 
 static constexpr int64_t kFlip = std::numeric_limits<int64_t>::min();
 
@@ -90,21 +92,18 @@ std::vector<torch::Tensor> dedup_keys(torch::Tensor keys) {
 std::vector<torch::Tensor> dedup_padded(torch::Tensor keys) {
   const int64_t nnz = keys.numel();
   auto opts = torch::TensorOptions().dtype(torch::kInt64).device(keys.device());
-  constexpr int64_t kFlipLocal = std::numeric_limits<int64_t>::min();
-  auto flipped = keys.bitwise_xor(kFlipLocal);
-  auto sorted = flipped.sort();
-  auto svals = std::get<0>(sorted);
-  auto perm = std::get<1>(sorted);
-  auto neq = torch::ones({nnz}, opts.dtype(torch::kBool));
-  neq.slice(0, 1, nnz) =
-      svals.slice(0, 1, nnz).ne(svals.slice(0, 0, nnz - 1));
+  // u64 radix sort needs no sign-flip: unsigned order IS key/owner order
+  auto sp = sort_pairs_u64(keys);
+  auto& svals = sp[0];
+  auto& perm = sp[1];
+  auto neq = neq_flags(svals);
   auto rank = neq.cumsum(0);
   rank.sub_(1);
   auto inverse = torch::empty({nnz}, opts);
   auto uniq = torch::zeros({nnz}, opts);
   auto ustarts = torch::empty({nnz + 1}, opts);
   auto u_count = torch::empty({1}, opts);
-  dedup_finalize(svals, perm, neq, rank, kFlipLocal, inverse, uniq, ustarts,
+  dedup_finalize(svals, perm, neq, rank, /*flip=*/0, inverse, uniq, ustarts,
                  u_count);
   return {uniq, inverse, perm, ustarts, u_count};
 }

@@ -27,6 +27,7 @@
 #include <ATen/hip/HIPContext.h>
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
+#include <hipcub/hipcub.hpp>
 
 #include "common.h"
 
@@ -441,6 +442,20 @@ __global__ void sign_prep_stack_kernel(
   }
 }
 
+__global__ void iota_i64_kernel(int64_t* __restrict__ out, int64_t n) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) out[i] = i;
+}
+
+// flags[i] = (sorted[i] != sorted[i-1]); flags[0] = 1  (boundary marks for
+// the rank scan — replaces a ones+slice+ne op chain)
+__global__ void neq_flags_kernel(const ull* __restrict__ sorted,
+                                 bool* __restrict__ flags, int64_t n) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  flags[i] = (i == 0) || (sorted[i] != sorted[i - 1]);
+}
+
 // Fixed-shape dedup epilogue: after sort + neq + rank=cumsum(neq)-1 (all
 // shape-static ATen ops), ONE pass finalizes inverse/uniq/ustarts into
 // nnz-padded buffers and writes the true unique count to a device scalar —
@@ -714,6 +729,48 @@ inline int n_blocks_for(int64_t work_items, int per_block) {
 static hipStream_t cur_stream() {
   return at::hip::getCurrentHIPStream().stream();
 }
+
+// u64 radix sort-pairs (keys ascending in UNSIGNED order — exactly the
+// dedup/owner-range order; torch.sort would need the sign-flip trick AND
+// runs a merge sort with a slow index-iota setup: ~3x the time)
+std::vector<torch::Tensor> sort_pairs_u64(torch::Tensor keys) {
+  const int64_t n = keys.numel();
+  auto opts = torch::TensorOptions().dtype(torch::kInt64).device(keys.device());
+  auto iota = torch::empty({n}, opts);
+  auto sorted = torch::empty_like(keys);
+  auto perm = torch::empty({n}, opts);
+  if (n == 0) return {sorted, perm};
+  hipStream_t st = cur_stream();
+  hipLaunchKernelGGL(iota_i64_kernel, dim3(n_blocks_for(n, 256)), dim3(256),
+                     0, st, iota.data_ptr<int64_t>(), n);
+  size_t temp_bytes = 0;
+  hipcub::DeviceRadixSort::SortPairs(
+      nullptr, temp_bytes, (const ull*)keys.data_ptr<int64_t>(),
+      (ull*)sorted.data_ptr<int64_t>(),
+      (const long long*)iota.data_ptr<int64_t>(),
+      (long long*)perm.data_ptr<int64_t>(), n, 0, 64, st);
+  auto temp = torch::empty(
+      {(int64_t)temp_bytes},
+      torch::TensorOptions().dtype(torch::kUInt8).device(keys.device()));
+  hipcub::DeviceRadixSort::SortPairs(
+      temp.data_ptr(), temp_bytes, (const ull*)keys.data_ptr<int64_t>(),
+      (ull*)sorted.data_ptr<int64_t>(),
+      (const long long*)iota.data_ptr<int64_t>(),
+      (long long*)perm.data_ptr<int64_t>(), n, 0, 64, st);
+  return {sorted, perm};
+}
+
+torch::Tensor neq_flags(torch::Tensor sorted) {
+  const int64_t n = sorted.numel();
+  auto flags = torch::empty(
+      {n}, torch::TensorOptions().dtype(torch::kBool).device(sorted.device()));
+  if (n == 0) return flags;
+  hipLaunchKernelGGL(neq_flags_kernel, dim3(n_blocks_for(n, 256)), dim3(256),
+                     0, cur_stream(), (const ull*)sorted.data_ptr<int64_t>(),
+                     flags.data_ptr<bool>(), n);
+  return flags;
+}
+
 
 void store_lookup(torch::Tensor table_keys, torch::Tensor ticks,
                   torch::Tensor arena, torch::Tensor query, torch::Tensor out,
@@ -1027,6 +1084,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "hashstack expansion + prefix-fold + splitmix64 key mixing");
   m.def("dedup_finalize", &dedup_finalize,
         "fixed-shape dedup epilogue (padded uniq/ustarts + device count)");
+  m.def("sort_pairs_u64", &sort_pairs_u64, "radix sort-pairs, u64 order");
+  m.def("neq_flags", &neq_flags, "sorted-run boundary flags");
   m.def("scatter_update", &scatter_update,
         "fused ordered grad scatter + optimizer update (no [U,dim] buffer)");
 }
