@@ -156,6 +156,129 @@ __global__ __launch_bounds__(256) void interact_bwd_mfma_kernel(
   }
 }
 
+// ---------------------------------------------- MFMA packed-input variants
+// The DLRM flagship path feeds the interaction from TWO sources: the bottom
+// MLP output x [B, D] (bf16) and the engine's slot-major sum base
+// [S*B, D] (f16, row (f-1)*B + b for feature f>=1).  Staging straight from
+// both skips the [B, F, D] cat + permute + f16->bf16 materialization the
+// model would otherwise run (and their backward copy cascade).
+__global__ __launch_bounds__(256) void interact_fwd_packed_kernel(
+    const short* __restrict__ x, const __half* __restrict__ base,
+    short* __restrict__ out, int B, int S, int D, int P) {
+  extern __shared__ short lds[];
+  const int F = S + 1;
+  const int waves = blockDim.x / 64;
+  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  const int Fp = ceil16(F), Dk = ceil32(D);
+  const int LDK = Dk + 8;
+  const int T = Fp / 16;
+  short* v = lds + wave * Fp * LDK;
+  for (int t = lane; t < Fp * LDK / 8; t += 64) *(bf16x8i*)&v[t * 8] = bf16x8i{};
+  const int fi = lane & 15;
+  const int fk8 = (lane >> 4) * 8;
+  const int d8 = D / 8;
+  for (int64_t b = (int64_t)blockIdx.x * waves + wave; b < B;
+       b += (int64_t)gridDim.x * waves) {
+    for (int t = lane; t < F * d8; t += 64) {
+      const int fr = t / d8, c8 = (t % d8) * 8;
+      bf16x8i val;
+      if (fr == 0) {
+        val = *(const bf16x8i*)&x[b * D + c8];
+      } else {
+        const __half* src = base + ((int64_t)(fr - 1) * B + b) * D + c8;
+#pragma unroll
+        for (int k = 0; k < 8; ++k) val[k] = if2bf(__half2float(src[k]));
+      }
+      *(bf16x8i*)&v[fr * LDK + c8] = val;
+    }
+    short* dst = out + b * P;
+    for (int ti = 0; ti < T; ++ti) {
+      for (int tj = 0; tj <= ti; ++tj) {
+        f32x4i acc = {};
+        for (int k = 0; k < Dk; k += 32) {
+          const bf16x8i a = *(const bf16x8i*)&v[(ti * 16 + fi) * LDK + k + fk8];
+          const bf16x8i c = *(const bf16x8i*)&v[(tj * 16 + fi) * LDK + k + fk8];
+          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, c, acc, 0, 0, 0);
+        }
+        const int j = tj * 16 + fi;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int i = ti * 16 + (lane >> 4) * 4 + r;
+          if (i < F && j < i) dst[i * (i - 1) / 2 + j] = if2bf(acc[r]);
+        }
+      }
+    }
+  }
+}
+
+__global__ __launch_bounds__(256) void interact_bwd_packed_kernel(
+    const short* __restrict__ g, const short* __restrict__ x,
+    const __half* __restrict__ base, short* __restrict__ dx,
+    __half* __restrict__ dbase, int B, int S, int D, int P) {
+  extern __shared__ short lds[];
+  const int F = S + 1;
+  const int waves = blockDim.x / 64;
+  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  const int Fp = ceil16(F), Fk = ceil32(F), Dp = ceil16(D);
+  const int LDA = Fk + 8;
+  short* A = lds + wave * (Fp + Dp) * LDA;
+  short* vt = A + Fp * LDA;
+  for (int t = lane; t < (Fp + Dp) * LDA / 8; t += 64)
+    *(bf16x8i*)&A[t * 8] = bf16x8i{};
+  const int Tm = Fp / 16, Tn = Dp / 16;
+  const int fi = lane & 15;
+  const int fk8 = (lane >> 4) * 8;
+  const int d8 = D / 8;
+  for (int64_t b = (int64_t)blockIdx.x * waves + wave; b < B;
+       b += (int64_t)gridDim.x * waves) {
+    for (int t = lane; t < F * d8; t += 64) {
+      const int fr = t / d8, c8 = (t % d8) * 8;
+      bf16x8i row;
+      if (fr == 0) {
+        row = *(const bf16x8i*)&x[b * D + c8];
+      } else {
+        const __half* src = base + ((int64_t)(fr - 1) * B + b) * D + c8;
+#pragma unroll
+        for (int k = 0; k < 8; ++k) row[k] = if2bf(__half2float(src[k]));
+      }
+#pragma unroll
+      for (int k = 0; k < 8; ++k) vt[(c8 + k) * LDA + fr] = row[k];
+    }
+    const short* gp = g + b * P;
+    for (int p = lane; p < P; p += 64) {
+      int i = (int)((1.0f + sqrtf(1.0f + 8.0f * (float)p)) * 0.5f);
+      while (i * (i - 1) / 2 > p) --i;
+      while ((i + 1) * i / 2 <= p) ++i;
+      const int j = p - i * (i - 1) / 2;
+      const short gv = gp[p];
+      A[i * LDA + j] = gv;
+      A[j * LDA + i] = gv;
+    }
+    for (int ti = 0; ti < Tm; ++ti) {
+      for (int td = 0; td < Tn; ++td) {
+        f32x4i acc = {};
+        for (int k = 0; k < Fk; k += 32) {
+          const bf16x8i a = *(const bf16x8i*)&A[(ti * 16 + fi) * LDA + k + fk8];
+          const bf16x8i c = *(const bf16x8i*)&vt[(td * 16 + fi) * LDA + k + fk8];
+          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, c, acc, 0, 0, 0);
+        }
+        const int d = td * 16 + fi;
+        if (d < D) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int i = ti * 16 + (lane >> 4) * 4 + r;
+            if (i == 0) {
+              dx[b * D + d] = if2bf(acc[r]);
+            } else if (i < F) {
+              dbase[((int64_t)(i - 1) * B + b) * D + d] = __float2half(acc[r]);
+            }
+          }
+        }
+      }
+    }
+  }
+}
+
 // ------------------------------------------------- scalar fallback (legacy)
 // one wave per sample; V_b staged in LDS; lanes split the P pairs
 __global__ __launch_bounds__(256) void interact_fwd_kernel(
@@ -329,6 +452,62 @@ torch::Tensor interact_bwd(torch::Tensor g, torch::Tensor V) {
   return dV;
 }
 
+torch::Tensor interact_fwd_packed(torch::Tensor x, torch::Tensor base) {
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && x.is_contiguous());
+  TORCH_CHECK(base.scalar_type() == torch::kFloat16 && base.is_contiguous());
+  const int B = (int)x.size(0), D = (int)x.size(1);
+  const int S = (int)(base.size(0) / B);
+  const int F = S + 1;
+  TORCH_CHECK(D % 8 == 0 && base.size(1) == D && base.size(0) == (int64_t)S * B);
+  const int P = F * (F - 1) / 2;
+  auto out = torch::empty(
+      {B, P}, torch::TensorOptions().dtype(torch::kBFloat16).device(x.device()));
+  const int Fp = ceil16(F), Dk = ceil32(D);
+  const int mfma_shorts = Fp * (Dk + 8);
+  const int waves = pick_waves(mfma_shorts);
+  TORCH_CHECK(waves > 0, "interact_fwd_packed: tile exceeds LDS");
+  const int grid = std::min((B + waves - 1) / waves, 8192);
+  hipLaunchKernelGGL(interact_fwd_packed_kernel, dim3(grid), dim3(waves * 64),
+                     waves * mfma_shorts * 2, icur_stream(),
+                     (const short*)x.data_ptr(),
+                     (const __half*)base.data_ptr<at::Half>(),
+                     (short*)out.data_ptr(), B, S, D, P);
+  return out;
+}
+
+std::vector<torch::Tensor> interact_bwd_packed(torch::Tensor g, torch::Tensor x,
+                                               torch::Tensor base) {
+  const int B = (int)x.size(0), D = (int)x.size(1);
+  const int S = (int)(base.size(0) / B);
+  const int F = S + 1;
+  const int P = F * (F - 1) / 2;
+  auto dx = torch::empty_like(x);
+  auto dbase = torch::empty_like(base);
+  const int Fp = ceil16(F), Fk = ceil32(F), Dp = ceil16(D);
+  const int mfma_shorts = (Fp + Dp) * (Fk + 8);
+  const int waves = pick_waves(mfma_shorts);
+  TORCH_CHECK(waves > 0, "interact_bwd_packed: tile exceeds LDS");
+  const int grid = std::min((B + waves - 1) / waves, 8192);
+  auto gc = g.contiguous();
+  hipLaunchKernelGGL(interact_bwd_packed_kernel, dim3(grid), dim3(waves * 64),
+                     waves * mfma_shorts * 2, icur_stream(),
+                     (const short*)gc.data_ptr(),
+                     (const short*)x.data_ptr(),
+                     (const __half*)base.data_ptr<at::Half>(),
+                     (short*)dx.data_ptr(), (__half*)dbase.data_ptr<at::Half>(),
+                     B, S, D, P);
+  return {dx, dbase};
+}
+
+// packed variants have no scalar fallback: MFMA images must fit
+bool interact_packed_feasible(int64_t F, int64_t D) {
+  if (D % 8 != 0) return false;
+  const int Fp = ceil16((int)F), Fk = ceil32((int)F), Dp = ceil16((int)D),
+            Dk = ceil32((int)D);
+  return pick_waves(Fp * (Dk + 8)) > 0 &&
+         pick_waves((Fp + Dp) * (Fk + 8)) > 0;
+}
+
 bool interact_feasible(int64_t F, int64_t D) {
   if (D % 8 != 0) return false;
   const int Fp = ceil16((int)F), Fk = ceil32((int)F), Dp = ceil16((int)D),
@@ -346,4 +525,10 @@ void init_interact(pybind11::module_& m) {
   m.def("interact_bwd", &interact_bwd, "interaction backward (dV)");
   m.def("interact_feasible", &interact_feasible,
         "true if a fused interaction kernel exists for (F, D)");
+  m.def("interact_packed_feasible", &interact_packed_feasible,
+        "true if the packed-input MFMA interaction fits (F, D)");
+  m.def("interact_fwd_packed", &interact_fwd_packed,
+        "interaction from (x bf16 [B,D], slot-major f16 base) directly");
+  m.def("interact_bwd_packed", &interact_bwd_packed,
+        "packed interaction backward -> (dx, dbase f16)");
 }

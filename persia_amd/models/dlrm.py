@@ -41,6 +41,29 @@ class _InteractFn(torch.autograd.Function):
         return native().interact_bwd(g.to(torch.bfloat16).contiguous(), v)
 
 
+class _InteractPackedFn(torch.autograd.Function):
+    """Packed-input interaction: consumes the bottom-MLP output x (bf16
+    [B, D]) and the engine's slot-major f16 sum base ([S*B, D]) DIRECTLY —
+    no [B, F, D] cat/permute/cast materialization on either pass."""
+
+    @staticmethod
+    def forward(ctx, x, base):
+        from persia_amd.ops import native
+
+        x = x.contiguous()
+        base = base.contiguous()
+        ctx.save_for_backward(x, base)
+        return native().interact_fwd_packed(x, base)
+
+    @staticmethod
+    def backward(ctx, g):
+        from persia_amd.ops import native
+
+        x, base = ctx.saved_tensors
+        dx, dbase = native().interact_bwd_packed(g, x, base)
+        return dx, dbase
+
+
 class DotInteraction(nn.Module):
     """Pairwise dot products of the (num_slots+1) feature vectors, strict
     lower triangle in torch.tril_indices order."""
@@ -98,9 +121,22 @@ class DLRM(nn.Module):
         dense = dense.to(dt)
         x = self.bottom(dense)  # [B, D]
         if torch.is_tensor(embedding_tensors):
+            B = x.shape[0]
+            if (
+                x.is_cuda
+                and x.dtype == torch.bfloat16
+                and embedding_tensors.dtype == torch.float16
+                and self.dim % 8 == 0
+            ):
+                from persia_amd.ops import native, native_available
+
+                if native_available() and native().interact_packed_feasible(
+                    self.num_sparse + 1, self.dim
+                ):
+                    inter = _InteractPackedFn.apply(x, embedding_tensors)
+                    return self.top(torch.cat([x, inter], dim=1)).squeeze(1)
             # packed slot-major [S*B, D] (the engine's fused sum output) —
             # ONE reshape instead of a 26-way stack
-            B = x.shape[0]
             emb = (
                 embedding_tensors.view(self.num_sparse, B, self.dim)
                 .permute(1, 0, 2)

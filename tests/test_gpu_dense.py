@@ -129,6 +129,41 @@ def test_interaction_matches_torch():
         )
 
 
+def test_interaction_packed_matches_reference():
+    """Packed-input variant (x bf16 + slot-major f16 base) == the cat/cast
+    construction it replaces."""
+    from persia_amd.ops import native
+
+    C = native()
+    torch.manual_seed(9)
+    for B, S, D in [(64, 26, 128), (512, 64, 8), (130, 12, 64)]:
+        F = S + 1
+        x = (torch.randn(B, D, device=_dev()) * 0.5).to(torch.bfloat16).contiguous()
+        base = (torch.randn(S * B, D, device=_dev()) * 0.5).to(torch.float16).contiguous()
+        out = C.interact_fwd_packed(x, base)
+        emb = base.view(S, B, D).permute(1, 0, 2).float()
+        V = torch.cat([x.float().unsqueeze(1), emb], dim=1)
+        prod = torch.bmm(V, V.transpose(1, 2))
+        li, lj = torch.tril_indices(F, F, offset=-1, device=_dev())
+        ref = prod[:, li, lj]
+        assert torch.allclose(out.float(), ref, atol=0.2, rtol=0.02), (
+            f"{B}x{S}x{D} fwd max err {(out.float() - ref).abs().max()}"
+        )
+        g = torch.randn_like(ref).to(torch.bfloat16)
+        dx, dbase = C.interact_bwd_packed(g.contiguous(), x, base)
+        V2 = V.detach().requires_grad_(True)
+        prod2 = torch.bmm(V2, V2.transpose(1, 2))
+        prod2[:, li, lj].backward(g.float())
+        ref_dx = V2.grad[:, 0]
+        ref_dbase = V2.grad[:, 1:].permute(1, 0, 2).reshape(S * B, D)
+        assert torch.allclose(dx.float(), ref_dx, atol=0.3, rtol=0.02), (
+            f"{B}x{S}x{D} dx max err {(dx.float() - ref_dx).abs().max()}"
+        )
+        assert torch.allclose(dbase.float(), ref_dbase, atol=0.3, rtol=0.02), (
+            f"{B}x{S}x{D} dbase max err {(dbase.float() - ref_dbase).abs().max()}"
+        )
+
+
 def test_interaction_autograd_in_dlrm_path():
     from persia_amd.models.dlrm import DotInteraction
 
