@@ -11,10 +11,50 @@ import torch
 import torch.nn as nn
 
 
+class _BlasLinearFn(torch.autograd.Function):
+    """nn.Linear math with the bias gradient on the coalesced split-M kernel
+    (csrc/dense.hip bias_grad) instead of torch's generic strided reduce."""
+
+    @staticmethod
+    def forward(ctx, x, w, b):
+        ctx.save_for_backward(x, w)
+        return torch.nn.functional.linear(x, w, b)
+
+    @staticmethod
+    def backward(ctx, g):
+        from persia_amd.ops import native
+
+        x, w = ctx.saved_tensors
+        g = g.contiguous()
+        dx = g @ w
+        dw = g.t() @ x
+        db = native().bias_grad(g).to(w.dtype)
+        return dx, dw, db
+
+
+class PALinear(nn.Linear):
+    """nn.Linear drop-in (same parameters/state_dict); routes the backward
+    bias reduction to the native kernel on GPU.  Gated by PA_BLAS_LINEAR=1."""
+
+    def forward(self, x):
+        import os
+
+        if (
+            x.is_cuda and x.dim() == 2 and self.bias is not None
+            and x.dtype == self.weight.dtype and x.dtype == torch.bfloat16
+            and os.environ.get("PA_BLAS_LINEAR", "0") == "1"
+        ):
+            from persia_amd.ops import native_available
+
+            if native_available():
+                return _BlasLinearFn.apply(x, self.weight, self.bias)
+        return super().forward(x)
+
+
 def _mlp(sizes: List[int], last_relu: bool = False) -> nn.Sequential:
     layers: List[nn.Module] = []
     for i in range(len(sizes) - 1):
-        layers.append(nn.Linear(sizes[i], sizes[i + 1]))
+        layers.append(PALinear(sizes[i], sizes[i + 1]))
         if i < len(sizes) - 2 or last_relu:
             layers.append(nn.ReLU())
     return nn.Sequential(*layers)
