@@ -670,20 +670,25 @@ __global__ void scatter_update_kernel(
     }
     if (!ok) continue;
     float* row = arena + (int64_t)slot * row_width;
+    // nontemporal arena traffic: each row is touched exactly once per step
+    // and this kernel overlaps the dense graph on another stream — keeping
+    // ~0.5 GB/step of one-shot data out of L2 leaves it to the GEMMs
     if (opt == 0) {  // SGD
       for (int t = 0, c = c0; c < dim; c += st, ++t) {
-        float w2 = row[c] - p0 * (acc[t] + p1 * row[c]);
+        const float w0 = __builtin_nontemporal_load(&row[c]);
+        float w2 = w0 - p0 * (acc[t] + p1 * w0);
         if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
-        row[c] = w2;
+        __builtin_nontemporal_store(w2, &row[c]);
       }
     } else if (opt == 1) {  // Adagrad
       if (p3 > 0.5f) {
         const float a0 = row[dim];
         float gsq = 0.0f;
         for (int t = 0, c = c0; c < dim; c += st, ++t) {
-          float w2 = row[c] - p0 * acc[t] * rsqrtf(a0 + p2);
+          const float w0 = __builtin_nontemporal_load(&row[c]);
+          float w2 = w0 - p0 * acc[t] * rsqrtf(a0 + p2);
           if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
-          row[c] = w2;
+          __builtin_nontemporal_store(w2, &row[c]);
           gsq += acc[t] * acc[t];
         }
         for (int off = sg / 2; off > 0; off >>= 1)
@@ -691,24 +696,27 @@ __global__ void scatter_update_kernel(
         if (c0 == 0) row[dim] = a0 * p1 + gsq / (float)dim;
       } else {
         for (int t = 0, c = c0; c < dim; c += st, ++t) {
-          const float a0 = row[dim + c];
-          float w2 = row[c] - p0 * acc[t] * rsqrtf(a0 + p2);
+          const float a0 = __builtin_nontemporal_load(&row[dim + c]);
+          const float w0 = __builtin_nontemporal_load(&row[c]);
+          float w2 = w0 - p0 * acc[t] * rsqrtf(a0 + p2);
           if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
-          row[c] = w2;
-          row[dim + c] = a0 * p1 + acc[t] * acc[t];
+          __builtin_nontemporal_store(w2, &row[c]);
+          __builtin_nontemporal_store(a0 * p1 + acc[t] * acc[t], &row[dim + c]);
         }
       }
     } else {  // Adam
       const float om1 = 1.0f - p1, om2 = 1.0f - p2;
       const float c1 = 1.0f / (1.0f - b1_power), c2 = 1.0f / (1.0f - b2_power);
       for (int t = 0, c = c0; c < dim; c += st, ++t) {
-        const float m = p1 * row[dim + c] + om1 * acc[t];
-        const float v = p2 * row[2 * dim + c] + om2 * acc[t] * acc[t];
-        float w2 = row[c] - p0 * (m * c1) / (p3 + sqrtf(v * c2));
+        const float m = p1 * __builtin_nontemporal_load(&row[dim + c]) + om1 * acc[t];
+        const float v =
+            p2 * __builtin_nontemporal_load(&row[2 * dim + c]) + om2 * acc[t] * acc[t];
+        const float w0 = __builtin_nontemporal_load(&row[c]);
+        float w2 = w0 - p0 * (m * c1) / (p3 + sqrtf(v * c2));
         if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
-        row[c] = w2;
-        row[dim + c] = m;
-        row[2 * dim + c] = v;
+        __builtin_nontemporal_store(w2, &row[c]);
+        __builtin_nontemporal_store(m, &row[dim + c]);
+        __builtin_nontemporal_store(v, &row[2 * dim + c]);
       }
     }
   }
