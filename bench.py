@@ -268,14 +268,6 @@ def main():
     timing = os.environ.get("PA_BENCH_TIMING", "0") == "1"
     tstats = {"get": 0.0, "copy": 0.0, "replay": 0.0, "apply": 0.0, "n": 0}
 
-    # sparse update on a side stream: the graphed step keeps its gradients in
-    # the STATIC buffer, so the scatter+optimizer kernel can overlap the next
-    # iteration's input copies / early dense work; the only ordering
-    # constraint is update-done before the next replay zeroes the grad buffer
-    upd_stream = torch.cuda.Stream() if (graph is not None and use_gpu) else None
-    ev_replay = torch.cuda.Event() if upd_stream is not None else None
-    ev_update = torch.cuda.Event() if upd_stream is not None else None
-
     def train_step(tb):
         if graph is not None:
             if timing:
@@ -286,18 +278,10 @@ def main():
                 static["label"].copy_(tb.label_tensors[0], non_blocking=True)
             if timing:
                 t1 = time.perf_counter()
-            torch.cuda.current_stream().wait_event(ev_update)
             graph.replay()
-            ev_replay.record()
             if timing:
                 t2 = time.perf_counter()
-            upd_stream.wait_event(ev_replay)
-            with torch.cuda.stream(upd_stream):
-                tb.record_stream(upd_stream)
-                engine.apply_gradients_base(
-                    tb, sum_base_grads=[static["base"].grad]
-                )
-            ev_update.record(upd_stream)
+            engine.apply_gradients_base(tb, sum_base_grads=[static["base"].grad])
             pipeline.release_permit()
             if timing:
                 t3 = time.perf_counter()
