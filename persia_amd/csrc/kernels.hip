@@ -722,6 +722,201 @@ __global__ void scatter_update_kernel(
   }
 }
 
+// Dual-key variant for full-wave dims (64 <= dim <= 256): one wave updates
+// TWO unique keys with interleaved loads — the single-key kernel is
+// latency-bound on its probe -> grads -> arena dependent chain (random 1-2 KB
+// row granules), so doubling the independent in-flight accesses per wave
+// buys memory-level parallelism the scheduler can't extract across
+// grid-stride iterations (ballots/shfl block software pipelining).
+__global__ void scatter_update2_kernel(
+    ull* __restrict__ table_keys, unsigned* __restrict__ ticks,
+    float* __restrict__ arena, const ull* __restrict__ uniq,
+    const __half* __restrict__ grads, const int64_t* __restrict__ perm,
+    const int64_t* __restrict__ ustarts, const int64_t* __restrict__ seg_id,
+    const float* __restrict__ seg_scale, int64_t n, int dim, int row_width,
+    int64_t n_buckets, int opt, float p0, float p1, float p2, float p3,
+    float b1_power, float b2_power, float weight_bound,
+    int* __restrict__ skipped, const long long* __restrict__ n_dev) {
+  if (n_dev) n = *n_dev;
+  const int wave = threadIdx.x / PA_WAVE;
+  const int lane = threadIdx.x % PA_WAVE;
+  const int waves_per_block = blockDim.x / PA_WAVE;
+  const int64_t mask = n_buckets - 1;
+  constexpr int WINDOW = PA_PROBE_BUCKETS * PA_BUCKET_SIZE;  // 32
+  const int half = lane >> 5;   // probe phase: lanes 0-31 key A, 32-63 key B
+  const int widx = lane & 31;
+  const int64_t n_waveitems = (n + 1) / 2;
+  for (int64_t w = (int64_t)blockIdx.x * waves_per_block + wave;
+       w < n_waveitems; w += (int64_t)gridDim.x * waves_per_block) {
+    const int64_t u = w * 2 + half;
+    const bool active = u < n;
+    const ull k = active ? uniq[u] : 0;
+    long long slot = -1;
+    if (active) {
+      const int64_t b = (int64_t)(k & (ull)mask);
+      const int p = widx / PA_BUCKET_SIZE;
+      const int s = widx % PA_BUCKET_SIZE;
+      const int64_t j = ((b + p) & mask) * PA_BUCKET_SIZE + s;
+      if (table_keys[j] == k) slot = j;
+    }
+    const unsigned long long found = __ballot(slot >= 0);
+    const unsigned fA = (unsigned)(found & 0xFFFFFFFFull);
+    const unsigned fB = (unsigned)(found >> 32);
+    bool okA = (w * 2 + 0) < n, okB = (w * 2 + 1) < n;
+    if (okA && fA == 0) { if (lane == 0) atomicAdd(&skipped[0], 1); okA = false; }
+    if (okB && fB == 0) { if (lane == 0) atomicAdd(&skipped[0], 1); okB = false; }
+    const long long slotA = okA ? __shfl(slot, __ffs(fA) - 1) : -1;
+    const long long slotB = okB ? __shfl(slot, 32 + __ffs(fB) - 1) : -1;
+    // gradient accumulation (whole wave per key, interleaved when both are
+    // single-position — the overwhelmingly common case)
+    float accA[4] = {}, accB[4] = {};
+    const int64_t loA = ustarts[w * 2], hiA = okA ? ustarts[w * 2 + 1] : loA;
+    const int64_t loB = okB ? ustarts[w * 2 + 1] : 0;
+    const int64_t hiB = okB ? ustarts[w * 2 + 2] : loB;
+    if (okA && okB && hiA - loA == 1 && hiB - loB == 1) {
+      const int64_t sA = seg_id[perm[loA]], sB = seg_id[perm[loB]];
+      const float scA = (sA >= 0) ? (seg_scale ? seg_scale[sA] : 1.0f) : 0.0f;
+      const float scB = (sB >= 0) ? (seg_scale ? seg_scale[sB] : 1.0f) : 0.0f;
+      const __half* gA = grads + (sA >= 0 ? sA : 0) * dim;
+      const __half* gB = grads + (sB >= 0 ? sB : 0) * dim;
+      for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
+        accA[t] = scA * __half2float(gA[c]);
+        accB[t] = scB * __half2float(gB[c]);
+      }
+    } else {
+      if (okA)
+        for (int64_t p = loA; p < hiA; ++p) {
+          const int64_t s = seg_id[perm[p]];
+          if (s < 0) continue;
+          const float sc = seg_scale ? seg_scale[s] : 1.0f;
+          if (sc == 0.0f) continue;
+          const __half* g = grads + s * dim;
+          for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t)
+            accA[t] += __half2float(g[c]) * sc;
+        }
+      if (okB)
+        for (int64_t p = loB; p < hiB; ++p) {
+          const int64_t s = seg_id[perm[p]];
+          if (s < 0) continue;
+          const float sc = seg_scale ? seg_scale[s] : 1.0f;
+          if (sc == 0.0f) continue;
+          const __half* g = grads + s * dim;
+          for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t)
+            accB[t] += __half2float(g[c]) * sc;
+        }
+    }
+    bool nanA = false, nanB = false;
+    for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
+      nanA |= isnan(accA[t]);
+      nanB |= isnan(accB[t]);
+    }
+    if (okA && __ballot(nanA) != 0) {
+      if (lane == 0) atomicAdd(&skipped[1], 1);
+      okA = false;
+    }
+    if (okB && __ballot(nanB) != 0) {
+      if (lane == 0) atomicAdd(&skipped[1], 1);
+      okB = false;
+    }
+    if (!okA && !okB) continue;
+    float* rowA = okA ? arena + (int64_t)slotA * row_width : nullptr;
+    float* rowB = okB ? arena + (int64_t)slotB * row_width : nullptr;
+    if (opt == 0) {  // SGD
+      for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
+        const float wA = okA ? __builtin_nontemporal_load(&rowA[c]) : 0.0f;
+        const float wB = okB ? __builtin_nontemporal_load(&rowB[c]) : 0.0f;
+        if (okA) {
+          float w2 = wA - p0 * (accA[t] + p1 * wA);
+          if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
+          __builtin_nontemporal_store(w2, &rowA[c]);
+        }
+        if (okB) {
+          float w2 = wB - p0 * (accB[t] + p1 * wB);
+          if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
+          __builtin_nontemporal_store(w2, &rowB[c]);
+        }
+      }
+    } else if (opt == 1) {  // Adagrad
+      if (p3 > 0.5f) {  // vectorwise-shared accumulator
+        const float a0A = okA ? rowA[dim] : 0.0f;
+        const float a0B = okB ? rowB[dim] : 0.0f;
+        float gsqA = 0.0f, gsqB = 0.0f;
+        for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
+          const float wA = okA ? __builtin_nontemporal_load(&rowA[c]) : 0.0f;
+          const float wB = okB ? __builtin_nontemporal_load(&rowB[c]) : 0.0f;
+          if (okA) {
+            float w2 = wA - p0 * accA[t] * rsqrtf(a0A + p2);
+            if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
+            __builtin_nontemporal_store(w2, &rowA[c]);
+            gsqA += accA[t] * accA[t];
+          }
+          if (okB) {
+            float w2 = wB - p0 * accB[t] * rsqrtf(a0B + p2);
+            if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
+            __builtin_nontemporal_store(w2, &rowB[c]);
+            gsqB += accB[t] * accB[t];
+          }
+        }
+#pragma unroll
+        for (int off = PA_WAVE / 2; off > 0; off >>= 1) {
+          gsqA += __shfl_down(gsqA, off);
+          gsqB += __shfl_down(gsqB, off);
+        }
+        if (lane == 0) {
+          if (okA) rowA[dim] = a0A * p1 + gsqA / (float)dim;
+          if (okB) rowB[dim] = a0B * p1 + gsqB / (float)dim;
+        }
+      } else {
+        for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
+          const float aA = okA ? __builtin_nontemporal_load(&rowA[dim + c]) : 0.0f;
+          const float aB = okB ? __builtin_nontemporal_load(&rowB[dim + c]) : 0.0f;
+          const float wA = okA ? __builtin_nontemporal_load(&rowA[c]) : 0.0f;
+          const float wB = okB ? __builtin_nontemporal_load(&rowB[c]) : 0.0f;
+          if (okA) {
+            float w2 = wA - p0 * accA[t] * rsqrtf(aA + p2);
+            if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
+            __builtin_nontemporal_store(w2, &rowA[c]);
+            __builtin_nontemporal_store(aA * p1 + accA[t] * accA[t], &rowA[dim + c]);
+          }
+          if (okB) {
+            float w2 = wB - p0 * accB[t] * rsqrtf(aB + p2);
+            if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
+            __builtin_nontemporal_store(w2, &rowB[c]);
+            __builtin_nontemporal_store(aB * p1 + accB[t] * accB[t], &rowB[dim + c]);
+          }
+        }
+      }
+    } else {  // Adam
+      const float om1 = 1.0f - p1, om2 = 1.0f - p2;
+      const float c1 = 1.0f / (1.0f - b1_power), c2 = 1.0f / (1.0f - b2_power);
+      for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
+        if (okA) {
+          const float m = p1 * __builtin_nontemporal_load(&rowA[dim + c]) + om1 * accA[t];
+          const float v =
+              p2 * __builtin_nontemporal_load(&rowA[2 * dim + c]) + om2 * accA[t] * accA[t];
+          const float w0 = __builtin_nontemporal_load(&rowA[c]);
+          float w2 = w0 - p0 * (m * c1) / (p3 + sqrtf(v * c2));
+          if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
+          __builtin_nontemporal_store(w2, &rowA[c]);
+          __builtin_nontemporal_store(m, &rowA[dim + c]);
+          __builtin_nontemporal_store(v, &rowA[2 * dim + c]);
+        }
+        if (okB) {
+          const float m = p1 * __builtin_nontemporal_load(&rowB[dim + c]) + om1 * accB[t];
+          const float v =
+              p2 * __builtin_nontemporal_load(&rowB[2 * dim + c]) + om2 * accB[t] * accB[t];
+          const float w0 = __builtin_nontemporal_load(&rowB[c]);
+          float w2 = w0 - p0 * (m * c1) / (p3 + sqrtf(v * c2));
+          if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
+          __builtin_nontemporal_store(w2, &rowB[c]);
+          __builtin_nontemporal_store(m, &rowB[dim + c]);
+          __builtin_nontemporal_store(v, &rowB[2 * dim + c]);
+        }
+      }
+    }
+  }
+}
+
 inline int n_blocks_for(int64_t work_items, int per_block) {
   int64_t b = (work_items + per_block - 1) / per_block;
   // >> 256 CUs needed to fill the chip; cap and grid-stride beyond
@@ -999,6 +1194,28 @@ void scatter_update(torch::Tensor table_keys, torch::Tensor ticks,
   const int row_width = (int)arena.size(1);
   const float* scale_ptr =
       seg_scale.numel() ? seg_scale.data_ptr<float>() : nullptr;
+  // full-wave dims: dual-key variant (2x memory-level parallelism); small
+  // dividing dims keep the sub-wave-packed single kernel (more keys/wave)
+  const bool dual = !(dim < PA_WAVE && (PA_WAVE % dim) == 0) && dim <= 256;
+  if (dual) {
+    hipLaunchKernelGGL(scatter_update2_kernel,
+                       dim3(n_blocks_for((n + 1) / 2, 4)), dim3(256), 0,
+                       cur_stream(), (ull*)table_keys.data_ptr<int64_t>(),
+                       (unsigned*)ticks.data_ptr<int32_t>(),
+                       arena.data_ptr<float>(),
+                       (const ull*)uniq.data_ptr<int64_t>(),
+                       (const __half*)grads.data_ptr<at::Half>(),
+                       perm.data_ptr<int64_t>(), ustarts.data_ptr<int64_t>(),
+                       seg_id.data_ptr<int64_t>(), scale_ptr, n, (int)dim,
+                       row_width, n_buckets, (int)opt, (float)params[0],
+                       (float)params[1], (float)params[2], (float)params[3],
+                       (float)b1_power, (float)b2_power, (float)weight_bound,
+                       skipped.data_ptr<int32_t>(),
+                       u_count.numel()
+                           ? (const long long*)u_count.data_ptr<int64_t>()
+                           : nullptr);
+    return;
+  }
   hipLaunchKernelGGL(scatter_update_kernel, dim3(n_blocks_for(n, 4)),
                      dim3(256), 0, cur_stream(),
                      (ull*)table_keys.data_ptr<int64_t>(),

@@ -373,9 +373,11 @@ def test_engine_gpu_matches_cpu_engine():
         gpu_eng.apply_gradients(tb_g, gg)
 
 
-def test_fused_scatter_update_matches_cpu():
+@pytest.mark.parametrize("dim", [16, 128])
+def test_fused_scatter_update_matches_cpu(dim):
     """apply_gradients_base single-GPU fast path (fused scatter_update
-    kernel) vs the CPU oracle engine."""
+    kernel; dim=16 exercises the sub-wave-packed kernel, dim=128 the
+    dual-key full-wave kernel) vs the CPU oracle engine."""
     from persia_amd.core.comm import DistContext
     from persia_amd.core.engine import EmbeddingEngine
     from persia_amd.core.schema import EmbeddingSchema, GlobalConfig, SlotConfig
@@ -386,7 +388,7 @@ def test_fused_scatter_update_matches_cpu():
     def mk(device):
         return EmbeddingEngine(
             schema=EmbeddingSchema(
-                slots={f"f{i}": SlotConfig(name=f"f{i}", dim=16) for i in range(3)}
+                slots={f"f{i}": SlotConfig(name=f"f{i}", dim=dim) for i in range(3)}
             ),
             hyper=EmbeddingConfig(emb_initialization=(-0.5, 0.5)),
             optimizer=Adagrad(lr=0.1),
@@ -410,7 +412,7 @@ def test_fused_scatter_update_matches_cpu():
     for step in range(3):
         tb_c = cpu.process_batch(batch(step))
         tb_g = gpu.process_batch(batch(step))
-        g = torch.randn(3 * 32, 16).to(torch.float16)
+        g = torch.randn(3 * 32, dim).to(torch.float16)
         tb_c.enable_training_views()
         tb_g.enable_training_views()
         tb_c._groups[0].sum_base.grad = g.clone()
