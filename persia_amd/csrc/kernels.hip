@@ -266,6 +266,73 @@ __global__ void init_scatter_kernel(float* __restrict__ arena,
   }
 }
 
+// Dual-key init/scatter for full-wave dims (see scatter_update2_kernel for
+// the rationale: the slots->arena->sums chain is latency-bound on random
+// row granules; two independent keys per wave double the in-flight loads).
+// The rare fresh-init path falls back to per-key sequential work.
+__global__ void init_scatter2_kernel(float* __restrict__ arena,
+                                     const ull* __restrict__ query,
+                                     const long long* __restrict__ slots,
+                                     const int* __restrict__ is_new,
+                                     const int64_t* __restrict__ perm,
+                                     const int64_t* __restrict__ ustarts,
+                                     __half* __restrict__ sums, int64_t n,
+                                     int dim, int row_width, double lo,
+                                     double hi, float state_init,
+                                     const long long* __restrict__ n_dev) {
+  if (n_dev) n = *n_dev;
+  const int wave = threadIdx.x / PA_WAVE;
+  const int lane = threadIdx.x % PA_WAVE;
+  const int waves_per_block = blockDim.x / PA_WAVE;
+  const int64_t n_waveitems = (n + 1) / 2;
+  for (int64_t w = (int64_t)blockIdx.x * waves_per_block + wave;
+       w < n_waveitems; w += (int64_t)gridDim.x * waves_per_block) {
+    const int64_t iA = w * 2, iB = w * 2 + 1;
+    const bool okB0 = iB < n;
+    const long long slotA = slots[iA];
+    const long long slotB = okB0 ? slots[iB] : -1;
+    // fresh rows: seeded init first (rare in steady state)
+    if (is_new[iA] && slotA >= 0) {
+      float* row = arena + (int64_t)slotA * row_width;
+      const uint64_t seed = pa_init_seed(pa_splitmix64_inv((uint64_t)query[iA]));
+      for (int c = lane; c < dim; c += PA_WAVE) row[c] = pa_init_val(seed, c, lo, hi);
+      for (int c = dim + lane; c < row_width; c += PA_WAVE) row[c] = state_init;
+    }
+    if (okB0 && is_new[iB] && slotB >= 0) {
+      float* row = arena + (int64_t)slotB * row_width;
+      const uint64_t seed = pa_init_seed(pa_splitmix64_inv((uint64_t)query[iB]));
+      for (int c = lane; c < dim; c += PA_WAVE) row[c] = pa_init_val(seed, c, lo, hi);
+      for (int c = dim + lane; c < row_width; c += PA_WAVE) row[c] = state_init;
+    }
+    const int64_t loA = ustarts[iA], hiA = ustarts[iA + 1];
+    const int64_t loB = okB0 ? ustarts[iB] : 0;
+    const int64_t hiB = okB0 ? ustarts[iB + 1] : 0;
+    const float* rowA = slotA >= 0 ? arena + (int64_t)slotA * row_width : nullptr;
+    const float* rowB = slotB >= 0 ? arena + (int64_t)slotB * row_width : nullptr;
+    if (hiA - loA == 1 && hiB - loB == 1) {  // common single-position case
+      __half* dstA = sums + perm[loA] * dim;
+      __half* dstB = sums + perm[loB] * dim;
+      for (int c = lane; c < dim; c += PA_WAVE) {
+        const float vA = rowA ? rowA[c] : 0.0f;
+        const float vB = rowB ? rowB[c] : 0.0f;
+        dstA[c] = __float2half(vA);
+        dstB[c] = __float2half(vB);
+      }
+    } else {
+      for (int64_t p = loA; p < hiA; ++p) {
+        __half* dst = sums + perm[p] * dim;
+        for (int c = lane; c < dim; c += PA_WAVE)
+          dst[c] = __float2half(rowA ? rowA[c] : 0.0f);
+      }
+      for (int64_t p = loB; p < hiB; ++p) {
+        __half* dst = sums + perm[p] * dim;
+        for (int c = lane; c < dim; c += PA_WAVE)
+          dst[c] = __float2half(rowB ? rowB[c] : 0.0f);
+      }
+    }
+  }
+}
+
 // ------------------------------------------------------------- sparse update
 
 // one wave per key: probe (lane-parallel over the 32-slot window) + fused
@@ -1043,14 +1110,27 @@ void store_lookup_sums(torch::Tensor table_keys, torch::Tensor ticks,
                      (long long*)slots.data_ptr<int64_t>(),
                      is_new.data_ptr<int32_t>(), nullptr, nullptr, nullptr,
                      n_dev);
-  hipLaunchKernelGGL(init_scatter_kernel, dim3(n_blocks_for(n, 4)), dim3(256),
-                     0, st, arena.data_ptr<float>(),
-                     (const ull*)query.data_ptr<int64_t>(),
-                     (const long long*)slots.data_ptr<int64_t>(),
-                     is_new.data_ptr<int32_t>(), perm.data_ptr<int64_t>(),
-                     ustarts.data_ptr<int64_t>(),
-                     (__half*)sums.data_ptr<at::Half>(), n, (int)dim,
-                     row_width, lo, hi, (float)state_init, n_dev);
+  const bool dual = !(dim < PA_WAVE && (PA_WAVE % dim) == 0);
+  if (dual) {
+    hipLaunchKernelGGL(init_scatter2_kernel,
+                       dim3(n_blocks_for((n + 1) / 2, 4)), dim3(256), 0, st,
+                       arena.data_ptr<float>(),
+                       (const ull*)query.data_ptr<int64_t>(),
+                       (const long long*)slots.data_ptr<int64_t>(),
+                       is_new.data_ptr<int32_t>(), perm.data_ptr<int64_t>(),
+                       ustarts.data_ptr<int64_t>(),
+                       (__half*)sums.data_ptr<at::Half>(), n, (int)dim,
+                       row_width, lo, hi, (float)state_init, n_dev);
+  } else {
+    hipLaunchKernelGGL(init_scatter_kernel, dim3(n_blocks_for(n, 4)),
+                       dim3(256), 0, st, arena.data_ptr<float>(),
+                       (const ull*)query.data_ptr<int64_t>(),
+                       (const long long*)slots.data_ptr<int64_t>(),
+                       is_new.data_ptr<int32_t>(), perm.data_ptr<int64_t>(),
+                       ustarts.data_ptr<int64_t>(),
+                       (__half*)sums.data_ptr<at::Half>(), n, (int)dim,
+                       row_width, lo, hi, (float)state_init, n_dev);
+  }
 }
 
 torch::Tensor store_probe(torch::Tensor table_keys, torch::Tensor ticks,
