@@ -984,6 +984,192 @@ __global__ void scatter_update2_kernel(
   }
 }
 
+// Quad-key variant for dim <= 128 (acc fits 2 cols/lane/key): 4 independent
+// probe->grads->arena chains per wave.  Probe: 16 lanes per key, each lane
+// scans 2 of the 32 window slots.
+__global__ void scatter_update4_kernel(
+    ull* __restrict__ table_keys, unsigned* __restrict__ ticks,
+    float* __restrict__ arena, const ull* __restrict__ uniq,
+    const __half* __restrict__ grads, const int64_t* __restrict__ perm,
+    const int64_t* __restrict__ ustarts, const int64_t* __restrict__ seg_id,
+    const float* __restrict__ seg_scale, int64_t n, int dim, int row_width,
+    int64_t n_buckets, int opt, float p0, float p1, float p2, float p3,
+    float b1_power, float b2_power, float weight_bound,
+    int* __restrict__ skipped, const long long* __restrict__ n_dev) {
+  if (n_dev) n = *n_dev;
+  const int wave = threadIdx.x / PA_WAVE;
+  const int lane = threadIdx.x % PA_WAVE;
+  const int waves_per_block = blockDim.x / PA_WAVE;
+  const int64_t mask = n_buckets - 1;
+  const int kq = lane >> 4;   // probe phase: 16 lanes per key
+  const int widx = lane & 15;
+  const int64_t n_waveitems = (n + 3) / 4;
+  for (int64_t w = (int64_t)blockIdx.x * waves_per_block + wave;
+       w < n_waveitems; w += (int64_t)gridDim.x * waves_per_block) {
+    const int64_t u = w * 4 + kq;
+    const bool active = u < n;
+    const ull k = active ? uniq[u] : 0;
+    long long slot = -1;
+    if (active) {
+      const int64_t b = (int64_t)(k & (ull)mask);
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        const int wi = widx + half * 16;
+        const int p = wi / PA_BUCKET_SIZE;
+        const int s = wi % PA_BUCKET_SIZE;
+        const int64_t j = ((b + p) & mask) * PA_BUCKET_SIZE + s;
+        if (table_keys[j] == k) slot = j;
+      }
+    }
+    const unsigned long long found = __ballot(slot >= 0);
+    bool ok[4];
+    long long slotk[4];
+    float* rowk[4];
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const unsigned fq = (unsigned)((found >> (16 * q)) & 0xFFFFull);
+      ok[q] = (w * 4 + q) < n;
+      if (ok[q] && fq == 0) {
+        if (lane == 0) atomicAdd(&skipped[0], 1);
+        ok[q] = false;
+      }
+      slotk[q] = ok[q] ? __shfl(slot, 16 * q + __ffs(fq) - 1) : -1;
+      rowk[q] = ok[q] ? arena + (int64_t)slotk[q] * row_width : nullptr;
+    }
+    // gradient accumulation (whole wave per key; interleaved single-position
+    // fast path is the overwhelmingly common case)
+    float acc[4][2] = {};
+    int64_t lop[4], hip[4];
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      lop[q] = ok[q] ? ustarts[w * 4 + q] : 0;
+      hip[q] = ok[q] ? ustarts[w * 4 + q + 1] : 0;
+    }
+    const bool all_single = ok[0] && ok[1] && ok[2] && ok[3] &&
+                            hip[0] - lop[0] == 1 && hip[1] - lop[1] == 1 &&
+                            hip[2] - lop[2] == 1 && hip[3] - lop[3] == 1;
+    if (all_single) {
+      const __half* gq[4];
+      float scq[4];
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const int64_t s = seg_id[perm[lop[q]]];
+        scq[q] = (s >= 0) ? (seg_scale ? seg_scale[s] : 1.0f) : 0.0f;
+        gq[q] = grads + (s >= 0 ? s : 0) * dim;
+      }
+      for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
+#pragma unroll
+        for (int q = 0; q < 4; ++q) acc[q][t] = scq[q] * __half2float(gq[q][c]);
+      }
+    } else {
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        if (!ok[q]) continue;
+        for (int64_t p = lop[q]; p < hip[q]; ++p) {
+          const int64_t s = seg_id[perm[p]];
+          if (s < 0) continue;
+          const float sc = seg_scale ? seg_scale[s] : 1.0f;
+          if (sc == 0.0f) continue;
+          const __half* g = grads + s * dim;
+          for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t)
+            acc[q][t] += __half2float(g[c]) * sc;
+        }
+      }
+    }
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      bool has_nan = false;
+      for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t)
+        has_nan |= isnan(acc[q][t]);
+      if (ok[q] && __ballot(has_nan) != 0) {
+        if (lane == 0) atomicAdd(&skipped[1], 1);
+        ok[q] = false;
+      }
+    }
+    if (!ok[0] && !ok[1] && !ok[2] && !ok[3]) continue;
+    if (opt == 0) {  // SGD
+      for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
+        float w0[4];
+#pragma unroll
+        for (int q = 0; q < 4; ++q)
+          w0[q] = ok[q] ? __builtin_nontemporal_load(&rowk[q][c]) : 0.0f;
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+          if (!ok[q]) continue;
+          float w2 = w0[q] - p0 * (acc[q][t] + p1 * w0[q]);
+          if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
+          __builtin_nontemporal_store(w2, &rowk[q][c]);
+        }
+      }
+    } else if (opt == 1) {  // Adagrad
+      if (p3 > 0.5f) {  // vectorwise-shared accumulator
+        float a0[4], gsq[4] = {};
+#pragma unroll
+        for (int q = 0; q < 4; ++q) a0[q] = ok[q] ? rowk[q][dim] : 0.0f;
+        for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
+          float w0[4];
+#pragma unroll
+          for (int q = 0; q < 4; ++q)
+            w0[q] = ok[q] ? __builtin_nontemporal_load(&rowk[q][c]) : 0.0f;
+#pragma unroll
+          for (int q = 0; q < 4; ++q) {
+            if (!ok[q]) continue;
+            float w2 = w0[q] - p0 * acc[q][t] * rsqrtf(a0[q] + p2);
+            if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
+            __builtin_nontemporal_store(w2, &rowk[q][c]);
+            gsq[q] += acc[q][t] * acc[q][t];
+          }
+        }
+#pragma unroll
+        for (int off = PA_WAVE / 2; off > 0; off >>= 1)
+#pragma unroll
+          for (int q = 0; q < 4; ++q) gsq[q] += __shfl_down(gsq[q], off);
+        if (lane == 0)
+#pragma unroll
+          for (int q = 0; q < 4; ++q)
+            if (ok[q]) rowk[q][dim] = a0[q] * p1 + gsq[q] / (float)dim;
+      } else {
+        for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
+          float w0[4], a0[4];
+#pragma unroll
+          for (int q = 0; q < 4; ++q) {
+            a0[q] = ok[q] ? __builtin_nontemporal_load(&rowk[q][dim + c]) : 0.0f;
+            w0[q] = ok[q] ? __builtin_nontemporal_load(&rowk[q][c]) : 0.0f;
+          }
+#pragma unroll
+          for (int q = 0; q < 4; ++q) {
+            if (!ok[q]) continue;
+            float w2 = w0[q] - p0 * acc[q][t] * rsqrtf(a0[q] + p2);
+            if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
+            __builtin_nontemporal_store(w2, &rowk[q][c]);
+            __builtin_nontemporal_store(a0[q] * p1 + acc[q][t] * acc[q][t],
+                                        &rowk[q][dim + c]);
+          }
+        }
+      }
+    } else {  // Adam
+      const float om1 = 1.0f - p1, om2 = 1.0f - p2;
+      const float c1 = 1.0f / (1.0f - b1_power), c2 = 1.0f / (1.0f - b2_power);
+      for (int t = 0, c = lane; c < dim; c += PA_WAVE, ++t) {
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+          if (!ok[q]) continue;
+          const float m =
+              p1 * __builtin_nontemporal_load(&rowk[q][dim + c]) + om1 * acc[q][t];
+          const float v = p2 * __builtin_nontemporal_load(&rowk[q][2 * dim + c]) +
+                          om2 * acc[q][t] * acc[q][t];
+          const float w0 = __builtin_nontemporal_load(&rowk[q][c]);
+          float w2 = w0 - p0 * (m * c1) / (p3 + sqrtf(v * c2));
+          if (weight_bound > 0.0f) w2 = fminf(fmaxf(w2, -weight_bound), weight_bound);
+          __builtin_nontemporal_store(w2, &rowk[q][c]);
+          __builtin_nontemporal_store(m, &rowk[q][dim + c]);
+          __builtin_nontemporal_store(v, &rowk[q][2 * dim + c]);
+        }
+      }
+    }
+  }
+}
+
 inline int n_blocks_for(int64_t work_items, int per_block) {
   int64_t b = (work_items + per_block - 1) / per_block;
   // >> 256 CUs needed to fill the chip; cap and grid-stride beyond
@@ -1276,7 +1462,27 @@ void scatter_update(torch::Tensor table_keys, torch::Tensor ticks,
       seg_scale.numel() ? seg_scale.data_ptr<float>() : nullptr;
   // full-wave dims: dual-key variant (2x memory-level parallelism); small
   // dividing dims keep the sub-wave-packed single kernel (more keys/wave)
-  const bool dual = !(dim < PA_WAVE && (PA_WAVE % dim) == 0) && dim <= 256;
+  const bool packed_small = (dim < PA_WAVE && (PA_WAVE % dim) == 0);
+  if (!packed_small && dim <= 128) {
+    hipLaunchKernelGGL(scatter_update4_kernel,
+                       dim3(n_blocks_for((n + 3) / 4, 4)), dim3(256), 0,
+                       cur_stream(), (ull*)table_keys.data_ptr<int64_t>(),
+                       (unsigned*)ticks.data_ptr<int32_t>(),
+                       arena.data_ptr<float>(),
+                       (const ull*)uniq.data_ptr<int64_t>(),
+                       (const __half*)grads.data_ptr<at::Half>(),
+                       perm.data_ptr<int64_t>(), ustarts.data_ptr<int64_t>(),
+                       seg_id.data_ptr<int64_t>(), scale_ptr, n, (int)dim,
+                       row_width, n_buckets, (int)opt, (float)params[0],
+                       (float)params[1], (float)params[2], (float)params[3],
+                       (float)b1_power, (float)b2_power, (float)weight_bound,
+                       skipped.data_ptr<int32_t>(),
+                       u_count.numel()
+                           ? (const long long*)u_count.data_ptr<int64_t>()
+                           : nullptr);
+    return;
+  }
+  const bool dual = !packed_small && dim <= 256;
   if (dual) {
     hipLaunchKernelGGL(scatter_update2_kernel,
                        dim3(n_blocks_for((n + 1) / 2, 4)), dim3(256), 0,
