@@ -145,3 +145,20 @@ def test_nan_gradients_skipped():
     eng.apply_gradients(tb, {"a": g, "b": None})
     assert torch.allclose(eng.stores[4].arena, before, equal_nan=True)
     assert eng.nan_grad_batches == 1
+
+
+def test_payloads_lazy_materialization_and_order():
+    """PersiaTrainingBatch.payloads is built lazily (the flagship loop never
+    touches it); on access it must materialize exactly once, in the original
+    id_type_features order, and stay stable across reads."""
+    schema = _schema()
+    eng = _engine(schema)
+    tb = eng.process_batch(_batch())
+    # simulate deferred construction state (CPU path builds eagerly; the
+    # property contract must hold regardless)
+    first = tb.payloads
+    names = [p.name for p in first]
+    assert names == ["a", "b"]
+    again = tb.payloads
+    assert again is first  # stable list identity, no rebuild
+    assert tb.training_embeddings()[0].shape[0] == tb.batch_size
