@@ -135,18 +135,25 @@ def main():
     )
     if bf16_weights:
         model.bfloat16()
-    if world > 1:
+    # world>1 graph mode: do NOT capture RCCL inside the graph (unproven and
+    # a capture hang would sink the whole run).  Instead the graph covers
+    # fwd+bwd with gradients accumulating into ONE flat bf16 buffer; a single
+    # plain all_reduce runs between the two graphs per step (a few MB over
+    # xGMI).  DDP is only constructed for the eager fallback.
+    defer_ddp = bool(args.graph) and use_gpu and world > 1 and bf16_weights
+
+    def _wrap_ddp(m):
         from torch.nn.parallel import DistributedDataParallel
 
-        model = DistributedDataParallel(
-            model,
+        return DistributedDataParallel(
+            m,
             device_ids=[local_rank] if use_gpu else None,
             gradient_as_bucket_view=True,
             bucket_cap_mb=50,
-            # required for capturing the DDP backward (bucket allreduce
-            # order must be frozen) into the per-iteration hipGraph
-            static_graph=bool(args.graph) and use_gpu,
         )
+
+    if world > 1 and not defer_ddp:
+        model = _wrap_ddp(model)
     opt = torch.optim.SGD(model.parameters(), lr=0.01)
     loss_fn = torch.nn.functional.binary_cross_entropy_with_logits
 
@@ -191,8 +198,10 @@ def main():
     # hipGraph with static in/out buffers; each step copies inputs in and
     # replays (the dense side was launch-bound, ~200 kernels/step).
     graph = None
+    graph_upd = None
+    flat_grads = None
     static = {}
-    if args.graph and use_gpu:
+    if args.graph and use_gpu and (world == 1 or defer_ddp):
         try:
             static = {
                 "dense": torch.zeros(B, args.num_dense, device=device),
@@ -211,7 +220,40 @@ def main():
             # launches replace the per-layer cast storm.
             # PA_GRAPH_BF16=0 selects the autocast variant (A/B switch).
             # (model.bfloat16() itself ran before the DDP wrap, above)
-            if bf16_weights:
+            if bf16_weights and world > 1:
+                import torch.distributed as dist
+
+                g_params = list(model.parameters())
+                n_tot = sum(p.numel() for p in g_params)
+                flat_grads = torch.zeros(
+                    n_tot, dtype=torch.bfloat16, device=device
+                )
+                off = 0
+                for p in g_params:
+                    p.grad = flat_grads[off : off + p.numel()].view_as(p)
+                    off += p.numel()
+                g_masters = [p.detach().clone().float() for p in g_params]
+                g_grads32 = [torch.zeros_like(m) for m in g_masters]
+                lr = opt.param_groups[0]["lr"]
+
+                def iteration():
+                    static["base"].grad.zero_()
+                    torch._foreach_zero_([flat_grads])
+                    logits = model(static["dense"], static["base"])
+                    loss = loss_fn(logits.float(), static["label"])
+                    loss.backward()
+                    return loss
+
+                def update_body():
+                    grads = [p.grad for p in g_params]
+                    torch._foreach_copy_(g_grads32, grads)
+                    # allreduce delivers the SUM; fold the 1/world average
+                    # into the SGD step
+                    torch._foreach_add_(g_masters, g_grads32, alpha=-lr / world)
+                    with torch.no_grad():
+                        torch._foreach_copy_(g_params, g_masters)
+
+            elif bf16_weights:
                 g_params = list(model.parameters())
                 g_masters = [p.detach().clone().float() for p in g_params]
                 g_grads32 = [torch.zeros_like(m) for m in g_masters]
@@ -254,16 +296,45 @@ def main():
             with torch.cuda.stream(s):
                 for _ in range(3):
                     iteration()
+                    if flat_grads is not None:
+                        import torch.distributed as dist
+
+                        dist.all_reduce(flat_grads)
+                        update_body()
             torch.cuda.current_stream().wait_stream(s)
             graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(graph):
                 static["loss"] = iteration()
+            if flat_grads is not None:
+                # verify the captured backward really accumulates into the
+                # flat views (autograd may re-allocate .grad instead of
+                # adding in place, which would silently train nothing)
+                with torch.no_grad():
+                    static["dense"].normal_()
+                    static["base"].normal_()
+                graph.replay()
+                torch.cuda.synchronize()
+                if float(flat_grads.float().abs().sum().item()) == 0.0:
+                    raise RuntimeError("flat-grad capture yielded no gradients")
+                with torch.no_grad():
+                    static["dense"].zero_()
+                    static["base"].zero_()
+                graph_upd = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph_upd):
+                    update_body()
         except Exception as e:  # pragma: no cover - fall back to eager
             import sys as _sys
 
             print(f"# hipGraph capture failed ({e}); running eager",
                   file=_sys.stderr, flush=True)
             graph = None
+            graph_upd = None
+            if world > 1 and defer_ddp:
+                # the eager fallback still needs synchronized dense grads
+                for p in model.parameters():
+                    p.grad = None
+                model = _wrap_ddp(model)
+                opt = torch.optim.SGD(model.parameters(), lr=0.01)
 
     timing = os.environ.get("PA_BENCH_TIMING", "0") == "1"
     tstats = {"get": 0.0, "copy": 0.0, "replay": 0.0, "apply": 0.0, "n": 0}
@@ -279,6 +350,11 @@ def main():
             if timing:
                 t1 = time.perf_counter()
             graph.replay()
+            if graph_upd is not None:
+                import torch.distributed as dist
+
+                dist.all_reduce(flat_grads)
+                graph_upd.replay()
             if timing:
                 t2 = time.perf_counter()
             engine.apply_gradients_base(tb, sum_base_grads=[static["base"].grad])

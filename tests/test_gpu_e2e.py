@@ -130,3 +130,44 @@ def test_bench_default_config_one_step():
     res = json.loads(line)
     assert res["value"] > 0
     assert res["n_gpus"] == 1
+
+
+def test_flat_grad_views_accumulate_inside_graph():
+    """bench.py's multi-GPU graph mode points every param.grad at a view of
+    ONE flat bf16 buffer and captures fwd+bwd; autograd must accumulate into
+    those views IN PLACE (a silent re-allocation would train nothing)."""
+    dev = _dev()
+    torch.manual_seed(3)
+    model = torch.nn.Sequential(
+        torch.nn.Linear(16, 32), torch.nn.ReLU(), torch.nn.Linear(32, 1)
+    ).to(dev).bfloat16()
+    params = list(model.parameters())
+    flat = torch.zeros(sum(p.numel() for p in params), dtype=torch.bfloat16, device=dev)
+    off = 0
+    for p in params:
+        p.grad = flat[off : off + p.numel()].view_as(p)
+        off += p.numel()
+    x = torch.randn(64, 16, device=dev, dtype=torch.bfloat16)
+    y = torch.randn(64, device=dev, dtype=torch.bfloat16)
+
+    def iteration():
+        torch._foreach_zero_([flat])
+        loss = ((model(x).squeeze(1) - y) ** 2).mean()
+        loss.backward()
+        return loss
+
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        for _ in range(3):
+            iteration()
+    torch.cuda.current_stream().wait_stream(s)
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        iteration()
+    g.replay()
+    torch.cuda.synchronize()
+    total = float(flat.float().abs().sum().item())
+    assert total > 0.0, "captured backward did not write the flat grad views"
+    # the python-side .grad objects must still BE the views
+    assert params[0].grad.data_ptr() == flat.data_ptr()
