@@ -136,7 +136,7 @@ def test_flat_grad_views_accumulate_inside_graph():
     """bench.py's multi-GPU graph mode points every param.grad at a view of
     ONE flat bf16 buffer and captures fwd+bwd; autograd must accumulate into
     those views IN PLACE (a silent re-allocation would train nothing)."""
-    dev = _dev()
+    dev = torch.device("cuda", 0)
     torch.manual_seed(3)
     model = torch.nn.Sequential(
         torch.nn.Linear(16, 32), torch.nn.ReLU(), torch.nn.Linear(32, 1)
