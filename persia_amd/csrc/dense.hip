@@ -234,14 +234,22 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
         w[h] = *(const bf16x8*)(A + (int64_t)m * K + k0 + s_c8);
     }
   };
+  // XOR-swizzled m-granule: element (n, m) lives at granule (m/8)^((n>>3)&7)
+  // inside row n.  The plain transposed scalar writes put all 64 lanes in 4
+  // banks (PMC: ~12 conflict cycles per LDS instruction); the swizzle
+  // spreads simultaneous writes across the full bank set (~2-way) while the
+  // b128 fragment reads keep their 16-byte alignment and 2-way pattern.
+  auto wswz = [](int n, int m) {
+    return n * WLD + ((((m >> 3) ^ ((n >> 3) & 7)) << 3) | (m & 7));
+  };
   auto wwrite = [&](int buf, bf16x8 (&v)[2], bf16x8 (&w)[2]) {
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
       const int m = s_m + h * 32;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) lds_dct[buf][(s_c8 + j) * WLD + m] = v[h][j];
+      for (int j = 0; j < 8; ++j) lds_dct[buf][wswz(s_c8 + j, m)] = v[h][j];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) lds_at[buf][(s_c8 + j) * WLD + m] = w[h][j];
+      for (int j = 0; j < 8; ++j) lds_at[buf][wswz(s_c8 + j, m)] = w[h][j];
     }
   };
 
@@ -262,11 +270,11 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
 #pragma unroll
       for (int an = 0; an < 2; ++an) {
         const bf16x8 a_frag = *(const bf16x8*)
-            &lds_dct[buf][(wr * 32 + an * 16 + fi) * WLD + ms + fk8];
+            &lds_dct[buf][wswz(wr * 32 + an * 16 + fi, ms + fk8)];
 #pragma unroll
         for (int bk = 0; bk < 2; ++bk) {
           const bf16x8 b_frag = *(const bf16x8*)
-              &lds_at[buf][(wc * 32 + bk * 16 + fi) * WLD + ms + fk8];
+              &lds_at[buf][wswz(wc * 32 + bk * 16 + fi, ms + fk8)];
           acc[an][bk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a_frag, b_frag, acc[an][bk], 0, 0, 0);
         }
