@@ -106,6 +106,15 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
       }
     }
   };
+  // XOR-swizzled k-granule for the B image: element (n, k) lives at granule
+  // (k/8) ^ (n>>3 masked to the granule count).  Without it the TRANS_B
+  // scalar transposed writes land all lanes in a handful of banks (same
+  // pathology PMC exposed in wgrad: ~12 conflict cycles per LDS op); the
+  // b128 fragment reads stay 16-byte aligned.
+  auto bswz = [](int n, int k) {
+    return n * LK +
+           ((((k >> 3) ^ ((n >> 3) & (TBK / 8 - 1))) << 3) | (k & 7));
+  };
   auto stage_write = [&](int buf, bf16x8 (&ra)[PA], bf16x8 (&rb)[PB]) {
 #pragma unroll
     for (int i = 0; i < PA; ++i) {
@@ -118,12 +127,12 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
       const int p = tid + i * 256;
       if (!TRANS_B) {
         const int r = p / ACPR, c8 = (p % ACPR) * 8;
-        *(bf16x8*)&lds_b[buf][r * LK + c8] = rb[i];
+        *(bf16x8*)&lds_b[buf][bswz(r, c8)] = rb[i];
       } else {
         const int r = p / (TBN / 8), c8 = (p % (TBN / 8)) * 8;  // k-row, n-col
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          lds_b[buf][(c8 + j) * LK + r] = rb[i][j];
+          lds_b[buf][bswz(c8 + j, r)] = rb[i][j];
       }
     }
   };
@@ -149,7 +158,7 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
 #pragma unroll
         for (int bn = 0; bn < NFRAG; ++bn) {
           const bf16x8 b_frag = *(const bf16x8*)
-              &lds_b[buf][(wc * WN + bn * 16 + fi) * LK + kk + fk8];
+              &lds_b[buf][bswz(wc * WN + bn * 16 + fi, kk + fk8)];
           acc[am][bn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a_frag, b_frag, acc[am][bn], 0, 0, 0);
         }
