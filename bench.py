@@ -54,9 +54,10 @@ def parse_args():
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--graph", type=int, default=1,
                    help="capture dense fwd+bwd in a hipGraph (1 GPU only)")
-    p.add_argument("--fused-dense", type=int, default=0,
+    p.add_argument("--fused-dense", type=int, default=1,
                    help="use the hand-written MFMA GEMM+bias+ReLU layers "
-                        "(currently ~0.95x the graphed hipBLASLt path)")
+                        "(beats the graphed hipBLASLt path since the "
+                        "XOR-swizzled staging fixed wgrad; 0 = hipBLASLt)")
     args = p.parse_args()
     preset = dict(PRESETS[args.preset])
     preset.setdefault("batch_size", 8192)  # x8 ranks = the MLPerf DLRM 64k global batch

@@ -16,9 +16,11 @@ import torch.nn.functional as F
 
 
 def _wgrad_via_blas() -> bool:
+    # default: the hand-written wgrad (XOR-swizzled LDS staging) — it beats
+    # hipBLASLt 2-4x on the DLRM shapes; "blas" kept as an A/B switch
     import os
 
-    return os.environ.get("PA_FUSED_WGRAD", "blas") == "blas"
+    return os.environ.get("PA_FUSED_WGRAD", "own") == "blas"
 
 
 def _pad32(x: torch.Tensor, dim: int) -> torch.Tensor:
