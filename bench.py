@@ -28,7 +28,9 @@ PRESETS = {
     "terabyte": dict(model="dlrm", num_sparse=26, num_dense=13, dim=128, rows=1e10),
     "dcn-spill": dict(model="dcn", num_sparse=26, num_dense=13, dim=64, rows=1e11,
                       spill_capacity=2e8),
-    "100t": dict(model="dlrm", num_sparse=64, num_dense=13, dim=8, rows=1e12),
+    # dim-8 towers don't suit the fused MFMA layers (skinny K, odd widths)
+    "100t": dict(model="dlrm", num_sparse=64, num_dense=13, dim=8, rows=1e12,
+                 fused_dense=0),
 }
 
 
@@ -54,14 +56,16 @@ def parse_args():
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--graph", type=int, default=1,
                    help="capture dense fwd+bwd in a hipGraph (1 GPU only)")
-    p.add_argument("--fused-dense", type=int, default=1,
+    p.add_argument("--fused-dense", type=int, default=None,
                    help="use the hand-written MFMA GEMM+bias+ReLU layers "
                         "(beats the graphed hipBLASLt path since the "
-                        "XOR-swizzled staging fixed wgrad; 0 = hipBLASLt)")
+                        "XOR-swizzled staging fixed wgrad; 0 = hipBLASLt; "
+                        "preset-dependent default)")
     args = p.parse_args()
     preset = dict(PRESETS[args.preset])
     preset.setdefault("batch_size", 8192)  # x8 ranks = the MLPerf DLRM 64k global batch
     preset.setdefault("spill_capacity", 0)
+    preset.setdefault("fused_dense", 1)
     for k, v in preset.items():
         if getattr(args, k, None) is None:
             setattr(args, k, v)
