@@ -112,14 +112,6 @@ __global__ __launch_bounds__(256) void interact_bwd_mfma_kernel(
   const int fi = lane & 15;
   const int fk8 = (lane >> 4) * 8;
   const int d8 = D / 8;
-  // XOR-swizzled feature-granule for the transposed V image: the plain
-  // scalar transposed writes land whole wavefronts in a few LDS banks
-  // (see the wgrad PMC finding); granule mask keeps the swizzle inside the
-  // (Fk/8)-granule row for non-power-of-two counts
-  const int swm = (((Fk >> 3) & 7) == 0) ? 7 : 3;
-  auto vswz = [&](int row, int col) {
-    return row * LDA + ((((col >> 3) ^ ((row >> 3) & swm)) << 3) | (col & 7));
-  };
   for (int64_t b = (int64_t)blockIdx.x * waves + wave; b < B;
        b += (int64_t)gridDim.x * waves) {
     // stage V transposed (scalar LDS writes; reads stay b128)
@@ -128,7 +120,7 @@ __global__ __launch_bounds__(256) void interact_bwd_mfma_kernel(
       const int fr = t / d8, c8 = (t % d8) * 8;
       const bf16x8i row = *(const bf16x8i*)&src[fr * D + c8];
 #pragma unroll
-      for (int k = 0; k < 8; ++k) vt[vswz(c8 + k, fr)] = row[k];
+      for (int k = 0; k < 8; ++k) vt[(c8 + k) * LDA + fr] = row[k];
     }
     // pair grads -> symmetric A (diag stays 0); previous sample's entries
     // are overwritten pairwise, padding stays 0 from the one-time zero fill
@@ -148,7 +140,7 @@ __global__ __launch_bounds__(256) void interact_bwd_mfma_kernel(
         f32x4i acc = {};
         for (int k = 0; k < Fk; k += 32) {
           const bf16x8i a = *(const bf16x8i*)&A[(ti * 16 + fi) * LDA + k + fk8];
-          const bf16x8i c = *(const bf16x8i*)&vt[vswz(td * 16 + fi, k + fk8)];
+          const bf16x8i c = *(const bf16x8i*)&vt[(td * 16 + fi) * LDA + k + fk8];
           acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, c, acc, 0, 0, 0);
         }
         const int d = td * 16 + fi;
@@ -237,10 +229,6 @@ __global__ __launch_bounds__(256) void interact_bwd_packed_kernel(
   const int fi = lane & 15;
   const int fk8 = (lane >> 4) * 8;
   const int d8 = D / 8;
-  const int swm = (((Fk >> 3) & 7) == 0) ? 7 : 3;
-  auto vswz = [&](int row, int col) {
-    return row * LDA + ((((col >> 3) ^ ((row >> 3) & swm)) << 3) | (col & 7));
-  };
   for (int64_t b = (int64_t)blockIdx.x * waves + wave; b < B;
        b += (int64_t)gridDim.x * waves) {
     for (int t = lane; t < F * d8; t += 64) {
@@ -254,7 +242,7 @@ __global__ __launch_bounds__(256) void interact_bwd_packed_kernel(
         for (int k = 0; k < 8; ++k) row[k] = if2bf(__half2float(src[k]));
       }
 #pragma unroll
-      for (int k = 0; k < 8; ++k) vt[vswz(c8 + k, fr)] = row[k];
+      for (int k = 0; k < 8; ++k) vt[(c8 + k) * LDA + fr] = row[k];
     }
     const short* gp = g + b * P;
     for (int p = lane; p < P; p += 64) {
@@ -271,7 +259,7 @@ __global__ __launch_bounds__(256) void interact_bwd_packed_kernel(
         f32x4i acc = {};
         for (int k = 0; k < Fk; k += 32) {
           const bf16x8i a = *(const bf16x8i*)&A[(ti * 16 + fi) * LDA + k + fk8];
-          const bf16x8i c = *(const bf16x8i*)&vt[vswz(td * 16 + fi, k + fk8)];
+          const bf16x8i c = *(const bf16x8i*)&vt[(td * 16 + fi) * LDA + k + fk8];
           acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, c, acc, 0, 0, 0);
         }
         const int d = td * 16 + fi;
