@@ -334,21 +334,6 @@ __global__ void bias_grad_kernel(const short* __restrict__ dC,
                                  int rows_per_block) {
   const int m_begin = blockIdx.x * rows_per_block;
   const int m_end = min(M, m_begin + rows_per_block);
-  if ((N & 7) == 0) {
-    // vector path: 16-byte loads (the scalar version moves 2 B per lane per
-    // instruction — an 8x bandwidth handicap)
-    for (int c8 = threadIdx.x * 8; c8 < N; c8 += blockDim.x * 8) {
-      float acc[8] = {};
-      for (int m = m_begin; m < m_end; ++m) {
-        const bf16x8 v = *(const bf16x8*)(dC + (int64_t)m * N + c8);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) acc[j] += bf2f(v[j]);
-      }
-#pragma unroll
-      for (int j = 0; j < 8; ++j) atomicAdd(&db[c8 + j], acc[j]);
-    }
-    return;
-  }
   for (int col = threadIdx.x; col < N; col += blockDim.x) {
     float acc = 0.0f;
     for (int m = m_begin; m < m_end; ++m) acc += bf2f(dC[(int64_t)m * N + col]);
