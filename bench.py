@@ -72,8 +72,32 @@ def parse_args():
     return args
 
 
+def _self_launch(args):
+    """Re-exec under torch.distributed.run when --gpus N (N>1) is given
+    directly (no torchrun env present): one rank per GPU over RCCL.  The
+    child ranks inherit the full original argv; their RANK/WORLD_SIZE env
+    stops the recursion.  Propagates the child's stdout (rank 0 prints the
+    JSON line) and exit code."""
+    import socket
+    import subprocess
+    import sys
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", f"--nproc-per-node={args.gpus}",
+        "--master-addr=127.0.0.1", f"--master-port={port}",
+        os.path.abspath(__file__), *sys.argv[1:],
+    ]
+    raise SystemExit(subprocess.call(cmd))
+
+
 def main():
     args = parse_args()
+    if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        _self_launch(args)
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
