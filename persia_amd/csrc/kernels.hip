@@ -135,6 +135,14 @@ __global__ void probe_claim_kernel(ull* __restrict__ table_keys,
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   const ull k = query[i];
+  if (k == PA_EMPTY_KEY) {
+    // a2a padding sentinel (distributed recv buffers): never probed — the
+    // empty key would match every free slot
+    out_slot[i] = -1;
+    out_new[i] = 0;
+    if (out_evict_idx) out_evict_idx[i] = -1;
+    return;
+  }
   long long slot;
   int is_new = 0;
   ull victim = 0;
@@ -165,12 +173,19 @@ __global__ void probe_claim_kernel(ull* __restrict__ table_keys,
   if (out_evict_idx) out_evict_idx[i] = eidx;
 }
 
-// one wave per key: init freshly claimed rows, then gather emb -> out (f32)
+// one wave per key: init freshly claimed rows, then gather emb -> out.
+// OutT = f32 (local path) or f16 (distributed wire: the gathered rows go
+// straight onto the xGMI all-to-all, so the wire cast is fused here).
+__device__ __forceinline__ void pa_store_out(float v, float* p) { *p = v; }
+__device__ __forceinline__ void pa_store_out(float v, __half* p) {
+  *p = __float2half(v);
+}
+template <typename OutT>
 __global__ void init_gather_kernel(float* __restrict__ arena,
                                    const ull* __restrict__ query,
                                    const long long* __restrict__ slots,
                                    const int* __restrict__ is_new,
-                                   float* __restrict__ out, int64_t n, int dim,
+                                   OutT* __restrict__ out, int64_t n, int dim,
                                    int row_width, double lo, double hi,
                                    float state_init,
                                    const long long* __restrict__ evict_idx,
@@ -189,9 +204,9 @@ __global__ void init_gather_kernel(float* __restrict__ arena,
     const int64_t i = w * G + sub;
     if (i >= n) continue;
     const long long slot = slots[i];
-    float* dst = out + i * dim;
+    OutT* dst = out + i * dim;
     if (slot < 0) {
-      for (int c = c0; c < dim; c += PA_WAVE) dst[c] = 0.0f;
+      for (int c = c0; c < dim; c += PA_WAVE) pa_store_out(0.0f, dst + c);
       continue;
     }
     float* row = arena + (int64_t)slot * row_width;
@@ -209,7 +224,7 @@ __global__ void init_gather_kernel(float* __restrict__ arena,
       for (int c = dim + c0; c < row_width; c += st)
         row[c] = state_init;
     }
-    for (int c = c0; c < dim; c += st) dst[c] = row[c];
+    for (int c = c0; c < dim; c += st) pa_store_out(row[c], dst + c);
   }
 }
 
@@ -355,6 +370,7 @@ __global__ void update_kernel(ull* __restrict__ table_keys,
   for (int64_t i = (int64_t)blockIdx.x * waves_per_block + wave; i < n;
        i += (int64_t)gridDim.x * waves_per_block) {
     const ull k = query[i];
+    if (k == PA_EMPTY_KEY) continue;  // a2a padding: silent skip
     // lane-parallel probe: lanes 0..31 each inspect one slot of the window
     long long slot = -1;
     const int64_t b = (int64_t)(k & (ull)mask);
@@ -656,6 +672,48 @@ __global__ void grad_scatter_kernel(const GradT* __restrict__ grads,
   }
 }
 
+// Indexed wire variant: out[out_idx[u], :] = acc in the WIRE dtype — packs
+// each unique key's reduced gradient straight into the a2a send buffer
+// (layout [world, cap] by owner rank), skipping both the [U, dim] f32
+// intermediate and a separate pack pass.  n_unique comes from n_dev (padded
+// sync-free dedup); invalid/overflowed entries point at the dummy tail row.
+// Small dims pack 64/dim keys per wave (sub-wave groups).
+template <typename OutT>
+__global__ void grad_scatter_idx_kernel(
+    const __half* __restrict__ grads, const int64_t* __restrict__ perm,
+    const int64_t* __restrict__ ustarts, const int64_t* __restrict__ seg_id,
+    const float* __restrict__ seg_scale, const int64_t* __restrict__ out_idx,
+    OutT* __restrict__ out, int64_t n_unique, int dim,
+    const long long* __restrict__ n_dev) {
+  if (n_dev) n_unique = *n_dev;
+  const int wave = threadIdx.x / PA_WAVE;
+  const int lane = threadIdx.x % PA_WAVE;
+  const int waves_per_block = blockDim.x / PA_WAVE;
+  const int G = (dim < PA_WAVE && (PA_WAVE % dim) == 0) ? PA_WAVE / dim : 1;
+  const int sub = (G > 1) ? lane / dim : 0;
+  const int c0 = (G > 1) ? lane % dim : lane;
+  const int st = (G > 1) ? dim : PA_WAVE;
+  const int64_t n_waveitems = (n_unique + G - 1) / G;
+  for (int64_t w = (int64_t)blockIdx.x * waves_per_block + wave;
+       w < n_waveitems; w += (int64_t)gridDim.x * waves_per_block) {
+    const int64_t u = w * G + sub;
+    if (u >= n_unique) continue;
+    const int64_t lo = ustarts[u], hi = ustarts[u + 1];
+    OutT* dst = out + out_idx[u] * dim;
+    for (int c = c0; c < dim; c += st) {
+      float acc = 0.0f;
+      for (int64_t p = lo; p < hi; ++p) {
+        const int64_t s = seg_id[perm[p]];
+        if (s >= 0) {
+          const float sc = seg_scale ? seg_scale[s] : 1.0f;
+          if (sc != 0.0f) acc += __half2float(grads[s * dim + c]) * sc;
+        }
+      }
+      pa_store_out(acc, dst + c);
+    }
+  }
+}
+
 // ------------------------------------------- fused scatter + optimizer update
 
 // One wave per unique sign: reduce its gradient from the per-segment grads
@@ -688,8 +746,10 @@ __global__ void scatter_update_kernel(
   for (int64_t w = (int64_t)blockIdx.x * waves_per_block + wave;
        w < n_waveitems; w += (int64_t)gridDim.x * waves_per_block) {
     const int64_t u = w * G + sub;
-    const bool active = u < n;
-    const ull k = active ? uniq[u] : 0;
+    const ull k = (u < n) ? uniq[u] : 0;
+    // empty key = a2a padding: silent skip (the empty key would match every
+    // free table slot in the probe)
+    const bool active = (u < n) && (k != PA_EMPTY_KEY);
     // subgroup-parallel probe over the 32-slot window
     long long slot = -1;
     if (active) {
@@ -816,8 +876,8 @@ __global__ void scatter_update2_kernel(
   for (int64_t w = (int64_t)blockIdx.x * waves_per_block + wave;
        w < n_waveitems; w += (int64_t)gridDim.x * waves_per_block) {
     const int64_t u = w * 2 + half;
-    const bool active = u < n;
-    const ull k = active ? uniq[u] : 0;
+    const ull k = (u < n) ? uniq[u] : 0;
+    const bool active = (u < n) && (k != PA_EMPTY_KEY);
     long long slot = -1;
     if (active) {
       const int64_t b = (int64_t)(k & (ull)mask);
@@ -827,9 +887,12 @@ __global__ void scatter_update2_kernel(
       if (table_keys[j] == k) slot = j;
     }
     const unsigned long long found = __ballot(slot >= 0);
+    const unsigned long long nonzero = __ballot(k != PA_EMPTY_KEY);
     const unsigned fA = (unsigned)(found & 0xFFFFFFFFull);
     const unsigned fB = (unsigned)(found >> 32);
-    bool okA = (w * 2 + 0) < n, okB = (w * 2 + 1) < n;
+    // empty-key (a2a padding) entries skip silently, not as misses
+    bool okA = (w * 2 + 0) < n && (nonzero & 1ull);
+    bool okB = (w * 2 + 1) < n && (nonzero >> 32 & 1ull);
     if (okA && fA == 0) { if (lane == 0) atomicAdd(&skipped[0], 1); okA = false; }
     if (okB && fB == 0) { if (lane == 0) atomicAdd(&skipped[0], 1); okB = false; }
     const long long slotA = okA ? __shfl(slot, __ffs(fA) - 1) : -1;
@@ -1007,8 +1070,8 @@ __global__ void scatter_update4_kernel(
   for (int64_t w = (int64_t)blockIdx.x * waves_per_block + wave;
        w < n_waveitems; w += (int64_t)gridDim.x * waves_per_block) {
     const int64_t u = w * 4 + kq;
-    const bool active = u < n;
-    const ull k = active ? uniq[u] : 0;
+    const ull k = (u < n) ? uniq[u] : 0;
+    const bool active = (u < n) && (k != PA_EMPTY_KEY);
     long long slot = -1;
     if (active) {
       const int64_t b = (int64_t)(k & (ull)mask);
@@ -1022,13 +1085,15 @@ __global__ void scatter_update4_kernel(
       }
     }
     const unsigned long long found = __ballot(slot >= 0);
+    const unsigned long long nonzero = __ballot(k != PA_EMPTY_KEY);
     bool ok[4];
     long long slotk[4];
     float* rowk[4];
 #pragma unroll
     for (int q = 0; q < 4; ++q) {
       const unsigned fq = (unsigned)((found >> (16 * q)) & 0xFFFFull);
-      ok[q] = (w * 4 + q) < n;
+      // empty-key (a2a padding) entries skip silently, not as misses
+      ok[q] = (w * 4 + q) < n && (nonzero >> (16 * q) & 1ull);
       if (ok[q] && fq == 0) {
         if (lane == 0) atomicAdd(&skipped[0], 1);
         ok[q] = false;
@@ -1259,14 +1324,27 @@ void store_lookup(torch::Tensor table_keys, torch::Tensor ticks,
                      spill ? evict_count.data_ptr<int32_t>() : nullptr,
                      spill ? (long long*)evict_idx.data_ptr<int64_t>() : nullptr,
                      n_dev);
-  hipLaunchKernelGGL(init_gather_kernel, dim3(n_blocks_for(n, 4)), dim3(256),
-                     0, st, arena.data_ptr<float>(),
-                     (const ull*)query.data_ptr<int64_t>(),
-                     (const long long*)slots.data_ptr<int64_t>(),
-                     is_new.data_ptr<int32_t>(), out.data_ptr<float>(), n,
-                     (int)dim, row_width, lo, hi, (float)state_init,
-                     spill ? (const long long*)evict_idx.data_ptr<int64_t>() : nullptr,
-                     spill ? evict_rows.data_ptr<float>() : nullptr, n_dev);
+  if (out.scalar_type() == torch::kFloat16) {
+    // f16 out: distributed wire path (rows go straight onto the a2a)
+    hipLaunchKernelGGL(init_gather_kernel<__half>, dim3(n_blocks_for(n, 4)),
+                       dim3(256), 0, st, arena.data_ptr<float>(),
+                       (const ull*)query.data_ptr<int64_t>(),
+                       (const long long*)slots.data_ptr<int64_t>(),
+                       is_new.data_ptr<int32_t>(),
+                       (__half*)out.data_ptr<at::Half>(), n, (int)dim,
+                       row_width, lo, hi, (float)state_init,
+                       spill ? (const long long*)evict_idx.data_ptr<int64_t>() : nullptr,
+                       spill ? evict_rows.data_ptr<float>() : nullptr, n_dev);
+  } else {
+    hipLaunchKernelGGL(init_gather_kernel<float>, dim3(n_blocks_for(n, 4)),
+                       dim3(256), 0, st, arena.data_ptr<float>(),
+                       (const ull*)query.data_ptr<int64_t>(),
+                       (const long long*)slots.data_ptr<int64_t>(),
+                       is_new.data_ptr<int32_t>(), out.data_ptr<float>(), n,
+                       (int)dim, row_width, lo, hi, (float)state_init,
+                       spill ? (const long long*)evict_idx.data_ptr<int64_t>() : nullptr,
+                       spill ? evict_rows.data_ptr<float>() : nullptr, n_dev);
+  }
 }
 
 // fast-path lookup: probe/claim + fused init/f16-scatter to sum rows
@@ -1444,6 +1522,41 @@ void grad_scatter(torch::Tensor grads, torch::Tensor perm,
 #undef PA_GS
 }
 
+void grad_scatter_idx(torch::Tensor grads, torch::Tensor perm,
+                      torch::Tensor ustarts, torch::Tensor seg_id,
+                      torch::Tensor seg_scale, torch::Tensor out_idx,
+                      torch::Tensor out, torch::Tensor u_count) {
+  const int64_t n_unique = out_idx.numel();  // padded upper bound
+  const int64_t dim = out.size(1);
+  if (n_unique == 0) return;
+  TORCH_CHECK(grads.scalar_type() == torch::kFloat16,
+              "grad_scatter_idx: f16 grads");
+  hipStream_t st = cur_stream();
+  const dim3 grid(n_blocks_for(n_unique, 4)), block(256);
+  const float* scale_ptr =
+      seg_scale.numel() ? seg_scale.data_ptr<float>() : nullptr;
+  const long long* n_dev =
+      u_count.numel() ? (const long long*)u_count.data_ptr<int64_t>() : nullptr;
+  if (out.scalar_type() == torch::kFloat16) {
+    hipLaunchKernelGGL(grad_scatter_idx_kernel<__half>, grid, block, 0, st,
+                       (const __half*)grads.data_ptr<at::Half>(),
+                       perm.data_ptr<int64_t>(), ustarts.data_ptr<int64_t>(),
+                       seg_id.data_ptr<int64_t>(), scale_ptr,
+                       out_idx.data_ptr<int64_t>(),
+                       (__half*)out.data_ptr<at::Half>(), n_unique, (int)dim,
+                       n_dev);
+  } else if (out.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL(grad_scatter_idx_kernel<float>, grid, block, 0, st,
+                       (const __half*)grads.data_ptr<at::Half>(),
+                       perm.data_ptr<int64_t>(), ustarts.data_ptr<int64_t>(),
+                       seg_id.data_ptr<int64_t>(), scale_ptr,
+                       out_idx.data_ptr<int64_t>(),
+                       out.data_ptr<float>(), n_unique, (int)dim, n_dev);
+  } else {
+    TORCH_CHECK(false, "grad_scatter_idx: out must be f16/f32");
+  }
+}
+
 void scatter_update(torch::Tensor table_keys, torch::Tensor ticks,
                     torch::Tensor arena, torch::Tensor uniq,
                     torch::Tensor grads, torch::Tensor perm,
@@ -1590,6 +1703,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("store_import", &store_import, "bulk insert rows (checkpoint load)");
   m.def("segment_sum", &segment_sum, "fused gather + per-sample summation");
   m.def("grad_scatter", &grad_scatter, "ordered per-sign gradient scatter");
+  m.def("grad_scatter_idx", &grad_scatter_idx,
+        "ordered per-sign gradient scatter packed into the a2a send layout");
   m.def("sign_prep", &sign_prep, "prefix-fold + splitmix64 key mixing");
   m.def("sign_prep_stack", &sign_prep_stack,
         "hashstack expansion + prefix-fold + splitmix64 key mixing");
