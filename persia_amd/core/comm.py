@@ -11,7 +11,7 @@ Backend "nccl" (= RCCL on ROCm) on GPU; "gloo" for CPU tests (world_size>1
 multi-process CPU tests run in CI — gloo supports all_to_all_single with
 uneven splits).
 """
-from typing import List, Optional
+from typing import List, Optional, Tuple
 
 import torch
 import torch.distributed as dist
@@ -89,6 +89,18 @@ class DistContext:
         )
         return out
 
+    def all_to_all_even(self, inp: torch.Tensor,
+                        out: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Fixed-shape all-to-all: every rank sends/receives exactly
+        ``inp.shape[0] // world_size`` rows per peer.  This is the sync-free
+        exchange of the fused distributed path (capacity-padded buckets; the
+        variable per-destination counts never reach the host).  ``out`` may
+        be a contiguous view (e.g. the prefix of a buffer with a dummy row)."""
+        if out is None:
+            out = torch.empty_like(inp)
+        dist.all_to_all_single(out, inp.contiguous(), group=self.group)
+        return out
+
     def barrier(self):
         if self.distributed:
             dist.barrier(group=self.group)
@@ -97,5 +109,28 @@ class DistContext:
         if not self.distributed:
             return x
         t = torch.tensor([x], dtype=torch.float64)
-        dist.all_reduce(t, group=self.group)
+        if not self._gloo_like():
+            # RCCL rejects CPU tensors: stage through the device like
+            # all_to_all_lengths does
+            dev = torch.device("cuda", torch.cuda.current_device())
+            t_d = t.to(dev)
+            dist.all_reduce(t_d, group=self.group)
+            t = t_d.cpu()
+        else:
+            dist.all_reduce(t, group=self.group)
         return float(t.item())
+
+    def allreduce_int_minmax(self, x: int) -> Tuple[int, int]:
+        """(min, max) of an int across ranks — one-time shape checks for the
+        padded a2a fast path (all ranks must agree on nnz/cap)."""
+        if not self.distributed:
+            return x, x
+        t = torch.tensor([x, -x], dtype=torch.int64)
+        if not self._gloo_like():
+            dev = torch.device("cuda", torch.cuda.current_device())
+            t_d = t.to(dev)
+            dist.all_reduce(t_d, op=dist.ReduceOp.MAX, group=self.group)
+            t = t_d.cpu()
+        else:
+            dist.all_reduce(t, op=dist.ReduceOp.MAX, group=self.group)
+        return -int(t[1].item()), int(t[0].item())

@@ -134,6 +134,11 @@ class _GroupCtx:
     n_sum_slots: int = 0
     send_counts: Optional[List[int]] = None
     recv_counts: Optional[List[int]] = None
+    # capacity-padded even-a2a fast path (sync-free distributed lookup):
+    a2a_plan: Optional["_GroupPlan"] = None
+    a2a_idx: Optional[torch.Tensor] = None  # [len(uniq)] uniq -> send-slot map
+    a2a_recv_keys: Optional[torch.Tensor] = None  # [world*cap] owner-side keys
+    a2a_owner_dedup: Optional[tuple] = None  # GPU: dedup_padded(recv_keys)
 
 
 class _GroupPlan:
@@ -154,6 +159,11 @@ class _GroupPlan:
         self.cat_offsets = torch.arange(S * B + 1, dtype=torch.int64, device=device)
         self.seg_id = torch.arange(S * B, dtype=torch.int64, device=device)
         self.empty_scale = torch.empty(0, dtype=torch.float32, device=device)
+        # padded-a2a constants (attached lazily by EmbeddingEngine._a2a_setup)
+        self.a2a_world: Optional[int] = None
+        self.a2a_cap: Optional[int] = None
+        self.a2a_ar: Optional[torch.Tensor] = None  # arange(nnz)
+        self.owner_seg_id: Optional[torch.Tensor] = None  # arange(world*cap)
 
 
 class _SplitSlots(torch.autograd.Function):
@@ -253,9 +263,10 @@ class PersiaTrainingBatch:
             if t.is_cuda:
                 yield t
         for g in self._groups:
+            a2a_owner = g.a2a_owner_dedup or ()
             for t in (g.uniq_keys, g.inverse, g.perm, g.ustarts, g.u_count,
                       g.sum_base, g.cat_offsets, g.seg_id, g.seg_lens,
-                      g.sqrt_mask):
+                      g.sqrt_mask, g.a2a_idx, g.a2a_recv_keys, *a2a_owner):
                 if t is not None and t.is_cuda:
                     yield t
             for sc in g.slots:
@@ -326,6 +337,13 @@ class EmbeddingEngine:
 
         self._prod_timing = _os.environ.get("PA_PROD_TIMING", "0") == "1"
         self._pt = {"prep": 0.0, "native": 0.0, "batch": 0.0, "n": 0}
+        # fused distributed path (capacity-padded even a2a, sync-free);
+        # PA_FUSED_DIST=0 falls back to the exact two-phase counts exchange,
+        # PA_FORCE_DIST=1 exercises the padded machinery at world_size=1
+        # (copy in place of the collectives) for single-GPU tests/profiling
+        self._fused_dist = _os.environ.get("PA_FUSED_DIST", "1") == "1"
+        self._force_dist = _os.environ.get("PA_FORCE_DIST", "0") == "1"
+        self._a2a_overflow = torch.zeros(1, dtype=torch.int64, device=device)
         from persia_amd.core.metrics import EngineMetrics
 
         self.metrics_enabled = bool(gconf.enable_metrics)
@@ -404,6 +422,107 @@ class EmbeddingEngine:
         rows_local = store.lookup(recv_keys, train).to(self.wire_dtype)
         rows = self.dist.all_to_all(rows_local, recv_counts, send_counts)
         return rows
+
+    # ------------------------------------- padded even-a2a (fused distributed)
+
+    def _a2a_setup(self, plan: "_GroupPlan") -> Tuple[int, int]:
+        """One-time per-plan constants for the capacity-padded a2a.
+
+        Every rank sends each peer a fixed ``cap``-row bucket (padding key 0 =
+        "no entry"), so the variable per-destination counts NEVER reach the
+        host — no ``.tolist()`` syncs in the hot loop (the reference's RPC
+        naturally carried lengths; RCCL ``all_to_all_single`` with even splits
+        is the xGMI-native equivalent).  Keys are hash-mixed ⇒ per-owner
+        counts concentrate at nnz/world (binomial, σ≈√(nnz/world)); the slack
+        of max(512, per/8) puts overflow beyond ~20σ.  A one-time allreduce
+        verifies all ranks agree on nnz (weak scaling keeps shapes equal)."""
+        if plan.a2a_cap is not None:
+            return plan.a2a_world, plan.a2a_cap
+        world = self.dist.world_size
+        nnz = plan.S * plan.B
+        if world > 1:
+            lo, hi = self.dist.allreduce_int_minmax(nnz)
+            if lo != hi:
+                raise RuntimeError(
+                    f"padded-a2a fast path needs equal per-rank batch shapes "
+                    f"(nnz min {lo} != max {hi}); set PA_FUSED_DIST=0"
+                )
+            per = max(1, nnz // world)
+            cap = per + max(512, per // 8)
+        else:
+            cap = nnz  # forced single-rank exercise: exact fit
+        plan.a2a_world = world
+        plan.a2a_cap = cap
+        plan.a2a_ar = torch.arange(nnz, dtype=torch.int64, device=self.device)
+        plan.owner_seg_id = torch.arange(
+            world * cap, dtype=torch.int64, device=self.device
+        )
+        return world, cap
+
+    def _a2a_route(self, plan: "_GroupPlan", uniq: torch.Tensor,
+                   u_count: Optional[torch.Tensor]):
+        """uniq (sorted, possibly nnz-padded with key 0) -> (send_keys
+        [world*cap + 1] with per-owner buckets zero-padded, idx [len(uniq)]
+        mapping each unique to its send slot; invalid/overflow -> the dummy
+        tail slot).  Pure device ops — sync-free."""
+        world, cap = plan.a2a_world, plan.a2a_cap
+        n = uniq.numel()
+        ar = plan.a2a_ar[:n]
+        owner = _owner_of_keys(uniq, world)
+        if u_count is not None:
+            # padding tail (key 0) routes to the invalid bucket `world`
+            owner = torch.where(ar < u_count, owner,
+                                torch.full_like(owner, world))
+        counts = torch.bincount(owner, minlength=world + 1)
+        starts = torch.cumsum(counts, 0) - counts
+        pos = ar - starts.gather(0, owner)
+        dummy = world * cap
+        bad = (owner >= world) | (pos >= cap)
+        idx = torch.where(bad, torch.full_like(pos, dummy), owner * cap + pos)
+        send = torch.zeros(dummy + 1, dtype=torch.int64, device=uniq.device)
+        send.scatter_(0, idx, uniq)
+        self._a2a_overflow.add_((counts[:world] > cap).sum())
+        return send, idx
+
+    def _a2a_exchange_fwd(self, plan: "_GroupPlan", group: _GroupCtx,
+                          train: bool):
+        """Padded forward exchange: route keys -> even a2a -> owner lookup in
+        wire dtype -> even a2a back.  Returns (rows_full [world*cap+1, dim]
+        with a zero dummy tail row, idx).  Fills group.a2a_* for backward."""
+        world, cap = self._a2a_setup(plan)
+        send_keys, idx = self._a2a_route(plan, group.uniq_keys, group.u_count)
+        comm = self.dist
+        if world > 1:
+            recv_keys = comm.all_to_all_even(send_keys[: world * cap])
+        else:
+            recv_keys = send_keys[: world * cap].clone()
+        store = self.stores[group.dim]
+        rows_local = store.lookup_wire(recv_keys, train, self.wire_dtype)
+        rows_full = torch.empty(
+            world * cap + 1, group.dim, dtype=rows_local.dtype,
+            device=self.device,
+        )
+        if world > 1:
+            comm.all_to_all_even(rows_local, out=rows_full[: world * cap])
+        else:
+            rows_full[: world * cap].copy_(rows_local)
+        rows_full[world * cap].zero_()
+        group.a2a_plan = plan
+        group.a2a_idx = idx
+        group.a2a_recv_keys = recv_keys
+        if self.device.type == "cuda":
+            # owner-side dedup for the backward merge, done HERE so the sort
+            # runs on the pipeline thread's stream (hidden under dense work)
+            from persia_amd.ops import native as _native
+
+            group.a2a_owner_dedup = tuple(_native().dedup_padded(recv_keys))
+        return rows_full, idx
+
+    def check_a2a_overflow(self) -> int:
+        """Batches whose per-owner bucket overflowed `cap` (their keys were
+        dropped to the dummy slot: rows read as zeros, grads discarded).
+        Non-zero means the slack heuristic failed — syncs the device."""
+        return int(self._a2a_overflow.sum().item())
 
     # ---------------------------------------------------------- forward path
 
@@ -530,7 +649,12 @@ class EmbeddingEngine:
             for i, f in enumerate(feats)
         ]
         store = self.stores[dim]
-        if not self.dist.distributed and store.spill is None:
+        dist_fast = (
+            (self.dist.distributed or self._force_dist)
+            and self._fused_dist
+            and store.spill is None
+        )
+        if not self.dist.distributed and store.spill is None and not self._force_dist:
             # whole lookup in ONE native call (C++ drives sign prep, dedup,
             # probe/insert, gather and the fused segment-sum)
             if self._prod_timing:
@@ -560,10 +684,9 @@ class EmbeddingEngine:
             return group
         else:
             keys_t = C.sign_prep(vals_t, plan.slot_starts, plan.prefixes, spacing_arg)
-            if not self.dist.distributed:
-                # spill path, single GPU: sync-free padded dedup (the spill
-                # restore phase has its own host consult, but the dedup needn't
-                # add two more stream syncs)
+            if dist_fast or not self.dist.distributed:
+                # sync-free padded dedup: both the fused distributed exchange
+                # and the single-GPU spill path read the unique count on-device
                 uniq_keys, inverse, perm, ustarts, u_count = C.dedup_padded(keys_t)
             else:
                 uniq_keys, inverse, perm, ustarts = _dedup(keys_t)
@@ -574,6 +697,22 @@ class EmbeddingEngine:
                 cat_offsets=plan.cat_offsets, seg_id=plan.seg_id,
                 n_sum_slots=plan.S,
             )
+            if dist_fast:
+                # capacity-padded even a2a (no host count syncs anywhere);
+                # compose the unpack gather into `inverse` so segment_sum
+                # reads the recv buffer directly — no [nnz, dim] row gather
+                rows_full, idx = self._a2a_exchange_fwd(plan, group, train)
+                inverse2 = idx.gather(0, inverse)
+                sums = C.segment_sum(
+                    rows_full, inverse2, plan.cat_offsets, plan.empty_scale
+                )
+                group.sum_base = sums
+                out._lazy_sum_groups.append(
+                    (group, [sc.name for sc in slot_ctxs], B)
+                )
+                if self._prod_timing:
+                    self._pt["native"] += time.perf_counter() - _t0
+                return group
             rows = self._exchange_rows(group, train)
             sums = C.segment_sum(
                 rows.contiguous(), inverse, plan.cat_offsets, plan.empty_scale
@@ -686,7 +825,36 @@ class EmbeddingEngine:
             dim=dim, uniq_keys=uniq_keys, inverse=inverse, perm=perm,
             ustarts=ustarts, slots=slot_ctxs,
         )
-        rows = self._exchange_rows(group, train)  # [U, dim]
+        # CPU mirror of the GPU fused distributed path: same capacity-padded
+        # even-a2a routing math, exact (unpadded) dedup — this is what the
+        # gloo multi-process tests exercise so the 8-GPU driver run's code
+        # path has CPU coverage
+        use_padded = (
+            not native
+            and (self.dist.distributed or self._force_dist)
+            and self._fused_dist
+            and self.stores[dim].spill is None
+            and all(getattr(f, "is_single", False) for f in feats)
+            and all(
+                sc.cfg.embedding_summation and sc.cfg.hash_stack_rounds == 0
+                for sc in slot_ctxs
+            )
+        )
+        if use_padded:
+            B = out.batch_size
+            names = tuple(sc.name for sc in slot_ctxs)
+            plan = self._plans.get((dim, names, B))
+            if plan is None:
+                prefixes_np = np.array(
+                    [self.schema.get_slot(n).index_prefix for n in names],
+                    dtype=np.uint64,
+                )
+                plan = _GroupPlan(dim, names, prefixes_np, B, dev)
+                self._plans[(dim, names, B)] = plan
+            rows_full, idx = self._a2a_exchange_fwd(plan, group, train)
+            rows = rows_full[idx]
+        else:
+            rows = self._exchange_rows(group, train)  # [U, dim]
 
         # ---- fused sum-slot postprocess
         sum_slots = [sc for sc in slot_ctxs if sc.cfg.embedding_summation]
@@ -783,6 +951,13 @@ class EmbeddingEngine:
             buf = torch.zeros(U, group.dim, dtype=torch.float32, device=self.device)
             any_grad = False
             sum_slots = [sc for sc in group.slots if sc.cfg.embedding_summation]
+            # all-None slot grads: skip the fused launch entirely so a fully
+            # skipped group never applies a zero-gradient optimizer step
+            # (reference skip-the-slot semantics, mod.rs:731-746)
+            if native and sum_slots and not any(
+                grads.get(sc.name) is not None for sc in sum_slots
+            ):
+                sum_slots = []
             if native and sum_slots:
                 # fused ordered scatter over ALL sum slots in one launch.
                 # Per-slot NaN skip (reference mod.rs:731-746) without a host
@@ -850,8 +1025,75 @@ class EmbeddingEngine:
                 continue
             self._route_and_update(group, buf)
 
+    def _a2a_backward_native(self, group: _GroupCtx, gbase: torch.Tensor,
+                             seg_scale: torch.Tensor, store, C) -> None:
+        """GPU fused distributed backward: per-uniq ordered grad reduction
+        written directly into the padded a2a send layout (f16 wire), even
+        all-to-all on the grad communicator, then owner-side fused
+        dedup+scatter+optimizer.  No [U, dim] buffer, no host syncs."""
+        plan = group.a2a_plan
+        world, cap = plan.a2a_world, plan.a2a_cap
+        send_g = torch.empty(
+            world * cap + 1, group.dim, dtype=torch.float16, device=self.device
+        )
+        uc = group.u_count if group.u_count is not None else self._empty_i64()
+        C.grad_scatter_idx(
+            gbase.contiguous(), group.perm, group.ustarts, group.seg_id,
+            seg_scale, group.a2a_idx, send_g, uc,
+        )
+        if world > 1:
+            recv_g = self.dist_grad.all_to_all_even(send_g[: world * cap])
+        else:
+            recv_g = send_g[: world * cap]
+        # positions whose recv key is 0 (bucket padding) were never written:
+        # scatter_update skips them on the key, before reading the grads
+        ou, _oinv, operm, oustarts, ou_count = group.a2a_owner_dedup
+        powers = store._adam_step_powers()
+        b1p, b2p = powers if powers else (0.0, 0.0)
+        C.scatter_update(
+            store.keys, store.ticks, store.arena, ou, recv_g, operm,
+            oustarts, plan.owner_seg_id, self._empty_f32(), group.dim,
+            store._opt_code, store._opt_params(), float(b1p), float(b2p),
+            float(self.hyper.weight_bound), store._skipped, ou_count,
+        )
+
+    def _route_and_update_padded(self, group: _GroupCtx, buf: torch.Tensor) -> None:
+        """CPU mirror of the padded backward (the gloo-tested twin of
+        _a2a_backward_native): pack per-uniq grads into the fixed bucket
+        layout, even a2a, owner-side merge, ONE optimizer application per
+        sign (sum-first, deterministic — same semantics as
+        _route_and_update)."""
+        plan = group.a2a_plan
+        world, cap = plan.a2a_world, plan.a2a_cap
+        store = self.stores[group.dim]
+        send_g = torch.zeros(
+            world * cap + 1, group.dim, dtype=self.wire_dtype,
+            device=self.device,
+        )
+        # buf is exact [U, dim] (a materialized group); idx may be nnz-padded
+        # — its first U entries are the valid uniques
+        send_g.index_copy_(
+            0, group.a2a_idx[: buf.shape[0]], buf.to(self.wire_dtype)
+        )
+        if world > 1:
+            recv_g = self.dist_grad.all_to_all_even(send_g[: world * cap])
+        else:
+            recv_g = send_g[: world * cap]
+        rk = group.a2a_recv_keys
+        m = rk != 0
+        if not bool(m.any()):
+            return
+        uniq_f, inv = torch.unique(rk[m] ^ _FLIP, sorted=True, return_inverse=True)
+        merged = torch.zeros(
+            uniq_f.numel(), group.dim, dtype=torch.float32, device=self.device
+        )
+        merged.index_add_(0, inv, recv_g[m].float())
+        self.skipped_grad_signs += store.update_gradients(uniq_f ^ _FLIP, merged)
+
     def _route_and_update(self, group: _GroupCtx, buf: torch.Tensor) -> None:
         store = self.stores[group.dim]
+        if group.a2a_idx is not None:
+            return self._route_and_update_padded(group, buf)
         if not self.dist.distributed:
             self.skipped_grad_signs += store.update_gradients(group.uniq_keys, buf)
             return
@@ -937,6 +1179,17 @@ class EmbeddingEngine:
                         )
                 store = self.stores[group.dim]
                 if (
+                    group.a2a_idx is not None
+                    and not raw_grads
+                    and hasattr(store, "_opt_code")
+                    and self.wire_dtype == torch.float16
+                ):
+                    # fused distributed backward: indexed scatter straight
+                    # into the a2a send layout, even a2a, owner-side fused
+                    # dedup+optimizer — zero host syncs
+                    self._a2a_backward_native(group, gbase, seg_scale, store, C)
+                    continue
+                if (
                     not self.dist.distributed
                     and not raw_grads
                     and hasattr(store, "_opt_code")  # HipEmbeddingStore
@@ -966,6 +1219,11 @@ class EmbeddingEngine:
                     group.seg_id, seg_scale, buf, 0,
                 )
             else:
+                if not raw_grads:
+                    # no sum-base grad and no raw grads: skip the group (a
+                    # zero-grad optimizer application would still move Adam /
+                    # Adagrad state — reference skips the slot instead)
+                    continue
                 self._materialize_group(group)
                 buf = torch.zeros(
                     group.uniq_keys.numel(), group.dim,

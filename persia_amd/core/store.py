@@ -193,6 +193,14 @@ class EmbeddingStoreBase:
         longer resident. Returns number skipped."""
         raise NotImplementedError
 
+    def lookup_wire(self, keys: torch.Tensor, train: bool,
+                    wire_dtype: torch.dtype) -> torch.Tensor:
+        """Owner-side lookup for the padded a2a exchange: ``keys`` may contain
+        the empty-key sentinel 0 (bucket padding) — those entries are silently
+        skipped (zeros out, no insert).  Rows come back in the WIRE dtype so
+        they go straight onto the xGMI all-to-all."""
+        raise NotImplementedError
+
     def export_rows(self) -> Tuple[np.ndarray, np.ndarray]:
         """-> (signs u64[n], inner f32[n, row_width]) of every resident row."""
         raise NotImplementedError
@@ -337,6 +345,16 @@ class CpuEmbeddingStore(EmbeddingStoreBase):
                     self.ticks[slot] = tick
                     out[i] = self.arena[slot, : self.dim]
         return out
+
+    def lookup_wire(self, keys: torch.Tensor, train: bool,
+                    wire_dtype: torch.dtype) -> torch.Tensor:
+        out = torch.zeros(keys.numel(), self.dim, dtype=torch.float32)
+        m = keys != 0
+        if bool(m.any()):
+            out[m] = self.lookup(keys[m], train)
+        else:
+            self.next_tick()  # tick parity with the non-empty case
+        return out.to(wire_dtype)
 
     def update_gradients(self, keys: torch.Tensor, grads: torch.Tensor) -> int:
         powers = self._adam_step_powers()
@@ -554,6 +572,25 @@ class HipEmbeddingStore(EmbeddingStoreBase):
             uc,
         )
         self._drain_evictions(ev)
+        return out
+
+    def lookup_wire(self, keys: torch.Tensor, train: bool,
+                    wire_dtype: torch.dtype) -> torch.Tensor:
+        """Padded-a2a owner lookup: key 0 entries (bucket padding) return
+        zeros and never claim slots (kernel-level skip); output is the wire
+        dtype directly (fused f16 cast in init_gather)."""
+        tick = self.next_tick()
+        n = keys.numel()
+        out = torch.empty(n, self.dim, dtype=wire_dtype, device=self.device)
+        lo, hi = self.hyper.emb_initialization
+        self._C.store_lookup(
+            self.keys, self.ticks, self.arena, keys, out, self.dim,
+            int(train), tick, float(lo), float(hi),
+            float(self.hyper.admit_probability),
+            float(self.optimizer.state_init(self.dim)), self.opt_space,
+            *self._no_evict,
+            torch.empty(0, dtype=torch.int64, device=self.device),
+        )
         return out
 
     def update_gradients(self, keys: torch.Tensor, grads: torch.Tensor) -> int:
