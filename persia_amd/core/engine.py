@@ -624,10 +624,13 @@ class EmbeddingEngine:
             self._plans[(dim, names, B)] = plan
         # per-batch pinned staging (allocated once per PersiaBatch, like a
         # loader-side pinned pool): repeat visits are a pure async H2D
-        pcache = getattr(src_batch, "_pinned_cache", None) if src_batch is not None else None
-        if pcache is not None and dim in pcache:
-            vals_t = pcache[dim].to(dev, non_blocking=True)
-        else:
+        def _upload():
+            pcache = (
+                getattr(src_batch, "_pinned_cache", None)
+                if src_batch is not None else None
+            )
+            if pcache is not None and dim in pcache:
+                return pcache[dim].to(dev, non_blocking=True)
             values_np = np.concatenate([f.values for f in feats])
             if src_batch is not None:
                 if pcache is None:
@@ -635,9 +638,9 @@ class EmbeddingEngine:
                 pin = torch.empty(len(values_np), dtype=torch.int64, pin_memory=True)
                 pin.numpy()[:] = values_np.view(np.int64)
                 pcache[dim] = pin
-                vals_t = pin.to(dev, non_blocking=True)
-            else:
-                vals_t = self._upload_values(dim, values_np)
+                return pin.to(dev, non_blocking=True)
+            return self._upload_values(dim, values_np)
+
         spacing = self.schema.feature_spacing
         spacing_arg = spacing if spacing < (1 << 63) else -1
         slot_ctxs = [
@@ -654,6 +657,13 @@ class EmbeddingEngine:
             and self._fused_dist
             and store.spill is None
         )
+        # spill-tier front half on a dedicated probe stream: the miss-mask
+        # readback in spill_restore then waits only for this batch's tiny
+        # probe work, not the lookup stream's multi-batch backlog (the
+        # dcn-spill ~1.9 ms/batch producer stall — round-1 profiles finding)
+        spill_front = (
+            store.spill is not None and not self.dist.distributed and train
+        )
         if not self.dist.distributed and store.spill is None and not self._force_dist:
             # whole lookup in ONE native call (C++ drives sign prep, dedup,
             # probe/insert, gather and the fused segment-sum)
@@ -661,6 +671,7 @@ class EmbeddingEngine:
                 self._pt["prep"] += time.perf_counter() - _t0
                 _t1 = time.perf_counter()
             lo, hi = self.hyper.emb_initialization
+            vals_t = _upload()
             sums, uniq_keys, inverse, perm, ustarts, u_count = C.lookup_local(
                 vals_t, plan.slot_starts, plan.prefixes, spacing_arg,
                 plan.cat_offsets, plan.empty_scale,
@@ -683,14 +694,36 @@ class EmbeddingEngine:
                 self._pt["native"] += time.perf_counter() - _t1
             return group
         else:
-            keys_t = C.sign_prep(vals_t, plan.slot_starts, plan.prefixes, spacing_arg)
-            if dist_fast or not self.dist.distributed:
-                # sync-free padded dedup: both the fused distributed exchange
-                # and the single-GPU spill path read the unique count on-device
-                uniq_keys, inverse, perm, ustarts, u_count = C.dedup_padded(keys_t)
+            if spill_front:
+                ps = store.probe_stream
+                cur = torch.cuda.current_stream()
+                with torch.cuda.stream(ps):
+                    vals_t = _upload()
+                    keys_t = C.sign_prep(
+                        vals_t, plan.slot_starts, plan.prefixes, spacing_arg
+                    )
+                    uniq_keys, inverse, perm, ustarts, u_count = C.dedup_padded(
+                        keys_t
+                    )
+                    store.spill_restore(uniq_keys, u_count)
+                cur.wait_stream(ps)
+                for t in (vals_t, uniq_keys, inverse, perm, ustarts, u_count):
+                    t.record_stream(cur)
             else:
-                uniq_keys, inverse, perm, ustarts = _dedup(keys_t)
-                u_count = None
+                vals_t = _upload()
+                keys_t = C.sign_prep(
+                    vals_t, plan.slot_starts, plan.prefixes, spacing_arg
+                )
+                if dist_fast or not self.dist.distributed:
+                    # sync-free padded dedup: both the fused distributed
+                    # exchange and the single-GPU spill path read the unique
+                    # count on-device
+                    uniq_keys, inverse, perm, ustarts, u_count = C.dedup_padded(
+                        keys_t
+                    )
+                else:
+                    uniq_keys, inverse, perm, ustarts = _dedup(keys_t)
+                    u_count = None
             group = _GroupCtx(
                 dim=dim, uniq_keys=uniq_keys, inverse=inverse, perm=perm,
                 ustarts=ustarts, u_count=u_count, slots=slot_ctxs,

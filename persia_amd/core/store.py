@@ -447,6 +447,8 @@ class HipEmbeddingStore(EmbeddingStoreBase):
         )
         self._skipped = torch.zeros(2, dtype=torch.int32, device=device)  # miss, nan
         self._opt_code = {"sgd": 0, "adagrad": 1, "adam": 2}[optimizer.kind]
+        self._probe_stream: Optional[torch.cuda.Stream] = None
+        self._restore_tick: Optional[int] = None
         self._no_evict = (
             torch.empty(0, dtype=torch.int64, device=device),
             torch.empty(0, dtype=torch.int32, device=device),
@@ -516,6 +518,53 @@ class HipEmbeddingStore(EmbeddingStoreBase):
     def __len__(self) -> int:
         return int((self.keys != 0).sum().item())
 
+    @property
+    def probe_stream(self) -> "torch.cuda.Stream":
+        """Side stream for the spill-restore front half: its backlog is only
+        this batch's tiny probe work, so the miss-mask readback does not wait
+        for the lookup stream's multi-batch queue (the dcn-spill producer
+        stall — profiles/README round-1 finding)."""
+        if self._probe_stream is None:
+            self._probe_stream = torch.cuda.Stream(device=self.device)
+        return self._probe_stream
+
+    def spill_restore(self, keys: torch.Tensor,
+                      u_count: Optional[torch.Tensor] = None) -> None:
+        """Phase 1 of a training lookup with a spill tier, split out so the
+        engine can run it on ``probe_stream``: probe for missing-but-spilled
+        rows, fetch them from the host tier and import into HBM.  The
+        following ``lookup`` call reuses this tick and skips its own phase 1.
+
+        Probes may race in-flight inserts/evictions on the lookup stream;
+        both races degrade to a cold-key re-init or a no-op fetch — the
+        bounded-staleness tolerance the table already guarantees."""
+        tick = self.next_tick()
+        self._restore_tick = tick
+        if not len(self.spill) or keys.numel() == 0:
+            return
+        uc = (
+            u_count
+            if u_count is not None
+            else torch.empty(0, dtype=torch.int64, device=self.device)
+        )
+        self._drain_pending()
+        slots = self._C.store_probe(self.keys, self.ticks, keys, tick, uc)
+        miss_keys = keys[slots < 0]  # syncs ONLY the caller's (probe) stream
+        if miss_keys.numel():
+            miss_np = miss_keys.cpu().numpy().view(np.uint64)
+            rows, found = self.spill.fetch(miss_np)
+            if found.any():
+                ev = self._evict_buffers(int(found.sum()))
+                found_keys = torch.from_numpy(
+                    miss_np[found].view(np.int64).copy()
+                ).to(self.device)
+                rows_t = torch.from_numpy(rows[found]).to(self.device)
+                self._C.store_import(
+                    self.keys, self.ticks, self.arena, found_keys, rows_t,
+                    tick, *ev,
+                )
+                self._drain_evictions(ev)
+
     def lookup(
         self, keys: torch.Tensor, train: bool,
         u_count: Optional[torch.Tensor] = None,
@@ -523,7 +572,12 @@ class HipEmbeddingStore(EmbeddingStoreBase):
         """``u_count``: optional device-side i64[1] valid-prefix length for
         nnz-padded ``keys`` (sync-free dedup) — the padding tail is the
         empty-key sentinel and is never probed or claimed."""
-        tick = self.next_tick()
+        restored = self._restore_tick is not None
+        if restored:
+            tick = self._restore_tick
+            self._restore_tick = None
+        else:
+            tick = self.next_tick()
         n = keys.numel()
         uc = (
             u_count
@@ -534,8 +588,9 @@ class HipEmbeddingStore(EmbeddingStoreBase):
         if self.spill is not None and train and n:
             self._drain_pending()
             ev = self._evict_buffers(n)
-        if self.spill is not None and train and n and len(self.spill):
-            # spill phase 1: restore missing-but-spilled rows into HBM first
+        if (not restored and self.spill is not None and train and n
+                and len(self.spill)):
+            # spill phase 1 inline (engine did not pre-restore)
             slots = self._C.store_probe(self.keys, self.ticks, keys, tick, uc)
             miss_keys = keys[slots < 0]
             if miss_keys.numel():
