@@ -163,6 +163,7 @@ class _GroupPlan:
         self.a2a_world: Optional[int] = None
         self.a2a_cap: Optional[int] = None
         self.a2a_ar: Optional[torch.Tensor] = None  # arange(nnz)
+        self.a2a_ones: Optional[torch.Tensor] = None  # ones(nnz) count weights
         self.owner_seg_id: Optional[torch.Tensor] = None  # arange(world*cap)
 
 
@@ -454,6 +455,7 @@ class EmbeddingEngine:
         plan.a2a_world = world
         plan.a2a_cap = cap
         plan.a2a_ar = torch.arange(nnz, dtype=torch.int64, device=self.device)
+        plan.a2a_ones = torch.ones(nnz, dtype=torch.int64, device=self.device)
         plan.owner_seg_id = torch.arange(
             world * cap, dtype=torch.int64, device=self.device
         )
@@ -473,7 +475,11 @@ class EmbeddingEngine:
             # padding tail (key 0) routes to the invalid bucket `world`
             owner = torch.where(ar < u_count, owner,
                                 torch.full_like(owner, world))
-        counts = torch.bincount(owner, minlength=world + 1)
+        # NOT torch.bincount: its CUDA kernel sizes the histogram via
+        # input.max().item() — a hidden device sync that would stall the
+        # producer thread behind the whole lookup-stream backlog
+        counts = torch.zeros(world + 1, dtype=torch.int64, device=uniq.device)
+        counts.scatter_add_(0, owner, plan.a2a_ones[:n])
         starts = torch.cumsum(counts, 0) - counts
         pos = ar - starts.gather(0, owner)
         dummy = world * cap
@@ -510,12 +516,10 @@ class EmbeddingEngine:
         group.a2a_plan = plan
         group.a2a_idx = idx
         group.a2a_recv_keys = recv_keys
-        if self.device.type == "cuda":
-            # owner-side dedup for the backward merge, done HERE so the sort
-            # runs on the pipeline thread's stream (hidden under dense work)
-            from persia_amd.ops import native as _native
-
-            group.a2a_owner_dedup = tuple(_native().dedup_padded(recv_keys))
+        # owner-side dedup for the backward merge happens lazily in
+        # _a2a_backward_native: the pipeline (producer) thread is the scaling
+        # bottleneck at world=8, the main thread has slack — and infer-mode
+        # batches never need it
         return rows_full, idx
 
     def check_a2a_overflow(self) -> int:
@@ -563,6 +567,42 @@ class EmbeddingEngine:
             self._pt["batch"] += time.perf_counter() - _tb0
             self._pt["n"] += 1
         return out
+
+    def prepare_host_batch(self, batch: PersiaBatch) -> None:
+        """Build a batch's pinned staging buffers host-side, without touching
+        the GPU streams.  The first visit of a batch otherwise pays
+        hipHostMalloc + copy inside the pipeline thread — ms-scale when the
+        host is still reclaiming a prior process's pinned pages, which
+        dominates short timed runs.  Idempotent; safe to call from the data
+        producer (the reference allocates its pinned pools at startup for the
+        same reason, cuda/pinned_memory_pool.rs)."""
+        if self.device.type != "cuda":
+            return
+        pc = getattr(batch, "_pinned_cache", None)
+        if pc is None:
+            pc = batch._pinned_cache = {}
+        for kind, items in (("nid", batch.non_id_type_features),
+                            ("lab", batch.labels)):
+            for i, x in enumerate(items):
+                if (kind, i) in pc:
+                    continue
+                src = torch.from_numpy(np.ascontiguousarray(x.data))
+                t = torch.empty_like(src, pin_memory=True)
+                t.copy_(src)
+                pc[(kind, i)] = t
+        feats_by_dim: Dict[int, List] = {}
+        for feat in batch.id_type_features:
+            cfg = self.schema.get_slot(feat.name)
+            feats_by_dim.setdefault(cfg.dim, []).append(feat)
+        for dim, feats in feats_by_dim.items():
+            if dim in pc or not all(
+                getattr(f, "is_single", False) for f in feats
+            ):
+                continue
+            values_np = np.concatenate([f.values for f in feats])
+            pin = torch.empty(len(values_np), dtype=torch.int64, pin_memory=True)
+            pin.numpy()[:] = values_np.view(np.int64)
+            pc[dim] = pin
 
     def _upload_aux(self, batch, cache_key, arr: np.ndarray) -> torch.Tensor:
         """Dense-feature/label upload: pageable H2D stalls the pipeline
@@ -1080,6 +1120,8 @@ class EmbeddingEngine:
             recv_g = send_g[: world * cap]
         # positions whose recv key is 0 (bucket padding) were never written:
         # scatter_update skips them on the key, before reading the grads
+        if group.a2a_owner_dedup is None:
+            group.a2a_owner_dedup = tuple(C.dedup_padded(group.a2a_recv_keys))
         ou, _oinv, operm, oustarts, ou_count = group.a2a_owner_dedup
         powers = store._adam_step_powers()
         b1p, b2p = powers if powers else (0.0, 0.0)
