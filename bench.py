@@ -469,9 +469,14 @@ def main():
         pt = engine._pt
         if pt["n"]:
             pn = pt["n"]
+            extra = "".join(
+                f" {k}={pt[k]/pn*1e3:.3f}"
+                for k in ("dedup", "exch", "route", "lkw", "sum") if k in pt
+            )
             print(
                 f"# producer ms/batch: total={pt['batch']/pn*1e3:.3f} "
-                f"prep={pt['prep']/pn*1e3:.3f} native={pt['native']/pn*1e3:.3f}",
+                f"prep={pt['prep']/pn*1e3:.3f} native={pt['native']/pn*1e3:.3f}"
+                + extra,
                 file=_s.stderr, flush=True,
             )
     samples = args.steps * B * world

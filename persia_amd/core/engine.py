@@ -496,14 +496,25 @@ class EmbeddingEngine:
         wire dtype -> even a2a back.  Returns (rows_full [world*cap+1, dim]
         with a zero dummy tail row, idx).  Fills group.a2a_* for backward."""
         world, cap = self._a2a_setup(plan)
+        if self._prod_timing:
+            _tr = time.perf_counter()
         send_keys, idx = self._a2a_route(plan, group.uniq_keys, group.u_count)
         comm = self.dist
         if world > 1:
             recv_keys = comm.all_to_all_even(send_keys[: world * cap])
         else:
             recv_keys = send_keys[: world * cap].clone()
+        if self._prod_timing:
+            self._pt["route"] = self._pt.get("route", 0.0) + (
+                time.perf_counter() - _tr
+            )
+            _tr = time.perf_counter()
         store = self.stores[group.dim]
         rows_local = store.lookup_wire(recv_keys, train, self.wire_dtype)
+        if self._prod_timing:
+            self._pt["lkw"] = self._pt.get("lkw", 0.0) + (
+                time.perf_counter() - _tr
+            )
         rows_full = torch.empty(
             world * cap + 1, group.dim, dtype=rows_local.dtype,
             device=self.device,
@@ -774,11 +785,25 @@ class EmbeddingEngine:
                 # capacity-padded even a2a (no host count syncs anywhere);
                 # compose the unpack gather into `inverse` so segment_sum
                 # reads the recv buffer directly — no [nnz, dim] row gather
+                if self._prod_timing:
+                    self._pt["dedup"] = self._pt.get("dedup", 0.0) + (
+                        time.perf_counter() - _t0
+                    )
+                    _tx = time.perf_counter()
                 rows_full, idx = self._a2a_exchange_fwd(plan, group, train)
+                if self._prod_timing:
+                    self._pt["exch"] = self._pt.get("exch", 0.0) + (
+                        time.perf_counter() - _tx
+                    )
+                    _tx = time.perf_counter()
                 inverse2 = idx.gather(0, inverse)
                 sums = C.segment_sum(
                     rows_full, inverse2, plan.cat_offsets, plan.empty_scale
                 )
+                if self._prod_timing:
+                    self._pt["sum"] = self._pt.get("sum", 0.0) + (
+                        time.perf_counter() - _tx
+                    )
                 group.sum_base = sums
                 out._lazy_sum_groups.append(
                     (group, [sc.name for sc in slot_ctxs], B)
