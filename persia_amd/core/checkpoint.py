@@ -14,9 +14,12 @@ documented little-endian record format (the speedy submodule is not
 verifiable — SURVEY §5 Checkpoint note):
 
     file   := magic "PAEMB1\\0\\0" | u64 num_records | u64 embedding_dim
-              | u64 row_width | records...
-    record := u64 sign | f32 inner[row_width]      (inner = emb ‖ opt_state,
-              reference emb_entry.rs:17-23 layout)
+              | u64 row_width
+              | u64 signs[num_records]                  (columnar: all signs)
+              | f32 inner[num_records * row_width]      (then all rows)
+    inner row = emb(dim) ‖ opt_state — reference emb_entry.rs:17-23 layout.
+    All fields little-endian; the two arrays are contiguous blocks, NOT
+    interleaved per record.
 
 Re-sharding on load when the dumped shard count differs from the current
 world size mirrors mod.rs:1150-1259: every rank scans all files and keeps the

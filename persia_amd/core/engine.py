@@ -1362,7 +1362,6 @@ class EmbeddingEngine:
 
         def run():
             try:
-                self._set_status("Dumping", 0.0)
                 dump_embedding(self, dst_dir)
                 self._set_status("Idle", 100.0)
             except Exception as e:
@@ -1371,6 +1370,9 @@ class EmbeddingEngine:
                 if blocking:
                     raise
 
+        # status set BEFORE the worker starts: a wait_for_emb_dumping issued
+        # right after a non-blocking dump must never observe a stale 'Idle'
+        self._set_status("Dumping", 0.0)
         if blocking:
             run()
         else:
@@ -1381,7 +1383,6 @@ class EmbeddingEngine:
 
         def run():
             try:
-                self._set_status("Loading", 0.0)
                 load_embedding(self, src_dir)
                 self._set_status("Idle", 100.0)
             except Exception as e:
@@ -1390,6 +1391,7 @@ class EmbeddingEngine:
                 if blocking:
                     raise
 
+        self._set_status("Loading", 0.0)  # before the thread (see dump)
         if blocking:
             run()
         else:

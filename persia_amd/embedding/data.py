@@ -43,34 +43,46 @@ _CODE_DTYPE = {v: k for k, v in _DTYPE_CODE.items()}
 
 
 def _id_type_data_check(data: np.ndarray, name: str) -> None:
-    assert isinstance(
-        data, np.ndarray
-    ), f"expected id_type_feature: {name} type to be np.ndarray but got type: {type(data)}"
-    assert (
-        data.ndim == 1
-    ), f"expected id_type_feature: {name} ndim equal to one but got ndim: {data.ndim}"
-    assert (
-        data.dtype == np.uint64
-    ), f"expected id_type_feature: {name} dtype as np.uint64 but got dtype: {data.dtype}"
+    """Same validation semantics as the reference (persia/embedding/data.py:
+    21-40): per-sample ID lists must be 1-D uint64 numpy arrays."""
+    assert isinstance(data, np.ndarray), (
+        f"id_type_feature {name!r}: each sample must be a numpy ndarray, "
+        f"not {type(data).__name__}"
+    )
+    assert data.ndim == 1, (
+        f"id_type_feature {name!r}: samples are 1-D ID lists, got a "
+        f"{data.ndim}-D array"
+    )
+    assert data.dtype == np.uint64, (
+        f"id_type_feature {name!r}: IDs must be uint64 (sign space), got "
+        f"{data.dtype}"
+    )
 
 
 def _ndarray_check(data: np.ndarray, name: str) -> None:
-    assert isinstance(
-        data, np.ndarray
-    ), f"input data {name}, type: {type(data)} no match numpy ndarray"
-    assert (
-        data.dtype.type in _ND_ARRAY_SUPPORT_TYPE
-    ), f"np.array only support dtype {_ND_ARRAY_SUPPORT_TYPE} but got {name} dtype {data.dtype}"
-    assert data.ndim > 0, f"{name} ndarray got ndim: {data.ndim} expect ndim greater than zero"
+    """Dense-tensor validation (reference persia/embedding/data.py:41-56)."""
+    assert isinstance(data, np.ndarray), (
+        f"{name!r}: dense data must be a numpy ndarray, not "
+        f"{type(data).__name__}"
+    )
+    assert data.dtype.type in _ND_ARRAY_SUPPORT_TYPE, (
+        f"{name!r}: dtype {data.dtype} is not wire-serializable; supported: "
+        f"{sorted(t.__name__ for t in _ND_ARRAY_SUPPORT_TYPE)}"
+    )
+    assert data.ndim > 0, f"{name!r}: dense data cannot be 0-dimensional"
 
 
 def _batch_size_check(batch_size: int, target: int, data_type: str, name: str) -> None:
-    assert (
-        batch_size == target
-    ), f"expected {data_type}: {name} batch_size equal to {target} but got {batch_size}"
-    assert (
-        batch_size <= MAX_BATCH_SIZE
-    ), f"expected {data_type}:{name} batch_size <= MAX_BATCH_SIZE: {MAX_BATCH_SIZE} but got {batch_size}"
+    """Every component of a PersiaBatch must agree on the sample count, which
+    is capped by the u16 wire coordinate (reference data.py:57-68)."""
+    assert batch_size == target, (
+        f"{data_type} {name!r}: has {batch_size} samples but the batch has "
+        f"{target}"
+    )
+    assert batch_size <= MAX_BATCH_SIZE, (
+        f"{data_type} {name!r}: {batch_size} samples exceeds the "
+        f"u16-coordinate cap of {MAX_BATCH_SIZE}"
+    )
 
 
 class IDTypeFeature:
