@@ -349,6 +349,32 @@ class EmbeddingEngine:
 
         self.metrics_enabled = bool(gconf.enable_metrics)
         self.metrics = EngineMetrics(self.metrics_enabled)
+        # hot-path observability: sampled every PA_METRICS_EVERY batches so
+        # the flagship loop never pays for timers/copies it does not use;
+        # GPU-side values (event timings, device counters) are harvested one
+        # sample late through pinned async copies — no stream syncs
+        self._metrics_every = max(1, int(_os.environ.get("PA_METRICS_EVERY", "64")))
+        self._batch_counter = 0
+        self._update_counter = 0
+        self._pending_metric = None
+        self._skipped_last = (0, 0)
+        self._evicted_last = 0
+        self.monitor = None
+        if self.metrics_enabled:
+            from persia_amd.core.monitor import DistinctIdMonitor
+
+            self.monitor = DistinctIdMonitor()
+        self.incremental = None
+
+    def enable_incremental_update(self, dst_dir: str,
+                                  buffer_size: int = 1_000_000):
+        """Attach an incremental-update dumper: every gradient update's
+        touched signs are recorded automatically and packets flush when the
+        dedup buffer fills (reference inc-update lib.rs:178-312)."""
+        from persia_amd.core.incremental import IncrementalUpdateDumper
+
+        self.incremental = IncrementalUpdateDumper(self, dst_dir, buffer_size)
+        return self.incremental
 
     def _empty_f32(self) -> torch.Tensor:
         if self._empty_scale is None:
@@ -544,6 +570,17 @@ class EmbeddingEngine:
     def process_batch(self, batch: PersiaBatch, train: Optional[bool] = None) -> PersiaTrainingBatch:
         if self._prod_timing:
             _tb0 = time.perf_counter()
+        self._batch_counter += 1
+        sample = (
+            self.metrics_enabled
+            and self._batch_counter % self._metrics_every == 1 % self._metrics_every
+        )
+        if sample:
+            _ms_t0 = time.perf_counter()
+            _ms_ev0 = None
+            if self.device.type == "cuda":
+                _ms_ev0 = torch.cuda.Event(enable_timing=True)
+                _ms_ev0.record()
         train = batch.requires_grad if train is None else train
         if self.gconf.job_type == "infer":
             train = False
@@ -574,10 +611,92 @@ class EmbeddingEngine:
         # lazily when .payloads is first materialized)
         out._order = {f.name: i for i, f in enumerate(batch.id_type_features)}
         out._sorted = False
+        if sample:
+            self._record_lookup_metrics(batch, out, _ms_t0, _ms_ev0)
         if self._prod_timing:
             self._pt["batch"] += time.perf_counter() - _tb0
             self._pt["n"] += 1
         return out
+
+    def _record_lookup_metrics(self, batch, out, t0: float, ev0) -> None:
+        """Sampled hot-path observability (reference worker/PS gauges,
+        embedding_worker_service/mod.rs:83-100, parameter mod.rs:27-79).
+        Device-side values are harvested one sample late (events/pinned
+        copies complete by then) so this never synchronizes a stream."""
+        m = self.metrics
+        # host producer time of this batch (prep + kernel issue + any waits)
+        m.lookup_preprocess_time_cost_sec.set(time.perf_counter() - t0)
+        # distinct-id estimates from the raw host-side id arrays
+        # (reference monitor.rs:29-114 samples the same point: pre-dedup ids)
+        if self.monitor is not None:
+            for feat in batch.id_type_features:
+                vals = getattr(feat, "values", None)
+                if vals is not None and len(vals):
+                    self.monitor.observe(feat.name, vals)
+            for name, est in self.monitor.estimates().items():
+                m.distinct_id_estimate.labels(name).set(est)
+        # unique-indices rate: exact on CPU; via the deferred u_count pinned
+        # copy on GPU (padded dedup keeps U device-side)
+        nnz = sum(
+            len(getattr(f, "values", ())) for f in batch.id_type_features
+        )
+        # harvest the PREVIOUS sample's deferred device-side values
+        pend = self._pending_metric
+        self._pending_metric = None
+        if pend is not None:
+            ev_a, ev_b, ucnt_pin, skip_pin, p_nnz, done = pend
+            if done.query():
+                if ev_a is not None and ev_b is not None:
+                    try:
+                        m.lookup_hashmap_time_cost_sec.set(
+                            ev_a.elapsed_time(ev_b) / 1e3
+                        )
+                    except Exception:
+                        pass
+                if ucnt_pin is not None and p_nnz:
+                    m.batch_unique_indices_rate.set(
+                        float(ucnt_pin.item()) / p_nnz
+                    )
+                if skip_pin is not None:
+                    miss, nan = int(skip_pin[0].item()), int(skip_pin[1].item())
+                    lm, ln = self._skipped_last
+                    if miss > lm:
+                        m.gradient_id_miss_count.inc(miss - lm)
+                    if nan > ln:
+                        m.nan_grad_skipped.inc(nan - ln)
+                    self._skipped_last = (miss, nan)
+        if self.device.type == "cuda":
+            ev1 = torch.cuda.Event(enable_timing=True)
+            ev1.record()
+            ucnt_pin = None
+            for g in out._groups:
+                if g.u_count is not None:
+                    ucnt_pin = torch.empty(1, dtype=torch.int64, pin_memory=True)
+                    ucnt_pin.copy_(g.u_count, non_blocking=True)
+                    break
+                # exact dedup: U known host-side without a sync
+                m.batch_unique_indices_rate.set(
+                    g.uniq_keys.numel() / max(1, nnz)
+                )
+            skip_pin = None
+            store = next(iter(self.stores.values()))
+            if hasattr(store, "_skipped"):
+                skip_pin = torch.empty(2, dtype=torch.int32, pin_memory=True)
+                skip_pin.copy_(store._skipped, non_blocking=True)
+            done = torch.cuda.Event()
+            done.record()
+            self._pending_metric = (ev0, ev1, ucnt_pin, skip_pin, nnz, done)
+        else:
+            U = sum(g.uniq_keys.numel() for g in out._groups)
+            m.batch_unique_indices_rate.set(U / max(1, nnz))
+            m.nan_grad_skipped.inc(0)
+        # eviction counter (host-side total maintained by the spill drain)
+        ev_tot = sum(
+            getattr(s, "evicted_total", 0) for s in self.stores.values()
+        )
+        if ev_tot > self._evicted_last:
+            m.evicted_count.inc(ev_tot - self._evicted_last)
+            self._evicted_last = ev_tot
 
     def prepare_host_batch(self, batch: PersiaBatch) -> None:
         """Build a batch's pinned staging buffers host-side, without touching
@@ -1038,6 +1157,14 @@ class EmbeddingEngine:
         """grads: slot name -> grad tensor ((B,dim) f16 for sum slots,
         (U_slot, dim) f32 for raw slots — the contract of ctx._on_backward,
         reference persia/ctx.py:926-1005) or None (skipped)."""
+        self._update_counter += 1
+        usample = (
+            self.metrics_enabled
+            and self._update_counter % self._metrics_every
+            == 1 % self._metrics_every
+        )
+        if usample:
+            _ut0 = time.perf_counter()
         native = self.device.type == "cuda"
         if native:
             from persia_amd.ops import native as _native
@@ -1122,6 +1249,10 @@ class EmbeddingEngine:
             if not any_grad:
                 continue
             self._route_and_update(group, buf)
+        if usample:
+            self.metrics.update_gradient_time_cost_sec.set(
+                time.perf_counter() - _ut0
+            )
 
     def _a2a_backward_native(self, group: _GroupCtx, gbase: torch.Tensor,
                              seg_scale: torch.Tensor, store, C) -> None:
@@ -1156,6 +1287,9 @@ class EmbeddingEngine:
             store._opt_code, store._opt_params(), float(b1p), float(b2p),
             float(self.hyper.weight_bound), store._skipped, ou_count,
         )
+        if self.incremental is not None:
+            # owner-side recording: this rank's shard's touched signs
+            self.incremental.record_keys(ou, ou_count)
 
     def _route_and_update_padded(self, group: _GroupCtx, buf: torch.Tensor) -> None:
         """CPU mirror of the padded backward (the gloo-tested twin of
@@ -1188,7 +1322,10 @@ class EmbeddingEngine:
             uniq_f.numel(), group.dim, dtype=torch.float32, device=self.device
         )
         merged.index_add_(0, inv, recv_g[m].float())
-        self.skipped_grad_signs += store.update_gradients(uniq_f ^ _FLIP, merged)
+        uk = uniq_f ^ _FLIP
+        self.skipped_grad_signs += store.update_gradients(uk, merged)
+        if self.incremental is not None:
+            self.incremental.record_keys(uk)
 
     def _route_and_update(self, group: _GroupCtx, buf: torch.Tensor) -> None:
         store = self.stores[group.dim]
@@ -1196,6 +1333,8 @@ class EmbeddingEngine:
             return self._route_and_update_padded(group, buf)
         if not self.dist.distributed:
             self.skipped_grad_signs += store.update_gradients(group.uniq_keys, buf)
+            if self.incremental is not None:
+                self.incremental.record_keys(group.uniq_keys)
             return
         comm = self.dist_grad  # main-thread communicator (see __init__)
         send_counts, recv_counts = group.send_counts, group.recv_counts
@@ -1216,7 +1355,10 @@ class EmbeddingEngine:
             uniq_f.numel(), group.dim, dtype=torch.float32, device=self.device
         )
         merged.index_add_(0, inv, grads_recv.float())
-        self.skipped_grad_signs += store.update_gradients(uniq_f ^ _FLIP, merged)
+        uk = uniq_f ^ _FLIP
+        self.skipped_grad_signs += store.update_gradients(uk, merged)
+        if self.incremental is not None:
+            self.incremental.record_keys(uk)
 
     def apply_gradients_base(
         self,
@@ -1233,6 +1375,14 @@ class EmbeddingEngine:
         ``raw_grads``: slot name -> (U_slot, dim) f32 for raw slots (the
         index_add_ de-dup output of ctx._on_backward)."""
         native = self.device.type == "cuda"
+        self._update_counter += 1
+        usample = (
+            self.metrics_enabled
+            and self._update_counter % self._metrics_every
+            == 1 % self._metrics_every
+        )
+        if usample:
+            _ut0 = time.perf_counter()
         if not native:
             # CPU: slice the base into the generic per-slot dict
             grads: Dict[str, Optional[torch.Tensor]] = dict(raw_grads or {})
@@ -1249,7 +1399,13 @@ class EmbeddingEngine:
                         gbase[b0 : b0 + nseg] if gbase is not None else None
                     )
                     b0 += nseg
-            return self.apply_gradients(training_batch, grads, loss_scale)
+            try:
+                return self.apply_gradients(training_batch, grads, loss_scale)
+            finally:
+                if usample:
+                    self.metrics.update_gradient_time_cost_sec.set(
+                        time.perf_counter() - _ut0
+                    )
 
         from persia_amd.ops import native as _native
 
@@ -1308,6 +1464,10 @@ class EmbeddingEngine:
                         if group.u_count is not None
                         else self._empty_i64(),
                     )
+                    if self.incremental is not None:
+                        self.incremental.record_keys(
+                            group.uniq_keys, group.u_count
+                        )
                     continue
                 self._materialize_group(group)
                 buf = torch.empty(
@@ -1348,6 +1508,12 @@ class EmbeddingEngine:
                         gf = gf / float(np.sqrt(rounds))
                     buf.index_add_(0, sc.slot_uniq_global, gf)
             self._route_and_update(group, buf)
+        if usample:
+            # host-side issue+wait time of the whole gradient push (the
+            # reference's update_gradient_time_cost_sec, mod.rs:83-100)
+            self.metrics.update_gradient_time_cost_sec.set(
+                time.perf_counter() - _ut0
+            )
 
     # ------------------------------------------------------------ checkpoint
 

@@ -29,7 +29,14 @@ DONE_MARKER = "inc_update_done"
 
 
 class IncrementalUpdateDumper:
-    """Train-side: collect touched signs, flush packets."""
+    """Train-side: collect the signs touched by gradient updates, flush
+    packets automatically when the dedup buffer fills (reference
+    lib.rs:178-312 buffers inside update_gradient the same way).
+
+    The engine calls :meth:`record_keys` from its update paths with the
+    DEVICE key tensor of the batch; the copy back to the host is asynchronous
+    (pinned staging + event, harvested on later calls), so recording costs
+    the hot loop one small async D2H, never a stream sync."""
 
     def __init__(self, engine: "EmbeddingEngine", dst_dir: str,
                  buffer_size: int = 1_000_000):
@@ -39,8 +46,12 @@ class IncrementalUpdateDumper:
         self._touched: Set[int] = set()
         self._lock = threading.Lock()
         self._seq = 0
+        self._pending: list = []  # [(pinned keys, event or None)]
+
+    # ------------------------------------------------------------- recording
 
     def record(self, signs: np.ndarray) -> None:
+        """Record touched SIGNS (u64 id space) directly."""
         flush_signs = None
         with self._lock:
             self._touched.update(int(s) for s in signs)
@@ -48,6 +59,47 @@ class IncrementalUpdateDumper:
                 flush_signs = self._take()
         if flush_signs is not None:
             self._dump(flush_signs)
+
+    def record_keys(self, keys, u_count=None) -> None:
+        """Record touched MIXED keys from a device (or CPU) int64 tensor.
+        GPU tensors are staged through a pinned async copy; ``u_count`` is the
+        device-side valid-prefix length of a padded dedup (the padding tail is
+        key 0 and is dropped at harvest)."""
+        import torch
+
+        if keys.is_cuda:
+            n = keys.numel()
+            pin = torch.empty(n + 1, dtype=torch.int64, pin_memory=True)
+            pin[:n].copy_(keys, non_blocking=True)
+            if u_count is not None:
+                pin[n:].copy_(u_count, non_blocking=True)
+            else:
+                pin[n] = n
+            ev = torch.cuda.Event()
+            ev.record()
+            with self._lock:
+                self._pending.append((pin, ev))
+            self._harvest(block=False)
+        else:
+            k_np = keys.numpy().view(np.uint64)
+            self.record(hashing.splitmix64_inv(k_np[k_np != 0]))
+
+    def _harvest(self, block: bool) -> None:
+        """Convert completed pending copies into signs."""
+        with self._lock:
+            ready, still = [], []
+            for pin, ev in self._pending:
+                if block:
+                    ev.synchronize()
+                if ev.query():
+                    ready.append(pin)
+                else:
+                    still.append((pin, ev))
+            self._pending = still
+        for pin in ready:
+            n_valid = int(pin[-1].item())
+            k_np = pin[:n_valid].numpy().view(np.uint64)
+            self.record(hashing.splitmix64_inv(k_np[k_np != 0]))
 
     def _take(self) -> Optional[np.ndarray]:
         if not self._touched:
@@ -57,6 +109,7 @@ class IncrementalUpdateDumper:
         return signs
 
     def flush(self) -> Optional[str]:
+        self._harvest(block=True)
         with self._lock:
             signs = self._take()
         if signs is None:
@@ -68,29 +121,22 @@ class IncrementalUpdateDumper:
         pkt_dir = os.path.join(self.dst_dir, f"inc_{ts}")
         os.makedirs(pkt_dir, exist_ok=True)
         rank = self.engine.dist.rank
-        # export current rows for the touched signs per dim store
-        keys = hashing.splitmix64(signs)
         for i, dim in enumerate(sorted(self.engine.stores.keys())):
             store = self.engine.stores[dim]
-            import torch
-
-            rows = store.lookup(
-                torch.from_numpy(keys.view(np.int64)).to(self.engine.device),
-                train=False,
-            )
-            present = ~(rows == 0).all(dim=1)
-            present_np = present.cpu().numpy()
-            if present_np.any():
-                # re-export full rows (emb + opt state) for resident signs
-                all_signs, all_inner = store.export_rows()
-                keep = np.isin(all_signs, signs[present_np])
+            # probe-based presence + row export for exactly the touched signs
+            # (store.export_keys; never a full-table export, no insert, and a
+            # legitimately all-zero row is still exported)
+            found_signs, inner = store.export_keys(signs)
+            if len(found_signs):
                 write_emb_file(
                     os.path.join(pkt_dir, f"{rank}_{i}.inc"),
-                    all_signs[keep], all_inner[keep], dim,
+                    found_signs, inner, dim,
                 )
         with open(os.path.join(pkt_dir, DONE_MARKER), "w", encoding="utf-8") as f:
             f.write(str(ts))
         self._seq += 1
+        if self.engine.metrics_enabled:
+            self.engine.metrics.inc_packets_dumped.inc(1)
         return pkt_dir
 
 

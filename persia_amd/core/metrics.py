@@ -134,11 +134,18 @@ class _Wrapped:
 def get_metrics_manager(enable: Optional[bool] = None) -> MetricsManager:
     global _manager
     with _lock:
-        if _manager is None:
+        if _manager is None or (enable and not _manager.enable):
             if enable is None:
                 enable = os.environ.get("PERSIA_ENABLE_METRICS", "0") in ("1", "true")
             _manager = MetricsManager(enable=bool(enable))
         return _manager
+
+
+def reset_metrics_manager() -> None:
+    """Drop the process-wide manager (tests)."""
+    global _manager
+    with _lock:
+        _manager = None
 
 
 class EngineMetrics:
@@ -157,3 +164,22 @@ class EngineMetrics:
         self.nan_grad_skipped = m.counter("nan_grad_skipped")
         self.index_miss_count = m.counter("index_miss_count")
         self.gradient_id_miss_count = m.counter("gradient_id_miss_count")
+        self.lookup_hashmap_time_cost_sec = m.gauge("lookup_hashmap_time_cost_sec")
+        self.evicted_count = m.counter("evicted_count")
+        self.distinct_id_estimate = m.gauge(
+            "distinct_id_estimate", "HLL distinct-sign estimate per feature"
+        )
+        self.inc_packets_dumped = m.counter("inc_packets_dumped")
+
+    def sample_values(self) -> Dict[str, float]:
+        """Current values of the unlabeled series (tests/debug)."""
+        out: Dict[str, float] = {}
+        for name, attr in vars(self).items():
+            m = getattr(attr, "_m", None)
+            if m is None:
+                continue
+            try:
+                out[name] = m.labels(feat="")._value.get()
+            except Exception:
+                pass
+        return out

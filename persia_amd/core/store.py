@@ -169,6 +169,9 @@ class EmbeddingStoreBase:
         self.n_buckets = _floor_pow2(max(1, capacity // BUCKET_SIZE))
         self.n_slots = self.n_buckets * BUCKET_SIZE
         self.tick = 1  # current batch counter (0 reserved)
+        # tick allocation may be hit from the pipeline thread (lookups) and
+        # the main thread (updates, incremental exports) concurrently
+        self._tick_lock = __import__("threading").Lock()
         self.spill = HostTier(spill_capacity, self.row_width) if spill_capacity > 0 else None
         # Adam per-group beta powers (persia-common optim.rs:147-216); kept
         # per-store and stepped once per update call (all slot-groups in a
@@ -212,9 +215,20 @@ class EmbeddingStoreBase:
         raise NotImplementedError
 
     def next_tick(self) -> int:
-        t = self.tick
-        self.tick += 1
+        with self._tick_lock:
+            t = self.tick
+            self.tick += 1
         return t
+
+    def export_keys(self, signs: np.ndarray) -> Tuple[np.ndarray, np.ndarray]:
+        """Export rows for specific signs: (found_signs, inner[found]).
+        Presence comes from the table probe (NOT a rows==0 heuristic — an
+        all-zero row is a legitimate value), without insert-on-miss and
+        without consuming a tick.  Used by the incremental-update dumper;
+        may run concurrently with training — a row mutated mid-read yields a
+        mixed old/new row, which the bounded-staleness freshness contract
+        tolerates (the reference holds per-sign locks instead)."""
+        raise NotImplementedError
 
     def _spill_export(self):
         """Host-tier rows for checkpointing (spilled rows are table state)."""
@@ -393,6 +407,14 @@ class CpuEmbeddingStore(EmbeddingStoreBase):
         self.arena[idx] = rows
         return skipped
 
+    def export_keys(self, signs: np.ndarray) -> Tuple[np.ndarray, np.ndarray]:
+        ks = hashing.splitmix64(signs.astype(np.uint64))
+        ks[ks == EMPTY_KEY] = _ZERO_REMAP
+        slots = np.array([self._probe(k) for k in ks], dtype=np.int64)
+        found = slots >= 0
+        inner = self.arena[torch.from_numpy(slots[found])].numpy().copy()
+        return signs[found], inner
+
     def export_rows(self) -> Tuple[np.ndarray, np.ndarray]:
         occ = np.nonzero(self.keys != EMPTY_KEY)[0]
         signs = hashing.splitmix64_inv(self.keys[occ])
@@ -449,6 +471,7 @@ class HipEmbeddingStore(EmbeddingStoreBase):
         self._opt_code = {"sgd": 0, "adagrad": 1, "adam": 2}[optimizer.kind]
         self._probe_stream: Optional[torch.cuda.Stream] = None
         self._restore_tick: Optional[int] = None
+        self.evicted_total = 0  # host-side eviction count (spill drains)
         self._no_evict = (
             torch.empty(0, dtype=torch.int64, device=device),
             torch.empty(0, dtype=torch.int32, device=device),
@@ -489,6 +512,7 @@ class HipEmbeddingStore(EmbeddingStoreBase):
         cnt = int(cnt_pin.item())
         if cnt == 0:
             return
+        self.evicted_total += cnt
         keys_np = keys_t[:cnt].cpu().numpy().view(np.uint64)
         rows_np = rows_t[:cnt].cpu().numpy()
         self.spill.insert(keys_np, rows_np)
@@ -672,6 +696,30 @@ class HipEmbeddingStore(EmbeddingStoreBase):
 
     def nan_row_count(self) -> int:
         return int(self._skipped[1].item())
+
+    def export_keys(self, signs: np.ndarray) -> Tuple[np.ndarray, np.ndarray]:
+        ks = hashing.splitmix64(signs.astype(np.uint64))
+        ks[ks == EMPTY_KEY] = _ZERO_REMAP
+        keys_t = torch.from_numpy(ks.view(np.int64)).to(self.device)
+        slots = self._C.store_probe(
+            self.keys, self.ticks, keys_t, self.tick,
+            torch.empty(0, dtype=torch.int64, device=self.device),
+        )
+        found = slots >= 0
+        inner = self.arena[slots[found]].cpu().numpy()
+        found_np = found.cpu().numpy()
+        out_signs, out_inner = signs[found_np], inner
+        if self.spill is not None and (~found_np).any():
+            # a touched sign may have been evicted to the host tier between
+            # update and flush; fetch + re-insert (peek) keeps it resident
+            self.flush_spill()
+            miss, miss_ks = signs[~found_np], ks[~found_np]
+            rows, sp_found = self.spill.fetch(miss_ks)
+            if sp_found.any():
+                self.spill.insert(miss_ks[sp_found], rows[sp_found])
+                out_signs = np.concatenate([out_signs, miss[sp_found]])
+                out_inner = np.concatenate([out_inner, rows[sp_found]])
+        return out_signs, out_inner
 
     def export_rows(self) -> Tuple[np.ndarray, np.ndarray]:
         self.flush_spill()
