@@ -111,6 +111,57 @@ def main(epochs=3, batch_size=128):
     return train_auc, test_auc
 
 
+def train_and_dump(dst_dir, epochs=1, batch_size=128, online_batches=60):
+    """Serving-path driver (the reference's train half of the TorchServe e2e,
+    .buildkite/e2e docker-compose.train + serve_client.py): train, dump a
+    full checkpoint, then KEEP training with incremental updates enabled so
+    the infer side can prove it applies the freshness stream on top of the
+    checkpoint.  Returns the incremental dir."""
+    import itertools
+
+    if REPRODUCIBLE:
+        setup_seed(3)
+    dense, ids, labels = make_dataset()
+    n_test = len(labels) // 5
+    train_data = (dense[n_test:], ids[n_test:], labels[n_test:])
+
+    model = DNN()
+    dense_opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    inc_dir = os.path.join(dst_dir, "inc")
+    with TrainCtx(
+        model=model,
+        embedding_optimizer=Adagrad(lr=0.05),
+        dense_optimizer=dense_opt,
+        embedding_config=EmbeddingConfig(),
+        embedding_schema=os.path.join(CONFIG_DIR, "embedding_config.yml"),
+        global_config=os.path.join(CONFIG_DIR, "global_config.yml"),
+        mixed_precision=False,
+    ) as ctx:
+        loss_fn = torch.nn.BCELoss()
+
+        def run_epochs(n, limit=None):
+            loader = DataLoader(
+                TrainDataset(*train_data, batch_size=batch_size, epochs=n),
+                reproducible=REPRODUCIBLE,
+                embedding_staleness=1 if REPRODUCIBLE else EMBEDDING_STALENESS,
+            )
+            it = iter(loader)
+            if limit is not None:
+                it = itertools.islice(it, limit)
+            for batch in it:
+                pred, lbls = ctx.forward(batch)
+                loss = loss_fn(pred.squeeze(1), lbls[0].squeeze(1).float())
+                ctx.backward(loss)
+
+        run_epochs(epochs)
+        ctx.dump_checkpoint(dst_dir)
+        # online phase: updates stream out as incremental packets
+        inc = ctx.engine.enable_incremental_update(inc_dir, buffer_size=2000)
+        run_epochs(1, limit=online_batches)
+        inc.flush()
+    return inc_dir
+
+
 if __name__ == "__main__":
     train_auc, test_auc = main()
     print(f"train_auc={train_auc!r} test_auc={test_auc!r}")

@@ -148,6 +148,25 @@ class NativeHostTier:
         return keys.numpy().view(np.uint64), rows.numpy()
 
 
+
+def _adapt_row_width(inner: np.ndarray, row_width: int,
+                     state_init: float = 0.0) -> np.ndarray:
+    """Adapt checkpoint rows to this store's row width: an infer-role store
+    (no optimizer state) truncates the dumped opt-state columns; a train
+    store loading an optimizer-less dump re-initializes the state columns
+    (reference infer PS keeps whatever inner was dumped and slices the emb
+    prefix on lookup — mod.rs:231-251; adapting at import keeps our fixed
+    arena layout)."""
+    cur = inner.shape[1]
+    if cur == row_width:
+        return inner
+    if cur > row_width:
+        return np.ascontiguousarray(inner[:, :row_width])
+    out = np.full((inner.shape[0], row_width), state_init, dtype=np.float32)
+    out[:, :cur] = inner
+    return out
+
+
 class EmbeddingStoreBase:
     """One rank's shard for one dim-group of slots."""
 
@@ -426,8 +445,8 @@ class CpuEmbeddingStore(EmbeddingStoreBase):
         return signs, inner
 
     def import_rows(self, signs: np.ndarray, inner: np.ndarray) -> None:
-        assert inner.shape[1] == self.row_width, (
-            f"checkpoint row width {inner.shape[1]} != store row width {self.row_width}"
+        inner = _adapt_row_width(
+            inner, self.row_width, float(self.optimizer.state_init(self.dim))
         )
         ks = hashing.splitmix64(signs.astype(np.uint64))
         ks[ks == EMPTY_KEY] = _ZERO_REMAP
@@ -734,7 +753,9 @@ class HipEmbeddingStore(EmbeddingStoreBase):
         return signs, inner
 
     def import_rows(self, signs: np.ndarray, inner: np.ndarray) -> None:
-        assert inner.shape[1] == self.row_width
+        inner = _adapt_row_width(
+            inner, self.row_width, float(self.optimizer.state_init(self.dim))
+        )
         ks = hashing.splitmix64(signs.astype(np.uint64))
         ks[ks == EMPTY_KEY] = _ZERO_REMAP
         keys_t = torch.from_numpy(ks.view(np.int64)).to(self.device)

@@ -29,9 +29,27 @@ def _run_once():
     return train_auc, test_auc
 
 
+# Pinned golden constants (the analog of the reference's
+# CPU_TEST_AUC = 0.8928645493226243, examples/src/adult-income/train.py:23):
+# a DETERMINISTICALLY-WRONG numerics regression now fails loudly instead of
+# passing the run-to-run reproducibility check.  Captured 2026-09-14 on the
+# synthetic stand-in dataset, world=1 CPU, REPRODUCIBLE=1, staleness=1.
+CPU_TRAIN_AUC = "0.8791403438530434"
+CPU_TEST_AUC = "0.8963082277978777"
+
+
 @pytest.mark.timeout(1200)
 def test_deterministic_auc_reproduces():
     a1 = _run_once()
     a2 = _run_once()
     assert a1 == a2, f"deterministic mode must reproduce bitwise: {a1} != {a2}"
-    assert float(a1[0].split("(")[-1].rstrip(")")) > 0.80
+    train_repr = a1[0].split("(")[-1].rstrip(")")
+    test_repr = a1[1].split("(")[-1].rstrip(")")
+    assert train_repr == CPU_TRAIN_AUC, (
+        f"train AUC drifted from the golden constant: {train_repr} != "
+        f"{CPU_TRAIN_AUC} — a cross-commit numerics regression"
+    )
+    assert test_repr == CPU_TEST_AUC, (
+        f"test AUC drifted from the golden constant: {test_repr} != "
+        f"{CPU_TEST_AUC}"
+    )
