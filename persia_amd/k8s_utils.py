@@ -3,8 +3,10 @@
 The reference ships a Rust operator with a ``PersiaJob`` CRD that deploys the
 4-role process topology (k8s/src/crd.rs:42-64).  In the MI355X architecture a
 job is one torchrun pod per node (8 ranks over RCCL) plus optional data-loader
-pods; this CLI turns a PersiaJob-style yaml spec into plain k8s manifests —
-no operator needed.
+pods.  ``gen`` renders the manifests for manual apply; ``gencrd`` and
+``operator`` (persia_amd/k8s_operator.py) provide the reconciling-controller
+workflow: create PersiaJob CRs and the operator creates/patches these same
+workloads and reflects status.
 """
 import sys
 
@@ -113,6 +115,33 @@ def gen(spec_file):
     with open(spec_file, "r", encoding="utf-8") as f:
         spec = yaml.safe_load(f)
     print(yaml.safe_dump_all(generate_manifests(spec)))
+
+
+@cli.command("gencrd")
+def gencrd():
+    """Print the PersiaJob CRD yaml (reference k8s/src/bin/gencrd.rs)."""
+    from persia_amd.k8s_operator import main as op_main
+
+    op_main(["gencrd"])
+
+
+@cli.command("operator")
+@click.option("--api-server", default=None, help="k8s API base URL (default: in-cluster)")
+@click.option("--namespace", default=None)
+@click.option("--token", default=None)
+@click.option("--period-sec", default=2.0, type=float)
+def operator(api_server, namespace, token, period_sec):
+    """Run the PersiaJob reconciling operator (k8s/src/bin/operator.rs)."""
+    from persia_amd.k8s_operator import main as op_main
+
+    argv = ["operator", f"--period-sec={period_sec}"]
+    if api_server:
+        argv.append(f"--api-server={api_server}")
+    if namespace:
+        argv.append(f"--namespace={namespace}")
+    if token:
+        argv.append(f"--token={token}")
+    op_main(argv)
 
 
 if __name__ == "__main__":
