@@ -47,6 +47,23 @@ class SGD(Optimizer):
 
 
 class Adam(Optimizer):
+    """Sparse Adam with bias correction via pre-computed beta powers.
+
+    β-power bookkeeping (documented divergence from the reference,
+    persia-common/src/optim.rs:147-216): the reference keeps one
+    (β1ᵗ, β2ᵗ) accumulator PER FEATURE GROUP (keyed by the masked sign
+    prefix) and advances it when that group receives an update batch.  Here
+    the accumulator lives per dim-group STORE and advances once per update
+    call.  The two are identical whenever every feature group appears in
+    every batch — the invariant of this architecture's dim-group batching
+    and of every shipped example/bench config.  They diverge only for jobs
+    that send partial batches (some slots absent), where a group skipping a
+    batch here still sees its bias-correction step advance: a slightly
+    smaller early-step correction, not a correctness issue (both schemes
+    converge to the same asymptotic update).  CPU and HIP backends implement
+    the same store-level scheme bit-identically (oracle-tested).
+    """
+
     kind = "adam"
 
     def __init__(
