@@ -341,6 +341,37 @@ __global__ void bias_grad_kernel(const short* __restrict__ dC,
   }
 }
 
+// Vector column-sum for power-of-two N (the DLRM layer widths): the matrix
+// is consumed as flat bf16x8 vectors with FULLY COALESCED 16-byte lane
+// loads; because the grid-stride (grid*256*8) is a multiple of N, every
+// thread's vector always covers the SAME 8 columns, so the whole pass
+// accumulates in 8 registers.  A block then reduces its 2048 lane-octets
+// into N partials through LDS and issues one atomicAdd per column.  The
+// scalar kernel above (2 B per lane per load) measured 24.5 us for an
+// 8 MB matrix — ~20x off the bandwidth bound this version targets.
+__global__ __launch_bounds__(256) void bias_grad_vec_kernel(
+    const short* __restrict__ dC, float* __restrict__ db, int64_t total8,
+    int N) {
+  extern __shared__ float lds_col[];  // [N]
+  const int n8 = N / 8;
+  for (int i = threadIdx.x; i < N; i += blockDim.x) lds_col[i] = 0.0f;
+  __syncthreads();
+  int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const int c8 = (int)((idx % n8) * 8);  // invariant: stride % n8 == 0
+  float acc[8] = {};
+  for (; idx < total8; idx += stride) {
+    const bf16x8 v = *(const bf16x8*)(dC + idx * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] += bf2f(v[j]);
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) atomicAdd(&lds_col[c8 + j], acc[j]);
+  __syncthreads();
+  for (int i = threadIdx.x; i < N; i += blockDim.x)
+    atomicAdd(&db[i], lds_col[i]);
+}
+
 }  // namespace
 
 // ============================================================ torch bindings
@@ -419,6 +450,23 @@ torch::Tensor bias_grad(torch::Tensor dC) {
   const int M = (int)dC.size(0), N = (int)dC.size(1);
   auto db = torch::zeros(
       {N}, torch::TensorOptions().dtype(torch::kFloat32).device(dC.device()));
+  // power-of-two N (every DLRM/DCN layer width): coalesced vector kernel.
+  // grid*256 must be a multiple of N/8 so each thread's column octet is
+  // loop-invariant; N/8 is a power of two <= 256's multiples, so any grid
+  // multiple of max(1, N/2048) works — use a 512-block grid (grid%*)
+  const bool pow2 = (N & (N - 1)) == 0 && N >= 8 && N <= 8192;
+  if (pow2) {
+    const int64_t total8 = (int64_t)M * N / 8;
+    int blocks = (int)std::min<int64_t>(512, (total8 + 255) / 256);
+    const int n8 = N / 8;
+    // round blocks up so (blocks*256) % n8 == 0 (n8 is a power of two)
+    if (n8 > 256) blocks = ((blocks + n8 / 256 - 1) / (n8 / 256)) * (n8 / 256);
+    hipLaunchKernelGGL(bias_grad_vec_kernel, dim3(blocks), dim3(256),
+                       N * sizeof(float), dcur_stream(),
+                       (const short*)dC.data_ptr(), db.data_ptr<float>(),
+                       total8, N);
+    return db;
+  }
   const int rows_per_block = std::max(8, (M + 511) / 512);
   const int blocks = (M + rows_per_block - 1) / rows_per_block;
   hipLaunchKernelGGL(bias_grad_kernel, dim3(blocks), dim3(256), 0,

@@ -57,9 +57,12 @@ def test_bias_grad_and_relu_bwd():
 
     C = native()
     torch.manual_seed(2)
+    for M, N in ((1000, 256), (8192, 512), (777, 360), (4096, 4096)):
+        dC = torch.randn(M, N, device=_dev()).to(torch.bfloat16)
+        db = C.bias_grad(dC.contiguous())
+        ref = dC.float().sum(0)
+        assert torch.allclose(db, ref, atol=0.5 + M / 4000, rtol=0.02), (M, N)
     dC = torch.randn(1000, 256, device=_dev()).to(torch.bfloat16)
-    db = C.bias_grad(dC.contiguous())
-    assert torch.allclose(db, dC.float().sum(0), atol=0.5, rtol=0.02)
     out = torch.randn(1000, 256, device=_dev()).to(torch.bfloat16)
     geff = C.relu_bwd(dC.contiguous(), out.contiguous())
     ref = dC.float() * (out.float() > 0)
