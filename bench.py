@@ -288,7 +288,7 @@ def main():
                     with torch.no_grad():
                         torch._foreach_copy_(g_params, g_masters)
 
-            elif bf16_weights:
+            elif bf16_weights and os.environ.get("PA_FLAT_PARAMS", "1") == "1":
                 # FLAT param/grad/master scheme: every weight is a view of
                 # one bf16 buffer, gradients accumulate into one flat bf16
                 # buffer, the f32 master SGD is 4 whole-buffer kernels.  The
@@ -325,6 +325,28 @@ def main():
                     flat_g.zero_()
                     return loss
 
+            elif bf16_weights:
+                # per-param foreach variant (PA_FLAT_PARAMS=0 A/B switch)
+                g_params = list(model.parameters())
+                g_masters = [p.detach().clone().float() for p in g_params]
+                g_grads32 = [torch.zeros_like(m) for m in g_masters]
+                lr = opt.param_groups[0]["lr"]
+                for p in g_params:
+                    p.grad = torch.zeros_like(p)
+
+                def iteration():
+                    static["base"].grad.zero_()
+                    logits = model(static["dense"], static["base"])
+                    loss = loss_fn(logits.float(), static["label"])
+                    loss.backward()
+                    grads = [p.grad for p in g_params]
+                    torch._foreach_copy_(g_grads32, grads)
+                    torch._foreach_add_(g_masters, g_grads32, alpha=-lr)
+                    with torch.no_grad():
+                        torch._foreach_copy_(g_params, g_masters)
+                    torch._foreach_zero_(grads)
+                    return loss
+
             else:
 
                 def iteration():
@@ -354,7 +376,7 @@ def main():
             graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(graph):
                 static["loss"] = iteration()
-            if bf16_weights and world == 1:
+            if bf16_weights and world == 1 and "flat_w" in locals():
                 # verify the captured backward writes the flat grad views
                 # (a re-allocated .grad would silently train nothing): one
                 # replay on random inputs must move the weights
