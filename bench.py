@@ -296,17 +296,24 @@ def main():
                 # launches per replay (4.7 us each on the profile — pure
                 # launch overhead for bias-sized tensors).
                 g_params = list(model.parameters())
-                n_tot = sum(p.numel() for p in g_params)
+                # 256 B-aligned slots: mis-aligned weight views knock
+                # hipBLASLt (and the fused kernels' 16 B vector loads) off
+                # their fast paths — measured 2x on the DCN preset
+                ALIGN = 128  # bf16 elements
+                offs, off = [], 0
+                for p in g_params:
+                    off = (off + ALIGN - 1) // ALIGN * ALIGN
+                    offs.append(off)
+                    off += p.numel()
+                n_tot = off
                 flat_w = torch.zeros(n_tot, dtype=torch.bfloat16, device=device)
                 flat_g = torch.zeros(n_tot, dtype=torch.bfloat16, device=device)
-                off = 0
-                for p in g_params:
+                for p, o in zip(g_params, offs):
                     n = p.numel()
-                    flat_w[off : off + n].copy_(p.detach().view(-1))
+                    flat_w[o : o + n].copy_(p.detach().view(-1))
                     with torch.no_grad():
-                        p.data = flat_w[off : off + n].view_as(p)
-                    p.grad = flat_g[off : off + n].view_as(p)
-                    off += n
+                        p.data = flat_w[o : o + n].view_as(p)
+                    p.grad = flat_g[o : o + n].view_as(p)
                 flat_master = flat_w.float()
                 flat_g32 = torch.zeros_like(flat_master)
                 lr = opt.param_groups[0]["lr"]
