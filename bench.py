@@ -26,11 +26,14 @@ PRESETS = {
                       batch_size=128),
     "criteo": dict(model="dlrm", num_sparse=26, num_dense=13, dim=128, rows=1e8),
     "terabyte": dict(model="dlrm", num_sparse=26, num_dense=13, dim=128, rows=1e10),
+    # flat_params=0 on the sparse-heavy presets: the flat scheme's gapless
+    # dense replay starves the concurrent sparse stream (scatter_update/sort
+    # stretched ~10x, measured) when that stream is the critical path
     "dcn-spill": dict(model="dcn", num_sparse=26, num_dense=13, dim=64, rows=1e11,
-                      spill_capacity=2e8),
+                      spill_capacity=2e8, flat_params=0),
     # dim-8 towers don't suit the fused MFMA layers (skinny K, odd widths)
     "100t": dict(model="dlrm", num_sparse=64, num_dense=13, dim=8, rows=1e12,
-                 fused_dense=0),
+                 fused_dense=0, flat_params=0),
 }
 
 
@@ -56,6 +59,9 @@ def parse_args():
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--graph", type=int, default=1,
                    help="capture dense fwd+bwd in a hipGraph (1 GPU only)")
+    p.add_argument("--flat-params", type=int, default=None,
+                   help="flat param/grad/master buffers in the graphed step "
+                        "(preset-dependent default)")
     p.add_argument("--fused-dense", type=int, default=None,
                    help="use the hand-written MFMA GEMM+bias+ReLU layers "
                         "(beats the graphed hipBLASLt path since the "
@@ -66,6 +72,7 @@ def parse_args():
     preset.setdefault("batch_size", 8192)  # x8 ranks = the MLPerf DLRM 64k global batch
     preset.setdefault("spill_capacity", 0)
     preset.setdefault("fused_dense", 1)
+    preset.setdefault("flat_params", 1)
     for k, v in preset.items():
         if getattr(args, k, None) is None:
             setattr(args, k, v)
@@ -288,7 +295,7 @@ def main():
                     with torch.no_grad():
                         torch._foreach_copy_(g_params, g_masters)
 
-            elif bf16_weights and os.environ.get("PA_FLAT_PARAMS", "1") == "1":
+            elif bf16_weights and bool(args.flat_params):
                 # FLAT param/grad/master scheme: every weight is a view of
                 # one bf16 buffer, gradients accumulate into one flat bf16
                 # buffer, the f32 master SGD is 4 whole-buffer kernels.  The
