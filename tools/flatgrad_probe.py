@@ -2,6 +2,11 @@
 fwd+bwd of an MLP with (none|preset|flat-view) .grad handling."""
 import time
 
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+
 import torch
 
 dev = "cuda"

@@ -3,6 +3,11 @@ building a deep stream backlog; a synchronizing op's host time jumps to the
 backlog length.  (PA route phase measured 2.27 ms/batch on the GPU.)"""
 import time
 
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+
 import torch
 
 dev = torch.device("cuda", 0)
