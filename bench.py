@@ -218,9 +218,23 @@ def main():
 
     # pinned staging built up-front (setup, untimed): first-visit
     # hipHostMalloc in the pipeline thread is ms-scale on a freshly
-    # reclaimed host and would dominate short timed runs
+    # reclaimed host and would dominate short runs
     for hb in host_batches:
         engine.prepare_host_batch(hb)
+
+    if world > 1:
+        # Serialize communicator creation: the default group (dense), the
+        # sparse a2a group (pipeline thread) and the grad group (main
+        # thread) all lazily build their NCCL communicator at first use —
+        # two ranks initializing different comms concurrently from
+        # different threads can deadlock.  One tiny collective per group,
+        # one thread, fixed order.
+        warm = torch.ones(1, device=device if use_gpu else None)
+        dist.all_reduce(warm)
+        engine.dist.barrier()
+        engine.dist_grad.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
 
     pipeline = ForwardPipeline(engine, staleness=args.staleness, out_buffer=args.staleness + 2)
     pipeline.start()
