@@ -161,7 +161,10 @@ def main():
             fused=bool(args.fused_dense) and use_gpu,
         ).to(device)
     else:
-        model = DCNv2(num_sparse=n_slots, num_dense=args.num_dense, dim=dim).to(device)
+        model = DCNv2(
+            num_sparse=n_slots, num_dense=args.num_dense, dim=dim,
+            fused=bool(args.fused_dense) and use_gpu,
+        ).to(device)
     # bf16 model weights + f32 master weights (see the graph block below):
     # must happen BEFORE the DDP wrap so the gradient buckets are built bf16
     # (halves allreduce bytes over xGMI as a bonus)

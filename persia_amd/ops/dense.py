@@ -83,14 +83,15 @@ class FusedLinearFn(torch.autograd.Function):
 
 
 class FusedLinear(nn.Module):
-    def __init__(self, in_features: int, out_features: int, relu: bool = True):
+    def __init__(self, in_features: int, out_features: int, relu: bool = True,
+                 bias: bool = True):
         super().__init__()
         assert out_features % 32 == 0, "FusedLinear needs out_features % 32 == 0"
         self.in_features = in_features
         self.out_features = out_features
         self.relu = relu
         self.weight = nn.Parameter(torch.empty(out_features, in_features))
-        self.bias = nn.Parameter(torch.zeros(out_features))
+        self.bias = nn.Parameter(torch.zeros(out_features), requires_grad=bias)
         nn.init.kaiming_uniform_(self.weight, a=5 ** 0.5)
 
     def forward(self, x):
