@@ -192,6 +192,109 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// NT GEMM v2: 2-buffer global_load_lds pipeline (guide §5 "minimum
+// 2-phase" + §5.4 rule 21).  The register-staged loop above drains
+// vmcnt(0)+lgkmcnt(0) inside every __syncthreads while its stage loads are
+// still in flight — PMC on the bench shape showed SQ_WAIT 9:1 over busy and
+// ~14% MFMA utilization.  Here the next K-tile's loads go straight to LDS
+// via glds DMA issued BEFORE the MFMA block, the barrier is a raw s_barrier
+// with explicit counted waits, and the LDS image is lane-linear with the
+// conflict-spreading XOR applied to the SOURCE addresses and the fragment
+// reads (the same involution on both sides — rule 21).
+// Constraints: K % 64 == 0, N % 128 == 0 (launcher falls back otherwise).
+template <int ACT, bool STORE_F32>
+__global__ __launch_bounds__(256) void gemm_nt_glds_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ B,
+    const float* __restrict__ bias, void* __restrict__ C, int M, int N,
+    int K) {
+  constexpr int TBK = 64;  // 128 B per image row
+  __shared__ short lds_a[2][128 * TBK];
+  __shared__ short lds_b[2][128 * TBK];
+  const int n_tiles_n = N / 128;
+  const int m0 = (blockIdx.x / n_tiles_n) * 128;
+  const int n0 = (blockIdx.x % n_tiles_n) * 128;
+  const int tid = threadIdx.x;
+  const int lane = tid % 64;
+  const int wave = tid / 64;
+  const int wr = wave / 2, wc = wave % 2;
+  const int fi = lane & 15;
+  const int g16 = lane >> 4;  // fragment k-granule (16 B units)
+
+  // wave w stages rows [w*32, w*32+32) of each image as 4 glds chunks of
+  // 8 rows (64 lanes x 16 B = 1 KB, lane-linear dest).  LDS granule g of
+  // row r holds LOGICAL granule g ^ (r & 7): the source address carries the
+  // XOR so the read side below is conflict-spread over 8 slots.
+  const int s_rl = lane >> 3;       // row within chunk (== row & 7)
+  const int s_gr = (lane & 7) ^ s_rl;  // pre-swizzled source granule
+  auto stage = [&](int bufi, int k0) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int row = wave * 32 + i * 8 + s_rl;
+      const int a_row = min(m0 + row, M - 1);
+      __builtin_amdgcn_global_load_lds(
+          (const uint32_t*)(A + (int64_t)a_row * K + k0 + s_gr * 8),
+          (uint32_t*)&lds_a[bufi][(wave * 32 + i * 8) * TBK], 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const uint32_t*)(B + (int64_t)(n0 + row) * K + k0 + s_gr * 8),
+          (uint32_t*)&lds_b[bufi][(wave * 32 + i * 8) * TBK], 16, 0, 0);
+    }
+  };
+  auto frag = [&](const short* base, int row, int gr) {
+    return *(const bf16x8*)&base[row * TBK + ((gr ^ (row & 7)) << 3)];
+  };
+
+  f32x4 acc[4][4] = {};
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  int buf = 0;
+  for (int k0 = 0; k0 < K; k0 += TBK) {
+    if (k0 + TBK < K) stage(buf ^ 1, k0 + TBK);  // DMA hides under MFMA
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 bfr[4];
+#pragma unroll
+      for (int bn = 0; bn < 4; ++bn)
+        bfr[bn] = frag(lds_b[buf], wc * 64 + bn * 16 + fi, kk * 4 + g16);
+#pragma unroll
+      for (int am = 0; am < 4; ++am) {
+        const bf16x8 afr = frag(lds_a[buf], wr * 64 + am * 16 + fi,
+                                kk * 4 + g16);
+#pragma unroll
+        for (int bn = 0; bn < 4; ++bn)
+          acc[am][bn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr, bfr[bn], acc[am][bn], 0, 0, 0);
+      }
+    }
+    // next tile's glds must have LANDED (vmcnt) and this tile's reads be
+    // done (lgkm) before any wave re-stages over them
+    asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    buf ^= 1;
+  }
+
+#pragma unroll
+  for (int am = 0; am < 4; ++am) {
+#pragma unroll
+    for (int bn = 0; bn < 4; ++bn) {
+      const int col = n0 + wc * 64 + bn * 16 + fi;
+      const float bval = bias ? bias[col] : 0.0f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = m0 + wr * 64 + am * 16 + (lane >> 4) * 4 + r;
+        if (row >= M) continue;
+        float v = acc[am][bn][r] + bval;
+        if (ACT == 1) v = fmaxf(v, 0.0f);
+        if (STORE_F32)
+          ((float*)C)[(int64_t)row * N + col] = v;
+        else
+          ((short*)C)[(int64_t)row * N + col] = f2bf(v);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // wgrad: dW[N,K] += sum_m dC[m][n] * A[m][k]   (dC [M,N], A [M,K] bf16;
 // dW f32, torch Linear weight layout).  MFMA with the reduction on the
 // memory-row axis: both tiles are staged TRANSPOSED ([n][m] / [k][m],
@@ -415,6 +518,24 @@ torch::Tensor gemm_nt_bias_act(torch::Tensor A, torch::Tensor B,
           .dtype(out_f32 ? torch::kFloat32 : torch::kBFloat16)
           .device(A.device()));
   const int m_tiles = (M + BM - 1) / BM;
+  // v2 glds pipeline when shapes allow (PA_GEMM_V2=0 reverts)
+  static const bool v2_on = [] {
+    const char* e = getenv("PA_GEMM_V2");
+    return !e || atoi(e) != 0;
+  }();
+  if (v2_on && !trans_b && K % 64 == 0 && N % 128 == 0 && M >= 128) {
+    const int grid = m_tiles * (N / 128);
+    const float* bias_p = bias.numel() ? bias.data_ptr<float>() : nullptr;
+#define PA_GEMM2(ACTV, F32V)                                                  \
+  hipLaunchKernelGGL((gemm_nt_glds_kernel<ACTV, F32V>), dim3(grid),           \
+                     dim3(256), 0, dcur_stream(),                             \
+                     (const bf16*)A.data_ptr(), (const bf16*)B.data_ptr(),    \
+                     bias_p, C.data_ptr(), M, N, K)
+    if (act == 1) { if (out_f32) PA_GEMM2(1, true); else PA_GEMM2(1, false); }
+    else          { if (out_f32) PA_GEMM2(0, true); else PA_GEMM2(0, false); }
+#undef PA_GEMM2
+    return C;
+  }
   // tile config: BK=64 halves barriers (needs K%64); BN=64 doubles the grid
   // (skinny layers underfill 256 CUs at BN=128); (128,64) exceeds 64KB LDS.
   const bool k64 = K % 64 == 0 && N % 64 == 0;

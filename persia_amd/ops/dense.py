@@ -25,9 +25,11 @@ def _wgrad_via_blas() -> bool:
     return os.environ.get("PA_FUSED_WGRAD", "own") == "blas"
 
 
-def _pad32(x: torch.Tensor, dim: int) -> torch.Tensor:
+def _pad32(x: torch.Tensor, dim: int, mult: int = 64) -> torch.Tensor:
+    """Zero-pad dim to a multiple of `mult` (64: the glds-pipelined GEMM v2
+    needs K % 64; zero K-columns are inert)."""
     k = x.shape[dim]
-    pad = (-k) % 32
+    pad = (-k) % mult
     if pad == 0:
         return x
     padding = [0, 0] * (x.dim() - 1 - dim) + [0, pad]
