@@ -523,7 +523,11 @@ torch::Tensor gemm_nt_bias_act(torch::Tensor A, torch::Tensor B,
     const char* e = getenv("PA_GEMM_V2");
     return !e || atoi(e) != 0;
   }();
-  if (v2_on && !trans_b && K % 64 == 0 && N % 128 == 0 && M >= 128) {
+  // shape cutover (measured sweep, tools/gemm_v2_sweep.py): v2 wins when the
+  // 128x128 grid fills the chip (>=400 blocks) and K amortizes the pipeline
+  // prologue; below that the v1 TBN=64 configs keep more CUs busy
+  if (v2_on && !trans_b && K % 64 == 0 && N % 128 == 0 && M >= 128 &&
+      K >= 256 && (int64_t)m_tiles * (N / 128) >= 400) {
     const int grid = m_tiles * (N / 128);
     const float* bias_p = bias.numel() ? bias.data_ptr<float>() : nullptr;
 #define PA_GEMM2(ACTV, F32V)                                                  \
