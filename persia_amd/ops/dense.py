@@ -93,7 +93,11 @@ class FusedLinear(nn.Module):
         self.out_features = out_features
         self.relu = relu
         self.weight = nn.Parameter(torch.empty(out_features, in_features))
-        self.bias = nn.Parameter(torch.zeros(out_features), requires_grad=bias)
+        if bias:
+            self.bias = nn.Parameter(torch.zeros(out_features))
+        else:  # constant zero bias: buffer, so parameter lists align with
+            # a bias-free nn.Linear twin
+            self.register_buffer("bias", torch.zeros(out_features))
         nn.init.kaiming_uniform_(self.weight, a=5 ** 0.5)
 
     def forward(self, x):
