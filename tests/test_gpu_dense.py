@@ -240,11 +240,13 @@ def test_fused_dcn_matches_eager():
         pe, pf = list(me.parameters()), list(mf.parameters())
         assert len(pe) == len(pf)
         for a, b in zip(pe[:-2], pf[:-2]):  # all but the head
+            # fused shapes pad BOTH axes that carry the feature width
+            # (cross u outputs the padded width); zero rows/cols are inert
             b.zero_()
             if a.dim() == 2:
-                b[:, : a.shape[1]].copy_(a)
+                b[: a.shape[0], : a.shape[1]].copy_(a)
             else:
-                b.copy_(a)
+                b[: a.shape[0]].copy_(a)
         # head: column blocks move with the pad (cat([xl_pad, d]))
         hw_e, hb_e = pe[-2], pe[-1]
         hw_f, hb_f = pf[-2], pf[-1]
