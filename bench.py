@@ -284,26 +284,34 @@ def main():
             if bf16_weights and world > 1:
                 import torch.distributed as dist
 
+                # flat grads PER DTYPE (FusedLinear biases stay f32 — see
+                # ops/dense.py _apply); each flat buffer gets one allreduce
                 g_params = list(model.parameters())
-                n_tot = sum(p.numel() for p in g_params)
-                flat_grads = torch.zeros(
-                    n_tot, dtype=torch.bfloat16, device=device
-                )
-                off = 0
+                flat_by_dt = {}
+                for dt in {p.dtype for p in g_params}:
+                    n_dt = sum(p.numel() for p in g_params if p.dtype == dt)
+                    flat_by_dt[dt] = [torch.zeros(n_dt, dtype=dt, device=device), 0]
                 for p in g_params:
-                    p.grad = flat_grads[off : off + p.numel()].view_as(p)
-                    off += p.numel()
+                    buf, off = flat_by_dt[p.dtype]
+                    p.grad = buf[off : off + p.numel()].view_as(p)
+                    flat_by_dt[p.dtype][1] = off + p.numel()
+                flat_grads = flat_by_dt[torch.bfloat16][0]
+                all_flats = [v[0] for v in flat_by_dt.values()]
                 g_masters = [p.detach().clone().float() for p in g_params]
                 g_grads32 = [torch.zeros_like(m) for m in g_masters]
                 lr = opt.param_groups[0]["lr"]
 
                 def iteration():
                     static["base"].grad.zero_()
-                    torch._foreach_zero_([flat_grads])
+                    torch._foreach_zero_(all_flats)
                     logits = model(static["dense"], static["base"])
                     loss = loss_fn(logits.float(), static["label"])
                     loss.backward()
                     return loss
+
+                def allreduce_grads():
+                    for f in all_flats:
+                        dist.all_reduce(f)
 
                 def update_body():
                     grads = [p.grad for p in g_params]
@@ -321,7 +329,14 @@ def main():
                 # per-param foreach variant split into ~37 tiny elementwise
                 # launches per replay (4.7 us each on the profile — pure
                 # launch overhead for bias-sized tensors).
-                g_params = list(model.parameters())
+                g_params = [p for p in model.parameters()
+                            if p.dtype == torch.bfloat16]
+                # f32 params (FusedLinear biases, ops/dense.py _apply): tiny,
+                # already full precision — plain foreach SGD, no masters
+                f32_params = [p for p in model.parameters()
+                              if p.dtype == torch.float32]
+                for p in f32_params:
+                    p.grad = torch.zeros_like(p)
                 # 256 B-aligned slots: mis-aligned weight views knock
                 # hipBLASLt (and the fused kernels' 16 B vector loads) off
                 # their fast paths — measured 2x on the DCN preset
@@ -356,6 +371,11 @@ def main():
                     with torch.no_grad():
                         flat_w.copy_(flat_master)
                     flat_g.zero_()
+                    if f32_params:
+                        g32 = [p.grad for p in f32_params]
+                        with torch.no_grad():
+                            torch._foreach_add_(f32_params, g32, alpha=-lr)
+                        torch._foreach_zero_(g32)
                     return loss
 
             elif bf16_weights:
@@ -401,9 +421,7 @@ def main():
                 for _ in range(3):
                     iteration()
                     if flat_grads is not None:
-                        import torch.distributed as dist
-
-                        dist.all_reduce(flat_grads)
+                        allreduce_grads()
                         update_body()
             torch.cuda.current_stream().wait_stream(s)
             graph = torch.cuda.CUDAGraph()
@@ -470,9 +488,7 @@ def main():
                 t1 = time.perf_counter()
             graph.replay()
             if graph_upd is not None:
-                import torch.distributed as dist
-
-                dist.all_reduce(flat_grads)
+                allreduce_grads()
                 graph_upd.replay()
             if timing:
                 t2 = time.perf_counter()

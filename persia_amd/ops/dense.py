@@ -100,6 +100,21 @@ class FusedLinear(nn.Module):
             self.register_buffer("bias", torch.zeros(out_features))
         nn.init.kaiming_uniform_(self.weight, a=5 ** 0.5)
 
+    def _apply(self, fn, recurse=True):
+        # keep the bias fp32 through .bfloat16()/.half(): the GEMM epilogue
+        # adds it in f32 and bias_grad produces f32 — a low-precision bias
+        # only inserts per-step cast kernels on BOTH passes (the profile
+        # showed ~24 bias-sized launches/step from exactly this)
+        r = super()._apply(fn, recurse)
+        if self.bias.dtype != torch.float32:
+            if isinstance(self.bias, nn.Parameter):
+                self.bias.data = self.bias.data.float()
+                if self.bias.grad is not None:
+                    self.bias.grad = self.bias.grad.float()
+            else:
+                self.bias = self.bias.float()
+        return r
+
     def forward(self, x):
         return FusedLinearFn.apply(x, self.weight, self.bias, 1 if self.relu else 0)
 
