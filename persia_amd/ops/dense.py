@@ -64,10 +64,19 @@ class FusedLinearFn(torch.autograd.Function):
         db = C.bias_grad(g)
         if db.dtype != ctx.b_dtype:
             db = db.to(ctx.b_dtype)
-        # dX = g @ W: trans_b path consumes the [N, Kp] weight directly
-        dx = C.gemm_nt_bias_act(
-            g, w_bf, torch.empty(0, device=g.device), 0, 0, 1
-        )
+        # dX = g @ W: trans_b path consumes the [N, Kp] weight directly; at
+        # the big square shape the transposed-staging kernel loses to an
+        # explicit W^T + the glds NT path (51 vs 33+8 us measured,
+        # tools/dgrad_probe.py)
+        if w_bf.shape[0] >= 1024 and w_bf.shape[1] >= 1024:
+            dx = C.gemm_nt_bias_act(
+                g, w_bf.t().contiguous(), torch.empty(0, device=g.device),
+                0, 0, 0,
+            )
+        else:
+            dx = C.gemm_nt_bias_act(
+                g, w_bf, torch.empty(0, device=g.device), 0, 0, 1
+            )
         if dx.shape[1] != ctx.k:
             dx = dx[:, : ctx.k]
         if _wgrad_via_blas():
