@@ -192,6 +192,130 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// wgrad v2: glds-staged tiles + ds_read_b64_tr_b16 hardware transpose reads.
+//
+// dW[n,k] = sum_m dC[m,n] * A[m,k] reduces over the LEADING dim of both
+// operands, so both MFMA fragments need COLUMN reads of row-major tiles —
+// the v1 kernel pays for that with 8 scalar transposed ds_writes per loaded
+// vector (its measured bottleneck).  Here both tiles stage LINEARLY through
+// global_load_lds into 16-column panels and the fragments come from the
+// gfx950 transpose-read.  The tr pattern delivers m-elements to lanes in a
+// DIFFERENT order than mfma_16x16x32's nominal k-layout — which is
+// irrelevant: the reduction dim is order-invariant as long as BOTH operands
+// use the same (lane, slot) -> m mapping, and both are read with the same
+// instruction from identically-shaped panels.
+// Constraints: M % 64 == 0, N % 128 == 0, K % 128 == 0 (launcher falls
+// back to v1 otherwise — which keeps the small/odd layers).
+using bf16x4t = __attribute__((ext_vector_type(4))) __bf16;
+
+__global__ __launch_bounds__(256) void wgrad_tr_kernel(
+    const short* __restrict__ dC, const short* __restrict__ A,
+    float* __restrict__ dW, int M, int N, int K, int splitm) {
+  constexpr int MT = 64;  // m-rows per buffer
+  // panel layout per image: [8 panels][MT rows][16 cols] bf16 = 16 KB
+  __shared__ short lds_c[2][8 * MT * 16];
+  __shared__ short lds_a[2][8 * MT * 16];
+  const int n_tiles_k = K / 128;
+  const int tile_id = blockIdx.x / splitm;
+  const int m_part = blockIdx.x % splitm;
+  const int n0 = (tile_id / n_tiles_k) * 128;
+  const int k0 = (tile_id % n_tiles_k) * 128;
+  const int tid = threadIdx.x;
+  const int wave = tid / 64, lane = tid % 64;
+  const int wr = wave / 2, wc = wave % 2;
+  const int fi = lane & 15;
+
+  const int m_chunk = ((M + splitm - 1) / splitm + MT - 1) / MT * MT;
+  const int m_begin = m_part * m_chunk;
+  const int m_end = min(M, m_begin + m_chunk);
+
+  // glds chunks: wave w stages panels [w*2, w*2+2) of each image (4 x 1 KB
+  // per image).  Chunk c of a panel covers m-rows [c*32, c*32+32); lane l
+  // holds m = c*32 + l/2, cols (l%2)*8..+8 — 16 contiguous global bytes.
+  const int s_m = lane >> 1;
+  const int s_n8 = (lane & 1) * 8;
+  auto stage = [&](int bufi, int m0g) {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      const int p = wave * 2 + (c >> 1);
+      const int mrow = (c & 1) * 32 + s_m;
+      const int64_t m_g = (int64_t)(m0g + mrow);
+      __builtin_amdgcn_global_load_lds(
+          (const uint32_t*)(dC + m_g * N + n0 + p * 16 + s_n8),
+          (uint32_t*)&lds_c[bufi][(p * MT + (c & 1) * 32) * 16], 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const uint32_t*)(A + m_g * K + k0 + p * 16 + s_n8),
+          (uint32_t*)&lds_a[bufi][(p * MT + (c & 1) * 32) * 16], 16, 0, 0);
+    }
+  };
+  // transpose-read an 8-m fragment of panel p starting at m-row ms.
+  // Measured semantics (tools/trprobe): the 16 lanes of a group concatenate
+  // their 8-byte slots into a 4x16 row-major block and each lane receives
+  // its column — so with per-lane address (row ms + 8*(lane>>4))*16 +
+  // (lane&15)*4 shorts, lane l ends up holding m = ms + 8*(l>>4) + j, the
+  // NOMINAL mfma_16x16x32 k-layout (both operands read identically).
+  const int t_l4 = (lane & 15) * 4;
+  const int t_g8 = (lane >> 4) * 8;
+  auto tfrag = [&](const short* img, int p, int ms) {
+    const short* b0 = &img[(p * MT + ms + t_g8) * 16 + t_l4];
+    bf16x4t lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+        (__attribute__((address_space(3))) bf16x4t*)(void*)const_cast<short*>(b0));
+    bf16x4t hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+        (__attribute__((address_space(3))) bf16x4t*)(void*)const_cast<short*>(b0 + 4 * 16));
+    bf16x8 r;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      r[j] = ((const short*)&lo)[j];
+      r[4 + j] = ((const short*)&hi)[j];
+    }
+    return r;
+  };
+
+  f32x4 acc[4][4] = {};
+  stage(0, m_begin);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  int buf = 0;
+  for (int m0g = m_begin; m0g < m_end; m0g += MT) {
+    if (m0g + MT < m_end) stage(buf ^ 1, m0g + MT);
+#pragma unroll
+    for (int ms = 0; ms < MT; ms += 32) {
+      bf16x8 bfr[4];
+#pragma unroll
+      for (int bk = 0; bk < 4; ++bk)
+        bfr[bk] = tfrag(lds_a[buf], wc * 4 + bk, ms);
+#pragma unroll
+      for (int an = 0; an < 4; ++an) {
+        const bf16x8 afr = tfrag(lds_c[buf], wr * 4 + an, ms);
+#pragma unroll
+        for (int bk = 0; bk < 4; ++bk)
+          acc[an][bk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr, bfr[bk], acc[an][bk], 0, 0, 0);
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    buf ^= 1;
+  }
+
+#pragma unroll
+  for (int an = 0; an < 4; ++an) {
+#pragma unroll
+    for (int bk = 0; bk < 4; ++bk) {
+      const int n = n0 + wr * 64 + an * 16 + (lane >> 4) * 4;
+      const int k = k0 + wc * 64 + bk * 16 + fi;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        if (splitm > 1)
+          atomicAdd(&dW[(int64_t)(n + r) * K + k], acc[an][bk][r]);
+        else
+          dW[(int64_t)(n + r) * K + k] = acc[an][bk][r];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // NT GEMM v2: 2-buffer global_load_lds pipeline (guide §5 "minimum
 // 2-phase" + §5.4 rule 21).  The register-staged loop above drains
 // vmcnt(0)+lgkmcnt(0) inside every __syncthreads while its stage loads are
@@ -626,6 +750,26 @@ torch::Tensor wgrad(torch::Tensor dC, torch::Tensor A) {
   const int M = (int)dC.size(0), N = (int)dC.size(1), K = (int)A.size(1);
   auto dW = torch::empty(
       {N, K}, torch::TensorOptions().dtype(torch::kFloat32).device(A.device()));
+  static const bool tr_on = [] {
+    const char* e = getenv("PA_WGRAD_TR");
+    return !e || atoi(e) != 0;
+  }();
+  // cutover (tools/wgrad_tr_probe.py): the 128x128-tile tr kernel wins only
+  // when its own grid fills the chip (1024x1024: 48.7 vs 66.0 us); below
+  // that the v1 64x64 tiles + split-M keep more CUs busy
+  if (tr_on && M % 64 == 0 && N % 128 == 0 && K % 128 == 0 &&
+      (N / 128) * (K / 128) >= 64) {
+    const int tiles2 = (N / 128) * (K / 128);
+    int splitm = 1;
+    while (tiles2 * splitm < 512 && splitm < 64 && (M / (splitm * 2)) >= 64)
+      splitm *= 2;
+    if (splitm > 1) dW.zero_();
+    hipLaunchKernelGGL(wgrad_tr_kernel, dim3(tiles2 * splitm), dim3(256), 0,
+                       dcur_stream(), (const short*)dC.data_ptr(),
+                       (const short*)A.data_ptr(), dW.data_ptr<float>(), M, N,
+                       K, splitm);
+    return dW;
+  }
   const int tiles = ((N + 63) / 64) * ((K + 63) / 64);
   int splitm = 1;
   while (tiles * splitm < 512 && splitm < 64 && (M / (splitm * 2)) >= 32)
