@@ -1628,7 +1628,13 @@ class ForwardPipeline:
         stream = None
         if self.engine.device.type == "cuda":
             torch.cuda.set_device(self.engine.device)
-            stream = torch.cuda.Stream(device=self.engine.device)
+            # PA_SPARSE_PRIORITY=-1 raises the lookup stream's scheduling
+            # priority: on presets where the sparse stream is the critical
+            # path (spill/dim-8 tables) the dense replay otherwise starves it
+            import os as _os
+
+            prio = int(_os.environ.get("PA_SPARSE_PRIORITY", "0"))
+            stream = torch.cuda.Stream(device=self.engine.device, priority=prio)
         import heapq
 
         pending_eof = False
