@@ -31,8 +31,12 @@ PRESETS = {
     # stretched ~10x, measured) when that stream is the critical path
     # fused_dense=0: hipBLASLt wins the DCN's wide (K~1700) cross/deep
     # GEMMs (measured 3.39M vs 2.34M fused end-to-end)
+    # sparse_priority=-1: dcn is sparse-stream-critical; raising that
+    # stream's priority measured 2.67 -> 2.98M (neutral on criteo, slightly
+    # negative on terabyte)
     "dcn-spill": dict(model="dcn", num_sparse=26, num_dense=13, dim=64, rows=1e11,
-                      spill_capacity=2e8, flat_params=0, fused_dense=0),
+                      spill_capacity=2e8, flat_params=0, fused_dense=0,
+                      sparse_priority=-1),
     # dim-8 towers don't suit the fused MFMA layers (skinny K, odd widths)
     "100t": dict(model="dlrm", num_sparse=64, num_dense=13, dim=8, rows=1e12,
                  fused_dense=0, flat_params=0),
@@ -61,6 +65,8 @@ def parse_args():
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--graph", type=int, default=1,
                    help="capture dense fwd+bwd in a hipGraph (1 GPU only)")
+    p.add_argument("--sparse-priority", type=int, default=None,
+                   help="lookup-stream priority (-1 = high; preset default)")
     p.add_argument("--flat-params", type=int, default=None,
                    help="flat param/grad/master buffers in the graphed step "
                         "(preset-dependent default)")
@@ -75,6 +81,7 @@ def parse_args():
     preset.setdefault("spill_capacity", 0)
     preset.setdefault("fused_dense", 1)
     preset.setdefault("flat_params", 1)
+    preset.setdefault("sparse_priority", 0)
     for k, v in preset.items():
         if getattr(args, k, None) is None:
             setattr(args, k, v)
@@ -124,6 +131,9 @@ def main():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29500")
         dist.init_process_group("nccl" if use_gpu else "gloo", rank=rank, world_size=world)
+
+    if args.sparse_priority:
+        os.environ.setdefault("PA_SPARSE_PRIORITY", str(args.sparse_priority))
 
     from persia_amd.core.comm import DistContext
     from persia_amd.core.engine import EmbeddingEngine, ForwardPipeline
