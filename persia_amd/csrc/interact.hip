@@ -28,6 +28,7 @@
 namespace {
 
 using bf16x8i = __attribute__((ext_vector_type(8))) short;
+using bf16x4i = __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16;
 using f32x4i = __attribute__((ext_vector_type(4))) float;
 
 __device__ __forceinline__ float ibf2f(short x) {
@@ -220,15 +221,40 @@ __global__ __launch_bounds__(256) void interact_bwd_packed_kernel(
   const int waves = blockDim.x / 64;
   const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
   const int Fp = ceil16(F), Fk = ceil32(F), Dp = ceil16(D);
-  const int LDA = Fk + 8;
-  short* A = lds + wave * (Fp + Dp) * LDA;
-  short* vt = A + Fp * LDA;
-  for (int t = lane; t < (Fp + Dp) * LDA / 8; t += 64)
+  const int LDA = Fk + 8;   // G image rows i, cols f'
+  const int LDV = Dp + 8;   // V image rows f', cols d (LINEAR: no transpose)
+  short* A = lds + wave * (Fp * LDA + Fk * LDV);
+  short* v = A + Fp * LDA;
+  for (int t = lane; t < (Fp * LDA + Fk * LDV) / 8; t += 64)
     *(bf16x8i*)&A[t * 8] = bf16x8i{};
   const int Tm = Fp / 16, Tn = Dp / 16;
   const int fi = lane & 15;
   const int fk8 = (lane >> 4) * 8;
   const int d8 = D / 8;
+  // V^T fragments come from ds_read_b64_tr_b16 on the LINEAR V image (the
+  // round-1 transposed scalar staging was the kernel's measured LDS
+  // bottleneck).  Addressing per the measured semantics (tools/trprobe +
+  // dense.hip wgrad_tr_kernel): lane m of a 16-lane group addresses the
+  // slot (row_base + m/4)*LDV + col_base + (m%4)*4 shorts; the group's
+  // slots form a 4x16 row-major block and lane m receives its column —
+  // with row_base = k + 8*(lane>>4) the fragment lands in the nominal
+  // mfma_16x16x32 k-layout, consistent with the row-read A fragment.
+  const int t_ro = ((lane & 15) >> 2);
+  const int t_co = (lane & 3) * 4;
+  auto vtr = [&](int kb, int td) {
+    const short* b0 = &v[(kb + fk8 + t_ro) * LDV + td * 16 + t_co];
+    bf16x4i lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+        (__attribute__((address_space(3))) bf16x4i*)(void*)const_cast<short*>(b0));
+    bf16x4i hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+        (__attribute__((address_space(3))) bf16x4i*)(void*)const_cast<short*>(b0 + 4 * LDV));
+    bf16x8i r;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      r[j] = ((const short*)&lo)[j];
+      r[4 + j] = ((const short*)&hi)[j];
+    }
+    return r;
+  };
   for (int64_t b = (int64_t)blockIdx.x * waves + wave; b < B;
        b += (int64_t)gridDim.x * waves) {
     for (int t = lane; t < F * d8; t += 64) {
@@ -241,8 +267,7 @@ __global__ __launch_bounds__(256) void interact_bwd_packed_kernel(
 #pragma unroll
         for (int k = 0; k < 8; ++k) row[k] = if2bf(__half2float(src[k]));
       }
-#pragma unroll
-      for (int k = 0; k < 8; ++k) vt[(c8 + k) * LDA + fr] = row[k];
+      *(bf16x8i*)&v[fr * LDV + c8] = row;
     }
     const short* gp = g + b * P;
     for (int p = lane; p < P; p += 64) {
@@ -259,7 +284,7 @@ __global__ __launch_bounds__(256) void interact_bwd_packed_kernel(
         f32x4i acc = {};
         for (int k = 0; k < Fk; k += 32) {
           const bf16x8i a = *(const bf16x8i*)&A[(ti * 16 + fi) * LDA + k + fk8];
-          const bf16x8i c = *(const bf16x8i*)&vt[(td * 16 + fi) * LDA + k + fk8];
+          const bf16x8i c = vtr(k, td);
           acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, c, acc, 0, 0, 0);
         }
         const int d = td * 16 + fi;
@@ -484,7 +509,7 @@ std::vector<torch::Tensor> interact_bwd_packed(torch::Tensor g, torch::Tensor x,
   auto dx = torch::empty_like(x);
   auto dbase = torch::empty_like(base);
   const int Fp = ceil16(F), Fk = ceil32(F), Dp = ceil16(D);
-  const int mfma_shorts = (Fp + Dp) * (Fk + 8);
+  const int mfma_shorts = Fp * (Fk + 8) + Fk * (Dp + 8);
   const int waves = pick_waves(mfma_shorts);
   TORCH_CHECK(waves > 0, "interact_bwd_packed: tile exceeds LDS");
   const int grid = std::min((B + waves - 1) / waves, 8192);
@@ -505,7 +530,7 @@ bool interact_packed_feasible(int64_t F, int64_t D) {
   const int Fp = ceil16((int)F), Fk = ceil32((int)F), Dp = ceil16((int)D),
             Dk = ceil32((int)D);
   return pick_waves(Fp * (Dk + 8)) > 0 &&
-         pick_waves((Fp + Dp) * (Fk + 8)) > 0;
+         pick_waves(Fp * (Fk + 8) + Fk * (Dp + 8)) > 0;
 }
 
 bool interact_feasible(int64_t F, int64_t D) {
