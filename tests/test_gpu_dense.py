@@ -10,7 +10,8 @@ def _dev():
     return torch.device("cuda", 0)
 
 
-@pytest.mark.parametrize("shape", [(256, 64, 128), (512, 96, 256), (4096, 480, 1024)])
+@pytest.mark.parametrize("shape", [(256, 64, 128), (512, 96, 256), (4096, 480, 1024),
+                                   (8192, 512, 1024), (8192, 1024, 1024)])
 @pytest.mark.parametrize("act", [0, 1])
 def test_gemm_nt_bias_act_matches_torch(shape, act):
     from persia_amd.ops import native
@@ -42,7 +43,10 @@ def test_wgrad_matches_torch():
 
     C = native()
     torch.manual_seed(1)
-    for M, N, K in [(4096, 512, 480), (1024, 64, 32), (4096, 1024, 1024)]:
+    # (8192, 1024, 1024) exercises the tr-read wgrad v2 (glds + transpose
+    # reads); the smaller shapes stay on the split-M v1 kernel
+    for M, N, K in [(4096, 512, 480), (1024, 64, 32), (4096, 1024, 1024),
+                    (8192, 1024, 1024), (8192, 512, 512)]:
         dC = (torch.randn(M, N, device=_dev()) * 0.1).to(torch.bfloat16)
         A = (torch.randn(M, K, device=_dev()) * 0.1).to(torch.bfloat16)
         dW = C.wgrad(dC.contiguous(), A.contiguous())
