@@ -333,40 +333,60 @@ def main():
                         torch._foreach_copy_(g_params, g_masters)
 
             elif bf16_weights and bool(args.flat_params):
-                # FLAT param/grad/master scheme: every weight is a view of
-                # one bf16 buffer, gradients accumulate into one flat bf16
-                # buffer, the f32 master SGD is 4 whole-buffer kernels.  The
-                # per-param foreach variant split into ~37 tiny elementwise
-                # launches per replay (4.7 us each on the profile — pure
-                # launch overhead for bias-sized tensors).
-                g_params = [p for p in model.parameters()
-                            if p.dtype == torch.bfloat16]
-                # f32 params (FusedLinear biases, ops/dense.py _apply): tiny,
-                # already full precision — plain foreach SGD, no masters
-                f32_params = [p for p in model.parameters()
-                              if p.dtype == torch.float32]
-                for p in f32_params:
-                    p.grad = torch.zeros_like(p)
-                # 256 B-aligned slots: mis-aligned weight views knock
-                # hipBLASLt (and the fused kernels' 16 B vector loads) off
-                # their fast paths — measured 2x on the DCN preset
-                ALIGN = 128  # bf16 elements
-                offs, off = [], 0
-                for p in g_params:
+                # SIDE-BAND flat scheme: every FusedLinear accumulates its
+                # weight/bias gradient straight into ONE f32 flat buffer via
+                # atomics inside wgrad/bias_grad (ops/dense.py _sideband) —
+                # no per-layer dw materialization, f32->bf16 cast, pad-column
+                # slice or AccumulateGrad add.  bf16 weights are views of an
+                # aligned flat bf16 buffer with an f32 master; the f32 grad
+                # buffer mirrors the weight layout so the SGD is one add +
+                # one cast-copy.  Non-fused params (the 1-logit head) keep
+                # preset bf16 grads, cast into their f32 slots per step.
+                from persia_amd.ops.dense import FusedLinear
+
+                fused_w, fused_b = [], []
+                for m in model.modules():
+                    if isinstance(m, FusedLinear):
+                        fused_w.append(m.weight)
+                        if isinstance(m.bias, torch.nn.Parameter):
+                            fused_b.append(m.bias)
+                fused_ids = {id(p) for p in fused_w + fused_b}
+                other = [p for p in model.parameters() if id(p) not in fused_ids]
+                bf_params = [p for p in model.parameters()
+                             if p.dtype == torch.bfloat16]
+                ALIGN = 128  # 256 B slots: misaligned views knock hipBLASLt
+                # and the 16 B vector loads off their fast paths (measured)
+                offs, off = {}, 0
+                for p in bf_params:
                     off = (off + ALIGN - 1) // ALIGN * ALIGN
-                    offs.append(off)
+                    offs[id(p)] = off
                     off += p.numel()
-                n_tot = off
-                flat_w = torch.zeros(n_tot, dtype=torch.bfloat16, device=device)
-                flat_g = torch.zeros(n_tot, dtype=torch.bfloat16, device=device)
-                for p, o in zip(g_params, offs):
-                    n = p.numel()
+                nw = off
+                for p in fused_b + [q for q in other if q.dtype == torch.float32]:
+                    off = (off + ALIGN - 1) // ALIGN * ALIGN
+                    offs[id(p)] = off
+                    off += p.numel()
+                flat_w = torch.zeros(nw, dtype=torch.bfloat16, device=device)
+                flat_g32 = torch.zeros(off, dtype=torch.float32, device=device)
+                for p in bf_params:
+                    o, n = offs[id(p)], p.numel()
                     flat_w[o : o + n].copy_(p.detach().view(-1))
                     with torch.no_grad():
                         p.data = flat_w[o : o + n].view_as(p)
-                    p.grad = flat_g[o : o + n].view_as(p)
+
+                def gview(p):
+                    return flat_g32[offs[id(p)] : offs[id(p)] + p.numel()].view_as(p)
+
+                for p in fused_w + fused_b:
+                    p._sideband_grad = gview(p)
+                other_bf = [p for p in other if p.dtype == torch.bfloat16]
+                other_f32 = [p for p in other if p.dtype == torch.float32]
+                for p in other_bf + other_f32:
+                    p.grad = torch.zeros_like(p)
+                o_grads = [p.grad for p in other_bf + other_f32]
+                o_views = [gview(p) for p in other_bf + other_f32]
+                fb_views = [p._sideband_grad for p in fused_b]
                 flat_master = flat_w.float()
-                flat_g32 = torch.zeros_like(flat_master)
                 lr = opt.param_groups[0]["lr"]
 
                 def iteration():
@@ -374,18 +394,16 @@ def main():
                     logits = model(static["dense"], static["base"])
                     loss = loss_fn(logits.float(), static["label"])
                     loss.backward()
-                    # same-dtype whole-buffer ops: cast-copy, f32 SGD, write
-                    # back, zero — 4 launches for the entire optimizer
-                    flat_g32.copy_(flat_g)
-                    flat_master.add_(flat_g32, alpha=-lr)
                     with torch.no_grad():
+                        for gsrc, gdst in zip(o_grads, o_views):
+                            gdst.copy_(gsrc)  # tiny head params
+                        flat_master.add_(flat_g32[:nw], alpha=-lr)
                         flat_w.copy_(flat_master)
-                    flat_g.zero_()
-                    if f32_params:
-                        g32 = [p.grad for p in f32_params]
-                        with torch.no_grad():
-                            torch._foreach_add_(f32_params, g32, alpha=-lr)
-                        torch._foreach_zero_(g32)
+                        if fused_b:
+                            torch._foreach_add_(fused_b, fb_views, alpha=-lr)
+                        flat_g32.zero_()
+                        if o_grads:
+                            torch._foreach_zero_(o_grads)
                     return loss
 
             elif bf16_weights:
@@ -476,6 +494,9 @@ def main():
                   file=_sys.stderr, flush=True)
             graph = None
             graph_upd = None
+            for p in model.parameters():  # sideband only valid under the graph
+                if hasattr(p, "_sideband_grad"):
+                    del p._sideband_grad
             if world > 1 and defer_ddp:
                 # the eager fallback still needs synchronized dense grads
                 for p in model.parameters():

@@ -210,7 +210,8 @@ using bf16x4t = __attribute__((ext_vector_type(4))) __bf16;
 
 __global__ __launch_bounds__(256) void wgrad_tr_kernel(
     const short* __restrict__ dC, const short* __restrict__ A,
-    float* __restrict__ dW, int M, int N, int K, int splitm) {
+    float* __restrict__ dW, int M, int N, int K, int splitm, int ldw,
+    int kmax, int accum) {
   constexpr int MT = 64;  // m-rows per buffer
   // panel layout per image: [8 panels][MT rows][16 cols] bf16 = 16 KB
   __shared__ short lds_c[2][8 * MT * 16];
@@ -304,12 +305,13 @@ __global__ __launch_bounds__(256) void wgrad_tr_kernel(
     for (int bk = 0; bk < 4; ++bk) {
       const int n = n0 + wr * 64 + an * 16 + (lane >> 4) * 4;
       const int k = k0 + wc * 64 + bk * 16 + fi;
+      if (k >= kmax) continue;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        if (splitm > 1)
-          atomicAdd(&dW[(int64_t)(n + r) * K + k], acc[an][bk][r]);
+        if (accum || splitm > 1)
+          atomicAdd(&dW[(int64_t)(n + r) * ldw + k], acc[an][bk][r]);
         else
-          dW[(int64_t)(n + r) * K + k] = acc[an][bk][r];
+          dW[(int64_t)(n + r) * ldw + k] = acc[an][bk][r];
       }
     }
   }
@@ -431,7 +433,8 @@ constexpr int WLD = WTM + PAD;
 
 __global__ __launch_bounds__(256) void wgrad_kernel(
     const short* __restrict__ dC, const short* __restrict__ A,
-    float* __restrict__ dW, int M, int N, int K, int splitm) {
+    float* __restrict__ dW, int M, int N, int K, int splitm, int ldw,
+    int kmax, int accum) {
   __shared__ short lds_dct[2][64 * WLD];  // [n][m]
   __shared__ short lds_at[2][64 * WLD];   // [k][m]
   // staging pieces: [WTM rows x 64 cols] / (256 threads x 8 shorts) = 2 each
@@ -526,15 +529,15 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
 #pragma unroll
     for (int bk = 0; bk < 2; ++bk) {
       const int k = k0 + wc * 32 + bk * 16 + fi;
-      if (k >= K) continue;
+      if (k >= kmax) continue;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int n = n0 + wr * 32 + an * 16 + (lane >> 4) * 4 + r;
         if (n >= N) continue;
-        if (splitm > 1)
-          atomicAdd(&dW[(int64_t)n * K + k], acc[an][bk][r]);
+        if (accum || splitm > 1)
+          atomicAdd(&dW[(int64_t)n * ldw + k], acc[an][bk][r]);
         else
-          dW[(int64_t)n * K + k] = acc[an][bk][r];
+          dW[(int64_t)n * ldw + k] = acc[an][bk][r];
       }
     }
   }
@@ -709,10 +712,8 @@ torch::Tensor relu_bwd(torch::Tensor g, torch::Tensor out) {
   return g_eff;
 }
 
-torch::Tensor bias_grad(torch::Tensor dC) {
+static void bias_grad_launch(torch::Tensor dC, torch::Tensor db) {
   const int M = (int)dC.size(0), N = (int)dC.size(1);
-  auto db = torch::zeros(
-      {N}, torch::TensorOptions().dtype(torch::kFloat32).device(dC.device()));
   // power-of-two N (every DLRM/DCN layer width): coalesced vector kernel.
   // grid*256 must be a multiple of N/8 so each thread's column octet is
   // loop-invariant; N/8 is a power of two <= 256's multiples, so any grid
@@ -735,21 +736,35 @@ torch::Tensor bias_grad(torch::Tensor dC) {
                        N * sizeof(float), dcur_stream(),
                        (const short*)dC.data_ptr(), db.data_ptr<float>(),
                        total8, N);
-    return db;
+    return;
   }
   const int rows_per_block = std::max(8, (M + 511) / 512);
   const int blocks = (M + rows_per_block - 1) / rows_per_block;
   hipLaunchKernelGGL(bias_grad_kernel, dim3(blocks), dim3(256), 0,
                      dcur_stream(), (const short*)dC.data_ptr(),
                      db.data_ptr<float>(), M, N, rows_per_block);
+}
+
+torch::Tensor bias_grad(torch::Tensor dC) {
+  auto db = torch::zeros(
+      {dC.size(1)},
+      torch::TensorOptions().dtype(torch::kFloat32).device(dC.device()));
+  bias_grad_launch(dC, db);
   return db;
 }
 
+// Side-band: atomically ACCUMULATE the column sums into the caller's
+// pre-zeroed f32 gradient slot (both kernels end in atomicAdd already).
+void bias_grad_into(torch::Tensor dC, torch::Tensor out) {
+  TORCH_CHECK(out.scalar_type() == torch::kFloat32 && out.is_contiguous());
+  TORCH_CHECK(out.numel() == dC.size(1));
+  bias_grad_launch(dC, out);
+}
+
 // dW [N,K] f32 = dC^T @ A
-torch::Tensor wgrad(torch::Tensor dC, torch::Tensor A) {
+static void wgrad_launch(torch::Tensor dC, torch::Tensor A, torch::Tensor out,
+                         int ldw, int kmax, int accum) {
   const int M = (int)dC.size(0), N = (int)dC.size(1), K = (int)A.size(1);
-  auto dW = torch::empty(
-      {N, K}, torch::TensorOptions().dtype(torch::kFloat32).device(A.device()));
   static const bool tr_on = [] {
     const char* e = getenv("PA_WGRAD_TR");
     return !e || atoi(e) != 0;
@@ -763,23 +778,40 @@ torch::Tensor wgrad(torch::Tensor dC, torch::Tensor A) {
     int splitm = 1;
     while (tiles2 * splitm < 512 && splitm < 64 && (M / (splitm * 2)) >= 64)
       splitm *= 2;
-    if (splitm > 1) dW.zero_();
+    if (splitm > 1 && !accum) out.zero_();
     hipLaunchKernelGGL(wgrad_tr_kernel, dim3(tiles2 * splitm), dim3(256), 0,
                        dcur_stream(), (const short*)dC.data_ptr(),
-                       (const short*)A.data_ptr(), dW.data_ptr<float>(), M, N,
-                       K, splitm);
-    return dW;
+                       (const short*)A.data_ptr(), out.data_ptr<float>(), M, N,
+                       K, splitm, ldw, kmax, accum);
+    return;
   }
   const int tiles = ((N + 63) / 64) * ((K + 63) / 64);
   int splitm = 1;
   while (tiles * splitm < 512 && splitm < 64 && (M / (splitm * 2)) >= 32)
     splitm *= 2;
-  if (splitm > 1) dW.zero_();
+  if (splitm > 1 && !accum) out.zero_();
   hipLaunchKernelGGL(wgrad_kernel, dim3(tiles * splitm), dim3(256), 0,
                      dcur_stream(), (const short*)dC.data_ptr(),
-                     (const short*)A.data_ptr(), dW.data_ptr<float>(), M, N, K,
-                     splitm);
+                     (const short*)A.data_ptr(), out.data_ptr<float>(), M, N,
+                     K, splitm, ldw, kmax, accum);
+}
+
+torch::Tensor wgrad(torch::Tensor dC, torch::Tensor A) {
+  const int N = (int)dC.size(1), K = (int)A.size(1);
+  auto dW = torch::empty(
+      {N, K}, torch::TensorOptions().dtype(torch::kFloat32).device(A.device()));
+  wgrad_launch(dC, A, dW, K, K, 0);
   return dW;
+}
+
+// Side-band accumulation: dW atomically ADDS into `out` (f32 [N, kmax],
+// row stride = out's width), the caller's pre-zeroed flat gradient slot —
+// no dW materialization, no cast, no pad-column slice, no AccumulateGrad
+// add.  `out` may be narrower than A (kmax < K drops the zero pad columns).
+void wgrad_into(torch::Tensor dC, torch::Tensor A, torch::Tensor out) {
+  TORCH_CHECK(out.scalar_type() == torch::kFloat32 && out.is_contiguous());
+  TORCH_CHECK(out.size(0) == dC.size(1) && out.size(1) <= A.size(1));
+  wgrad_launch(dC, A, out, (int)out.size(1), (int)out.size(1), 1);
 }
 
 void init_dense(pybind11::module_& m) {
@@ -788,4 +820,8 @@ void init_dense(pybind11::module_& m) {
   m.def("relu_bwd", &relu_bwd, "g * (out > 0)");
   m.def("bias_grad", &bias_grad, "column-sum bias gradient");
   m.def("wgrad", &wgrad, "dW = dC^T @ A (f32 out)");
+  m.def("wgrad_into", &wgrad_into,
+        "dW accumulated straight into a flat f32 grad slot");
+  m.def("bias_grad_into", &bias_grad_into,
+        "bias grad accumulated straight into a flat f32 grad slot");
 }

@@ -50,6 +50,12 @@ class FusedLinearFn(torch.autograd.Function):
         ctx.k = x.shape[1]
         ctx.w_dtype = weight.dtype
         ctx.b_dtype = bias.dtype
+        # side-band gradient targets (set by the bench's flat-buffer scheme):
+        # when present, backward ACCUMULATES dw/db straight into these f32
+        # slots via atomics and returns None — no dw materialization, cast,
+        # pad-slice or AccumulateGrad add per layer
+        ctx.w_side = getattr(weight, "_sideband_grad", None)
+        ctx.b_side = getattr(bias, "_sideband_grad", None)
         return out
 
     @staticmethod
@@ -61,9 +67,13 @@ class FusedLinearFn(torch.autograd.Function):
         g = g.to(torch.bfloat16).contiguous()
         if ctx.act == 1:
             g = C.relu_bwd(g, out)
-        db = C.bias_grad(g)
-        if db.dtype != ctx.b_dtype:
-            db = db.to(ctx.b_dtype)
+        if ctx.b_side is not None:
+            C.bias_grad_into(g, ctx.b_side)
+            db = None
+        else:
+            db = C.bias_grad(g)
+            if db.dtype != ctx.b_dtype:
+                db = db.to(ctx.b_dtype)
         # dX = g @ W: trans_b path consumes the [N, Kp] weight directly; at
         # the big square shape the transposed-staging kernel loses to an
         # explicit W^T + the glds NT path (51 vs 33+8 us measured,
@@ -79,7 +89,10 @@ class FusedLinearFn(torch.autograd.Function):
             )
         if dx.shape[1] != ctx.k:
             dx = dx[:, : ctx.k]
-        if _wgrad_via_blas():
+        if ctx.w_side is not None:
+            C.wgrad_into(g, x_bf, ctx.w_side)
+            dw = None
+        elif _wgrad_via_blas():
             # hybrid: the hand-written fwd/dgrad kernels beat hipBLASLt on
             # these shapes but the wgrad kernel trails it (transpose-staging
             # bound) — let the library run the one GEMM it wins
@@ -88,7 +101,7 @@ class FusedLinearFn(torch.autograd.Function):
             dw = C.wgrad(g, x_bf)
             if dw.dtype != ctx.w_dtype:
                 dw = dw.to(ctx.w_dtype)
-        if dw.shape[1] != ctx.k:
+        if dw is not None and dw.shape[1] != ctx.k:
             dw = dw[:, : ctx.k].contiguous()
         return dx, dw, db, None
 
