@@ -276,3 +276,43 @@ def test_fused_dcn_matches_eager():
         (o_e - o_f).abs().max().item()
     )
     assert all(g > 0 for g in g_f), "fused DCN: some weight got no gradient"
+
+
+def test_sideband_grad_accumulation_matches_standard():
+    """FusedLinear with _sideband_grad targets: the atomically-accumulated
+    flat f32 gradients must equal the standard autograd path's."""
+    from persia_amd.ops.dense import FusedLinear
+
+    torch.manual_seed(11)
+    M, K, N = 1024, 480, 512
+    x = torch.randn(M, K, device=_dev(), dtype=torch.bfloat16)
+    layer = FusedLinear(K, N, relu=True).to(_dev())
+    layer.bfloat16()  # bias stays f32 (_apply override)
+    assert layer.bias.dtype == torch.float32
+
+    # standard path
+    out = layer(x)
+    g = torch.randn_like(out.float()).to(torch.bfloat16)
+    out.backward(g)
+    dw_std = layer.weight.grad.float().clone()
+    db_std = layer.bias.grad.clone()
+
+    # sideband path
+    layer.weight.grad = None
+    layer.bias.grad = None
+    wslot = torch.zeros(N, K, dtype=torch.float32, device=_dev())
+    bslot = torch.zeros(N, dtype=torch.float32, device=_dev())
+    layer.weight._sideband_grad = wslot
+    layer.bias._sideband_grad = bslot
+    out2 = layer(x)
+    out2.backward(g)
+    assert layer.weight.grad is None and layer.bias.grad is None
+    assert torch.allclose(wslot, dw_std, atol=0.5, rtol=0.05), (
+        (wslot - dw_std).abs().max().item()
+    )
+    assert torch.allclose(bslot, db_std, atol=0.5, rtol=0.02)
+    # accumulation: second backward ADDS
+    out3 = layer(x)
+    out3.backward(g)
+    assert torch.allclose(wslot, 2 * dw_std, atol=1.0, rtol=0.05)
+    del layer.weight._sideband_grad, layer.bias._sideband_grad
