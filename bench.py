@@ -26,20 +26,14 @@ PRESETS = {
                       batch_size=128),
     "criteo": dict(model="dlrm", num_sparse=26, num_dense=13, dim=128, rows=1e8),
     "terabyte": dict(model="dlrm", num_sparse=26, num_dense=13, dim=128, rows=1e10),
-    # flat_params=0 on the sparse-heavy presets: the flat scheme's gapless
-    # dense replay starves the concurrent sparse stream (scatter_update/sort
-    # stretched ~10x, measured) when that stream is the critical path
-    # fused_dense=0: hipBLASLt wins the DCN's wide (K~1700) cross/deep
-    # GEMMs (measured 3.39M vs 2.34M fused end-to-end)
     # sparse_priority=-1: dcn is sparse-stream-critical; raising that
     # stream's priority measured 2.67 -> 2.98M (neutral on criteo, slightly
-    # negative on terabyte)
+    # negative on terabyte).  After the side-band flat-gradient scheme the
+    # fused dense path wins on EVERY preset (dcn 3.0 -> 3.26M, 100t
+    # 5.25 -> 5.56M re-measured same box).
     "dcn-spill": dict(model="dcn", num_sparse=26, num_dense=13, dim=64, rows=1e11,
-                      spill_capacity=2e8, flat_params=0, fused_dense=0,
-                      sparse_priority=-1),
-    # dim-8 towers don't suit the fused MFMA layers (skinny K, odd widths)
-    "100t": dict(model="dlrm", num_sparse=64, num_dense=13, dim=8, rows=1e12,
-                 fused_dense=0, flat_params=0),
+                      spill_capacity=2e8, sparse_priority=-1),
+    "100t": dict(model="dlrm", num_sparse=64, num_dense=13, dim=8, rows=1e12),
 }
 
 
