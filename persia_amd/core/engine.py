@@ -337,7 +337,10 @@ class EmbeddingEngine:
         import os as _os
 
         self._prod_timing = _os.environ.get("PA_PROD_TIMING", "0") == "1"
-        self._pt = {"prep": 0.0, "native": 0.0, "batch": 0.0, "n": 0}
+        # timers skip the first batches: one-time ATen kernel-module loads
+        # (~hundreds of ms) otherwise dominate the averages
+        self._pt = {"prep": 0.0, "native": 0.0, "batch": 0.0, "n": 0,
+                    "skip": 20}
         # fused distributed path (capacity-padded even a2a, sync-free);
         # PA_FUSED_DIST=0 falls back to the exact two-phase counts exchange,
         # PA_FORCE_DIST=1 exercises the padded machinery at world_size=1
@@ -530,14 +533,14 @@ class EmbeddingEngine:
             recv_keys = comm.all_to_all_even(send_keys[: world * cap])
         else:
             recv_keys = send_keys[: world * cap].clone()
-        if self._prod_timing:
+        if self._prod_timing and getattr(self, "_prod_timing_live", True):
             self._pt["route"] = self._pt.get("route", 0.0) + (
                 time.perf_counter() - _tr
             )
             _tr = time.perf_counter()
         store = self.stores[group.dim]
         rows_local = store.lookup_wire(recv_keys, train, self.wire_dtype)
-        if self._prod_timing:
+        if self._prod_timing and getattr(self, "_prod_timing_live", True):
             self._pt["lkw"] = self._pt.get("lkw", 0.0) + (
                 time.perf_counter() - _tr
             )
@@ -569,6 +572,11 @@ class EmbeddingEngine:
 
     def process_batch(self, batch: PersiaBatch, train: Optional[bool] = None) -> PersiaTrainingBatch:
         if self._prod_timing:
+            if self._pt["skip"] > 0:
+                self._pt["skip"] -= 1
+                self._prod_timing_live = False
+            else:
+                self._prod_timing_live = True
             _tb0 = time.perf_counter()
         self._batch_counter += 1
         sample = (
@@ -613,7 +621,7 @@ class EmbeddingEngine:
         out._sorted = False
         if sample:
             self._record_lookup_metrics(batch, out, _ms_t0, _ms_ev0)
-        if self._prod_timing:
+        if self._prod_timing and self._prod_timing_live:
             self._pt["batch"] += time.perf_counter() - _tb0
             self._pt["n"] += 1
         return out
@@ -837,8 +845,9 @@ class EmbeddingEngine:
         if not self.dist.distributed and store.spill is None and not self._force_dist:
             # whole lookup in ONE native call (C++ drives sign prep, dedup,
             # probe/insert, gather and the fused segment-sum)
-            if self._prod_timing:
+            if self._prod_timing and self._prod_timing_live:
                 self._pt["prep"] += time.perf_counter() - _t0
+            if self._prod_timing:
                 _t1 = time.perf_counter()
             lo, hi = self.hyper.emb_initialization
             vals_t = _upload()
@@ -904,13 +913,13 @@ class EmbeddingEngine:
                 # capacity-padded even a2a (no host count syncs anywhere);
                 # compose the unpack gather into `inverse` so segment_sum
                 # reads the recv buffer directly — no [nnz, dim] row gather
-                if self._prod_timing:
+                if self._prod_timing and getattr(self, "_prod_timing_live", True):
                     self._pt["dedup"] = self._pt.get("dedup", 0.0) + (
                         time.perf_counter() - _t0
                     )
                     _tx = time.perf_counter()
                 rows_full, idx = self._a2a_exchange_fwd(plan, group, train)
-                if self._prod_timing:
+                if self._prod_timing and getattr(self, "_prod_timing_live", True):
                     self._pt["exch"] = self._pt.get("exch", 0.0) + (
                         time.perf_counter() - _tx
                     )
@@ -919,7 +928,7 @@ class EmbeddingEngine:
                 sums = C.segment_sum(
                     rows_full, inverse2, plan.cat_offsets, plan.empty_scale
                 )
-                if self._prod_timing:
+                if self._prod_timing and getattr(self, "_prod_timing_live", True):
                     self._pt["sum"] = self._pt.get("sum", 0.0) + (
                         time.perf_counter() - _tx
                     )
@@ -927,7 +936,7 @@ class EmbeddingEngine:
                 out._lazy_sum_groups.append(
                     (group, [sc.name for sc in slot_ctxs], B)
                 )
-                if self._prod_timing:
+                if self._prod_timing and self._prod_timing_live:
                     self._pt["native"] += time.perf_counter() - _t0
                 return group
             rows = self._exchange_rows(group, train)
