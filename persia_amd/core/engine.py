@@ -497,6 +497,19 @@ class EmbeddingEngine:
         mapping each unique to its send slot; invalid/overflow -> the dummy
         tail slot).  Pure device ops — sync-free."""
         world, cap = plan.a2a_world, plan.a2a_cap
+        if self.device.type == "cuda":
+            # fused route: one bounds kernel (the owner partition of the
+            # sorted keys is a range partition -> binary searches) + one
+            # packed scatter — replaces the ~10-dispatch torch chain below
+            # (measured 0.51 ms/batch of producer issue time)
+            from persia_amd.ops import native as _native
+
+            send, idx = _native().a2a_route(
+                uniq,
+                u_count if u_count is not None else self._empty_i64(),
+                world, cap, self._a2a_overflow,
+            )
+            return send, idx
         n = uniq.numel()
         ar = plan.a2a_ar[:n]
         owner = _owner_of_keys(uniq, world)

@@ -525,6 +525,62 @@ __global__ void sign_prep_stack_kernel(
   }
 }
 
+// ------------------------------------------------------- padded a2a routing
+// The owner partition of the SORTED uniq keys is a range partition (owner =
+// (hi32*world)>>32 is monotone in the key), so per-owner segment starts are
+// world binary searches — one tiny kernel — and the pack is one grid-stride
+// scatter.  Replaces a ~10-dispatch torch chain that cost the producer
+// thread ~0.5 ms/batch of issue time.
+__global__ void a2a_bounds_kernel(const ull* __restrict__ uniq,
+                                  const long long* __restrict__ n_dev,
+                                  long long n_pad, int world,
+                                  long long* __restrict__ starts) {
+  const long long n = n_dev ? *n_dev : n_pad;
+  const int r = threadIdx.x;
+  if (r > world) return;
+  if (r == 0) { starts[0] = 0; return; }
+  if (r == world) { starts[world] = n; return; }
+  // smallest key with owner >= r: hi32 >= ceil(r * 2^32 / world)
+  const ull hi_min =
+      (ull)((((unsigned __int128)r << 32) + (ull)world - 1) / (ull)world);
+  const ull thresh = hi_min << 32;
+  long long lo = 0, hi = n;
+  while (lo < hi) {
+    const long long mid = (lo + hi) >> 1;
+    if (uniq[mid] < thresh) lo = mid + 1; else hi = mid;
+  }
+  starts[r] = lo;
+}
+
+__global__ void a2a_scatter_kernel(const ull* __restrict__ uniq,
+                                   const long long* __restrict__ n_dev,
+                                   long long n_pad,
+                                   const long long* __restrict__ starts,
+                                   int world, long long cap,
+                                   ull* __restrict__ send,
+                                   int64_t* __restrict__ idx,
+                                   long long* __restrict__ overflow) {
+  const long long n = n_dev ? *n_dev : n_pad;
+  const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n_pad) return;
+  const long long dummy = (long long)world * cap;
+  if (i >= n) {  // dedup padding tail
+    idx[i] = dummy;
+    return;
+  }
+  const ull k = uniq[i];
+  const int o = (int)(((k >> 32) * (ull)world) >> 32);
+  const long long pos = i - starts[o];
+  long long d = dummy;
+  if (pos < cap) {
+    d = (long long)o * cap + pos;
+    send[d] = k;
+  } else {
+    atomicAdd((unsigned long long*)overflow, 1ull);
+  }
+  idx[i] = d;
+}
+
 __global__ void iota_i64_kernel(int64_t* __restrict__ out, int64_t n) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i < n) out[i] = i;
@@ -1633,6 +1689,33 @@ void scatter_update(torch::Tensor table_keys, torch::Tensor ticks,
                          : nullptr);
 }
 
+// -> (send [world*cap+1] zero-padded, idx [n_pad]); overflow accumulates
+// into the caller's persistent device counter
+std::vector<torch::Tensor> a2a_route(torch::Tensor uniq, torch::Tensor u_count,
+                                     int64_t world, int64_t cap,
+                                     torch::Tensor overflow) {
+  const int64_t n_pad = uniq.numel();
+  auto opts = torch::TensorOptions().dtype(torch::kInt64).device(uniq.device());
+  auto send = torch::zeros({world * cap + 1}, opts);
+  auto idx = torch::empty({n_pad}, opts);
+  auto starts = torch::empty({world + 1}, opts);
+  hipStream_t st = cur_stream();
+  const long long* n_dev =
+      u_count.numel() ? (const long long*)u_count.data_ptr<int64_t>() : nullptr;
+  hipLaunchKernelGGL(a2a_bounds_kernel, dim3(1), dim3(world + 1), 0, st,
+                     (const ull*)uniq.data_ptr<int64_t>(), n_dev, n_pad,
+                     (int)world, (long long*)starts.data_ptr<int64_t>());
+  if (n_pad)
+    hipLaunchKernelGGL(a2a_scatter_kernel, dim3(n_blocks_for(n_pad, 256)),
+                       dim3(256), 0, st, (const ull*)uniq.data_ptr<int64_t>(),
+                       n_dev, n_pad,
+                       (const long long*)starts.data_ptr<int64_t>(), (int)world,
+                       cap, (ull*)send.data_ptr<int64_t>(),
+                       idx.data_ptr<int64_t>(),
+                       (long long*)overflow.data_ptr<int64_t>());
+  return {send, idx};
+}
+
 torch::Tensor sign_prep(torch::Tensor values, torch::Tensor slot_starts,
                         torch::Tensor prefixes, int64_t spacing) {
   const int64_t n = values.numel();
@@ -1711,6 +1794,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dedup_finalize", &dedup_finalize,
         "fixed-shape dedup epilogue (padded uniq/ustarts + device count)");
   m.def("sort_pairs_u64", &sort_pairs_u64, "radix sort-pairs, u64 order");
+  m.def("a2a_route", &a2a_route,
+        "padded-a2a key routing: range-partition bounds + packed scatter");
   m.def("neq_flags", &neq_flags, "sorted-run boundary flags");
   m.def("scatter_update", &scatter_update,
         "fused ordered grad scatter + optimizer update (no [U,dim] buffer)");
