@@ -569,10 +569,14 @@ class EmbeddingEngine:
         group.a2a_plan = plan
         group.a2a_idx = idx
         group.a2a_recv_keys = recv_keys
-        # owner-side dedup for the backward merge happens lazily in
-        # _a2a_backward_native: the pipeline (producer) thread is the scaling
-        # bottleneck at world=8, the main thread has slack — and infer-mode
-        # batches never need it
+        if train and self.device.type == "cuda":
+            # owner-side dedup for the backward merge, issued HERE so the
+            # sort overlaps the dense step on the pipeline stream (the fused
+            # a2a_route freed the producer budget this costs);
+            # _a2a_backward_native falls back lazily if absent (infer skips)
+            from persia_amd.ops import native as _native
+
+            group.a2a_owner_dedup = tuple(_native().dedup_padded(recv_keys))
         return rows_full, idx
 
     def check_a2a_overflow(self) -> int:
