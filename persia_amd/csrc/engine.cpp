@@ -180,16 +180,49 @@ void update_local(torch::Tensor grads, torch::Tensor perm,
 // loops in the pipeline thread (reference: remote CPU parameter-server
 // shards, embedding_parameter_server/lib.rs).
 #include <list>
+#include <memory>
 #include <unordered_map>
+#include <vector>
 
 struct NativeHostTier {
+  // Row payloads live in fixed-size slabs with a free-list (two heap
+  // allocations PER INSERT — list node + row vector — dominated the
+  // original implementation and fragmented the host heap at 1e8-row
+  // scale); the key map is sharded 16 ways so rehash pauses are 1/16th
+  // the size.  LRU stays one global list of (key, slot).
+  static constexpr int64_t CHUNK_ROWS = 1 << 16;
+  static constexpr int SHARDS = 16;
   int64_t capacity, row_width;
-  std::list<std::pair<uint64_t, std::vector<float>>> lru;  // front = oldest
+  std::vector<std::unique_ptr<float[]>> chunks;
+  std::vector<int64_t> free_slots;
+  int64_t next_fresh = 0;  // slots handed out so far (before any free)
+  std::list<std::pair<uint64_t, int64_t>> lru;  // (key, slot); front = oldest
   std::unordered_map<uint64_t,
-                     std::list<std::pair<uint64_t, std::vector<float>>>::iterator>
-      map;
+                     std::list<std::pair<uint64_t, int64_t>>::iterator>
+      map[SHARDS];
 
   NativeHostTier(int64_t cap, int64_t rw) : capacity(cap), row_width(rw) {}
+
+  float* rowp(int64_t slot) {
+    return chunks[slot / CHUNK_ROWS].get() + (slot % CHUNK_ROWS) * row_width;
+  }
+  int64_t alloc_slot() {
+    if (!free_slots.empty()) {
+      const int64_t s = free_slots.back();
+      free_slots.pop_back();
+      return s;
+    }
+    if (next_fresh >= (int64_t)chunks.size() * CHUNK_ROWS)
+      chunks.emplace_back(new float[CHUNK_ROWS * row_width]);
+    return next_fresh++;
+  }
+  auto& shard(uint64_t key) { return map[key & (SHARDS - 1)]; }
+
+  void erase_entry(std::list<std::pair<uint64_t, int64_t>>::iterator it) {
+    free_slots.push_back(it->second);
+    shard(it->first).erase(it->first);
+    lru.erase(it);
+  }
 
   void insert(torch::Tensor keys, torch::Tensor rows) {
     TORCH_CHECK(!keys.is_cuda() && !rows.is_cuda(), "HostTier: cpu tensors");
@@ -201,19 +234,22 @@ struct NativeHostTier {
     const int64_t k = kc.numel();
     for (int64_t i = 0; i < k; ++i) {
       const uint64_t key = (uint64_t)kp[i];
-      auto it = map.find(key);
-      if (it != map.end()) {
+      auto& m = shard(key);
+      auto it = m.find(key);
+      int64_t slot;
+      if (it != m.end()) {  // refresh: reuse the slot, move to back
+        slot = it->second->second;
         lru.erase(it->second);
-        map.erase(it);
+        lru.emplace_back(key, slot);
+        it->second = std::prev(lru.end());
+      } else {
+        slot = alloc_slot();
+        lru.emplace_back(key, slot);
+        m[key] = std::prev(lru.end());
       }
-      lru.emplace_back(key, std::vector<float>(rp + i * row_width,
-                                               rp + (i + 1) * row_width));
-      map[key] = std::prev(lru.end());
+      std::copy(rp + i * row_width, rp + (i + 1) * row_width, rowp(slot));
     }
-    while ((int64_t)lru.size() > capacity) {
-      map.erase(lru.front().first);
-      lru.pop_front();
-    }
+    while ((int64_t)lru.size() > capacity) erase_entry(lru.begin());
   }
 
   std::tuple<torch::Tensor, torch::Tensor> fetch(torch::Tensor keys) {
@@ -225,13 +261,14 @@ struct NativeHostTier {
     float* rp = rows.data_ptr<float>();
     bool* fp = found.data_ptr<bool>();
     for (int64_t i = 0; i < k; ++i) {
-      auto it = map.find((uint64_t)kp[i]);
-      if (it == map.end()) continue;
-      std::copy(it->second->second.begin(), it->second->second.end(),
-                rp + i * row_width);
+      const uint64_t key = (uint64_t)kp[i];
+      auto& m = shard(key);
+      auto it = m.find(key);
+      if (it == m.end()) continue;
+      const float* src = rowp(it->second->second);
+      std::copy(src, src + row_width, rp + i * row_width);
       fp[i] = true;
-      lru.erase(it->second);
-      map.erase(it);
+      erase_entry(it->second);
     }
     return {rows, found};
   }
@@ -245,17 +282,23 @@ struct NativeHostTier {
     int64_t i = 0;
     for (auto& e : lru) {
       kp[i] = (int64_t)e.first;
-      std::copy(e.second.begin(), e.second.end(), rp + i * row_width);
+      const float* src = rowp(e.second);
+      std::copy(src, src + row_width, rp + i * row_width);
       ++i;
     }
     return {keys, rows};
   }
 
   int64_t size() const { return (int64_t)lru.size(); }
-  bool contains(int64_t key) const { return map.count((uint64_t)key) != 0; }
+  bool contains(int64_t key) {
+    return shard((uint64_t)key).count((uint64_t)key) != 0;
+  }
   void clear() {
     lru.clear();
-    map.clear();
+    for (auto& m : map) m.clear();
+    chunks.clear();
+    free_slots.clear();
+    next_fresh = 0;
   }
 };
 
