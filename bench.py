@@ -285,48 +285,7 @@ def main():
             # launches replace the per-layer cast storm.
             # PA_GRAPH_BF16=0 selects the autocast variant (A/B switch).
             # (model.bfloat16() itself ran before the DDP wrap, above)
-            if bf16_weights and world > 1:
-                import torch.distributed as dist
-
-                # flat grads PER DTYPE (FusedLinear biases stay f32 — see
-                # ops/dense.py _apply); each flat buffer gets one allreduce
-                g_params = list(model.parameters())
-                flat_by_dt = {}
-                for dt in {p.dtype for p in g_params}:
-                    n_dt = sum(p.numel() for p in g_params if p.dtype == dt)
-                    flat_by_dt[dt] = [torch.zeros(n_dt, dtype=dt, device=device), 0]
-                for p in g_params:
-                    buf, off = flat_by_dt[p.dtype]
-                    p.grad = buf[off : off + p.numel()].view_as(p)
-                    flat_by_dt[p.dtype][1] = off + p.numel()
-                flat_grads = flat_by_dt[torch.bfloat16][0]
-                all_flats = [v[0] for v in flat_by_dt.values()]
-                g_masters = [p.detach().clone().float() for p in g_params]
-                g_grads32 = [torch.zeros_like(m) for m in g_masters]
-                lr = opt.param_groups[0]["lr"]
-
-                def iteration():
-                    static["base"].grad.zero_()
-                    torch._foreach_zero_(all_flats)
-                    logits = model(static["dense"], static["base"])
-                    loss = loss_fn(logits.float(), static["label"])
-                    loss.backward()
-                    return loss
-
-                def allreduce_grads():
-                    for f in all_flats:
-                        dist.all_reduce(f)
-
-                def update_body():
-                    grads = [p.grad for p in g_params]
-                    torch._foreach_copy_(g_grads32, grads)
-                    # allreduce delivers the SUM; fold the 1/world average
-                    # into the SGD step
-                    torch._foreach_add_(g_masters, g_grads32, alpha=-lr / world)
-                    with torch.no_grad():
-                        torch._foreach_copy_(g_params, g_masters)
-
-            elif bf16_weights and bool(args.flat_params):
+            if bf16_weights and (world > 1 or bool(args.flat_params)):
                 # SIDE-BAND flat scheme: every FusedLinear accumulates its
                 # weight/bias gradient straight into ONE f32 flat buffer via
                 # atomics inside wgrad/bias_grad (ops/dense.py _sideband) —
@@ -383,6 +342,19 @@ def main():
                 flat_master = flat_w.float()
                 lr = opt.param_groups[0]["lr"]
 
+                def sgd_body():
+                    # allreduce (world>1) delivers the SUM: fold the 1/world
+                    # average into the step
+                    with torch.no_grad():
+                        flat_master.add_(flat_g32[:nw], alpha=-lr / world)
+                        flat_w.copy_(flat_master)
+                        if fused_b:
+                            torch._foreach_add_(fused_b, fb_views,
+                                                alpha=-lr / world)
+                        flat_g32.zero_()
+                        if o_grads:
+                            torch._foreach_zero_(o_grads)
+
                 def iteration():
                     static["base"].grad.zero_()
                     logits = model(static["dense"], static["base"])
@@ -391,14 +363,21 @@ def main():
                     with torch.no_grad():
                         for gsrc, gdst in zip(o_grads, o_views):
                             gdst.copy_(gsrc)  # tiny head params
-                        flat_master.add_(flat_g32[:nw], alpha=-lr)
-                        flat_w.copy_(flat_master)
-                        if fused_b:
-                            torch._foreach_add_(fused_b, fb_views, alpha=-lr)
-                        flat_g32.zero_()
-                        if o_grads:
-                            torch._foreach_zero_(o_grads)
+                    if world == 1:
+                        sgd_body()
                     return loss
+
+                if world > 1:
+                    import torch.distributed as dist
+
+                    # two-graph scheme: fwd+bwd graph, ONE f32 allreduce of
+                    # the side-band flat gradients between the graphs, then
+                    # the SGD graph (RCCL never captured)
+                    flat_grads = flat_g32
+                    update_body = sgd_body
+
+                    def allreduce_grads():
+                        dist.all_reduce(flat_g32)
 
             elif bf16_weights:
                 # per-param foreach variant (PA_FLAT_PARAMS=0 A/B switch)
