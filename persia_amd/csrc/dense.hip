@@ -557,6 +557,40 @@ __global__ void relu_bwd_kernel(const short* __restrict__ g,
   *(bf16x8*)(g_eff + i) = r;
 }
 
+// relu backward FUSED with the bias column-sum: the masked gradient is the
+// exact tensor bias_grad would re-read (16 MB round-trip + a launch per
+// layer saved).  Same column-invariance + LDS-reduce + small atomic fan-in
+// structure as bias_grad_vec_kernel; power-of-two N only.
+__global__ __launch_bounds__(256) void relu_bwd_bias_kernel(
+    const short* __restrict__ g, const short* __restrict__ out,
+    short* __restrict__ g_eff, float* __restrict__ db, int64_t total8,
+    int N) {
+  extern __shared__ float lds_col[];  // [N]
+  const int n8 = N / 8;
+  for (int i = threadIdx.x; i < N; i += blockDim.x) lds_col[i] = 0.0f;
+  __syncthreads();
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int c8 = (int)((idx % n8) * 8);  // invariant: stride % n8 == 0
+  float acc[8] = {};
+  for (; idx < total8; idx += stride) {
+    const bf16x8 gv = *(const bf16x8*)(g + idx * 8);
+    const bf16x8 ov = *(const bf16x8*)(out + idx * 8);
+    bf16x8 r;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      r[j] = bf2f(ov[j]) > 0.0f ? gv[j] : (short)0;
+      acc[j] += bf2f(r[j]);
+    }
+    *(bf16x8*)(g_eff + idx * 8) = r;
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) atomicAdd(&lds_col[c8 + j], acc[j]);
+  __syncthreads();
+  for (int i = threadIdx.x; i < N; i += blockDim.x)
+    atomicAdd(&db[i], lds_col[i]);
+}
+
 // db[n] = sum_m dC[m][n]: each block reduces a chunk of rows over ALL cols
 // (threads stride the row -> fully coalesced); one atomicAdd per (block,col).
 __global__ void bias_grad_kernel(const short* __restrict__ dC,
@@ -700,6 +734,27 @@ torch::Tensor gemm_nt_bias_act(torch::Tensor A, torch::Tensor B,
   return C;
 }
 
+// -> g_eff; db accumulated into the given f32 buffer (pre-zeroed slot or
+// torch.zeros).  Falls back to the unfused pair for non-pow2 N.
+torch::Tensor relu_bwd_bias(torch::Tensor g, torch::Tensor out,
+                            torch::Tensor db) {
+  const int64_t M = g.size(0), N = g.size(1);
+  TORCH_CHECK(db.scalar_type() == torch::kFloat32 && db.numel() == N);
+  const bool pow2 = (N & (N - 1)) == 0 && N >= 8 && N <= 8192;
+  TORCH_CHECK(pow2, "relu_bwd_bias: power-of-two N only");
+  auto g_eff = torch::empty_like(g);
+  const int64_t total8 = M * N / 8;
+  const int n8 = (int)(N / 8);
+  int blocks = (int)std::min<int64_t>(256, (total8 + 255) / 256);
+  if (n8 > 256) blocks = ((blocks + n8 / 256 - 1) / (n8 / 256)) * (n8 / 256);
+  hipLaunchKernelGGL(relu_bwd_bias_kernel, dim3(blocks), dim3(256),
+                     (int)N * sizeof(float), dcur_stream(),
+                     (const short*)g.data_ptr(), (const short*)out.data_ptr(),
+                     (short*)g_eff.data_ptr(), db.data_ptr<float>(), total8,
+                     (int)N);
+  return g_eff;
+}
+
 torch::Tensor relu_bwd(torch::Tensor g, torch::Tensor out) {
   auto g_eff = torch::empty_like(g);
   const int64_t n = g.numel();
@@ -818,6 +873,8 @@ void init_dense(pybind11::module_& m) {
   m.def("gemm_nt_bias_act", &gemm_nt_bias_act,
         "bf16 MFMA GEMM (A @ B^T) with fused bias+activation");
   m.def("relu_bwd", &relu_bwd, "g * (out > 0)");
+  m.def("relu_bwd_bias", &relu_bwd_bias,
+        "relu backward fused with the bias column-sum accumulation");
   m.def("bias_grad", &bias_grad, "column-sum bias gradient");
   m.def("wgrad", &wgrad, "dW = dC^T @ A (f32 out)");
   m.def("wgrad_into", &wgrad_into,

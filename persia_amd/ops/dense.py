@@ -65,15 +65,23 @@ class FusedLinearFn(torch.autograd.Function):
         C = native()
         x_bf, w_bf, out = ctx.saved_tensors
         g = g.to(torch.bfloat16).contiguous()
-        if ctx.act == 1:
-            g = C.relu_bwd(g, out)
-        if ctx.b_side is not None:
-            C.bias_grad_into(g, ctx.b_side)
+        N = g.shape[1]
+        pow2 = N >= 8 and (N & (N - 1)) == 0
+        if ctx.act == 1 and pow2 and ctx.b_side is not None:
+            # one fused pass: relu mask + bias column sums straight into the
+            # side-band slot (saves re-reading the 16 MB masked gradient)
+            g = C.relu_bwd_bias(g, out, ctx.b_side)
             db = None
         else:
-            db = C.bias_grad(g)
-            if db.dtype != ctx.b_dtype:
-                db = db.to(ctx.b_dtype)
+            if ctx.act == 1:
+                g = C.relu_bwd(g, out)
+            if ctx.b_side is not None:
+                C.bias_grad_into(g, ctx.b_side)
+                db = None
+            else:
+                db = C.bias_grad(g)
+                if db.dtype != ctx.b_dtype:
+                    db = db.to(ctx.b_dtype)
         # dX = g @ W: trans_b path consumes the [N, Kp] weight directly; at
         # the big square shape the transposed-staging kernel loses to an
         # explicit W^T + the glds NT path (51 vs 33+8 us measured,
