@@ -25,9 +25,10 @@ def _wgrad_via_blas() -> bool:
     return os.environ.get("PA_FUSED_WGRAD", "own") == "blas"
 
 
-def _pad32(x: torch.Tensor, dim: int, mult: int = 64) -> torch.Tensor:
-    """Zero-pad dim to a multiple of `mult` (64: the glds-pipelined GEMM v2
-    needs K % 64; zero K-columns are inert)."""
+def _pad32(x: torch.Tensor, dim: int, mult: int = 128) -> torch.Tensor:
+    """Zero-pad dim to a multiple of `mult` (the glds-pipelined GEMM v2
+    needs K % 64 and the tr-read wgrad K % 128; zero K-columns are inert —
+    the extra flops are cheaper than falling off the v2 kernels)."""
     k = x.shape[dim]
     pad = (-k) % mult
     if pad == 0:
