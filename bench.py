@@ -557,6 +557,17 @@ def main():
         elapsed = float(t.item())
 
     pipeline.stop()
+    if world > 1:
+        # the padded-a2a fast path drops keys to a dummy slot on bucket
+        # overflow (probability ~exp(-20 sigma) with the default slack);
+        # surface it loudly if the heuristic ever failed
+        ovf = engine.check_a2a_overflow()
+        if ovf:
+            import sys as _sy
+
+            print(f"# WARNING: padded-a2a bucket overflow on {ovf} batches "
+                  f"(keys dropped; widen the cap slack)",
+                  file=_sy.stderr, flush=True)
     if timing and tstats["n"]:
         import sys as _s
 
