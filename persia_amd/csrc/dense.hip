@@ -328,17 +328,20 @@ __global__ __launch_bounds__(256) void wgrad_tr_kernel(
 // conflict-spreading XOR applied to the SOURCE addresses and the fragment
 // reads (the same involution on both sides — rule 21).
 // Constraints: K % 64 == 0, N % 128 == 0 (launcher falls back otherwise).
-template <int ACT, bool STORE_F32>
+template <int ACT, bool STORE_F32, int TBN>
 __global__ __launch_bounds__(256) void gemm_nt_glds_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     const float* __restrict__ bias, void* __restrict__ C, int M, int N,
     int K) {
   constexpr int TBK = 64;  // 128 B per image row
+  constexpr int WN = TBN / 2;      // per-wave N extent (2x2 wave grid)
+  constexpr int NFRAG = WN / 16;
+  constexpr int BCH = TBN / 32;    // B glds chunks per wave
   __shared__ short lds_a[2][128 * TBK];
-  __shared__ short lds_b[2][128 * TBK];
-  const int n_tiles_n = N / 128;
+  __shared__ short lds_b[2][TBN * TBK];
+  const int n_tiles_n = N / TBN;
   const int m0 = (blockIdx.x / n_tiles_n) * 128;
-  const int n0 = (blockIdx.x % n_tiles_n) * 128;
+  const int n0 = (blockIdx.x % n_tiles_n) * TBN;
   const int tid = threadIdx.x;
   const int lane = tid % 64;
   const int wave = tid / 64;
@@ -360,16 +363,20 @@ __global__ __launch_bounds__(256) void gemm_nt_glds_kernel(
       __builtin_amdgcn_global_load_lds(
           (const uint32_t*)(A + (int64_t)a_row * K + k0 + s_gr * 8),
           (uint32_t*)&lds_a[bufi][(wave * 32 + i * 8) * TBK], 16, 0, 0);
+    }
+#pragma unroll
+    for (int i = 0; i < BCH; ++i) {
+      const int brow = wave * (TBN / 4) + i * 8 + s_rl;
       __builtin_amdgcn_global_load_lds(
-          (const uint32_t*)(B + (int64_t)(n0 + row) * K + k0 + s_gr * 8),
-          (uint32_t*)&lds_b[bufi][(wave * 32 + i * 8) * TBK], 16, 0, 0);
+          (const uint32_t*)(B + (int64_t)(n0 + brow) * K + k0 + s_gr * 8),
+          (uint32_t*)&lds_b[bufi][(wave * (TBN / 4) + i * 8) * TBK], 16, 0, 0);
     }
   };
   auto frag = [&](const short* base, int row, int gr) {
     return *(const bf16x8*)&base[row * TBK + ((gr ^ (row & 7)) << 3)];
   };
 
-  f32x4 acc[4][4] = {};
+  f32x4 acc[4][NFRAG] = {};
   stage(0, 0);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
@@ -378,16 +385,16 @@ __global__ __launch_bounds__(256) void gemm_nt_glds_kernel(
     if (k0 + TBK < K) stage(buf ^ 1, k0 + TBK);  // DMA hides under MFMA
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      bf16x8 bfr[4];
+      bf16x8 bfr[NFRAG];
 #pragma unroll
-      for (int bn = 0; bn < 4; ++bn)
-        bfr[bn] = frag(lds_b[buf], wc * 64 + bn * 16 + fi, kk * 4 + g16);
+      for (int bn = 0; bn < NFRAG; ++bn)
+        bfr[bn] = frag(lds_b[buf], wc * WN + bn * 16 + fi, kk * 4 + g16);
 #pragma unroll
       for (int am = 0; am < 4; ++am) {
         const bf16x8 afr = frag(lds_a[buf], wr * 64 + am * 16 + fi,
                                 kk * 4 + g16);
 #pragma unroll
-        for (int bn = 0; bn < 4; ++bn)
+        for (int bn = 0; bn < NFRAG; ++bn)
           acc[am][bn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afr, bfr[bn], acc[am][bn], 0, 0, 0);
       }
@@ -402,8 +409,8 @@ __global__ __launch_bounds__(256) void gemm_nt_glds_kernel(
 #pragma unroll
   for (int am = 0; am < 4; ++am) {
 #pragma unroll
-    for (int bn = 0; bn < 4; ++bn) {
-      const int col = n0 + wc * 64 + bn * 16 + fi;
+    for (int bn = 0; bn < NFRAG; ++bn) {
+      const int col = n0 + wc * WN + bn * 16 + fi;
       const float bval = bias ? bias[col] : 0.0f;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -687,17 +694,30 @@ torch::Tensor gemm_nt_bias_act(torch::Tensor A, torch::Tensor B,
   // shape cutover (measured sweep, tools/gemm_v2_sweep.py): v2 wins when the
   // 128x128 grid fills the chip (>=400 blocks) and K amortizes the pipeline
   // prologue; below that the v1 TBN=64 configs keep more CUs busy
-  if (v2_on && !trans_b && K % 64 == 0 && N % 128 == 0 && M >= 128 &&
-      K >= 256 && (int64_t)m_tiles * (N / 128) >= 400) {
-    const int grid = m_tiles * (N / 128);
+  const bool v2_shape = v2_on && !trans_b && K % 64 == 0 && M >= 128 &&
+                        K >= 256;
+  const bool v2_128 = v2_shape && N % 128 == 0 &&
+                      (int64_t)m_tiles * (N / 128) >= 400;
+  // TBN=64 variant: mid-width layers whose 128-wide grid underfills
+  const bool v2_64 = v2_shape && !v2_128 && N % 64 == 0 &&
+                     (int64_t)m_tiles * (N / 64) >= 400;
+  if (v2_128 || v2_64) {
+    const int tbn = v2_128 ? 128 : 64;
+    const int grid = m_tiles * (N / tbn);
     const float* bias_p = bias.numel() ? bias.data_ptr<float>() : nullptr;
-#define PA_GEMM2(ACTV, F32V)                                                  \
-  hipLaunchKernelGGL((gemm_nt_glds_kernel<ACTV, F32V>), dim3(grid),           \
+#define PA_GEMM2(ACTV, F32V, TBNV)                                            \
+  hipLaunchKernelGGL((gemm_nt_glds_kernel<ACTV, F32V, TBNV>), dim3(grid),     \
                      dim3(256), 0, dcur_stream(),                             \
                      (const bf16*)A.data_ptr(), (const bf16*)B.data_ptr(),    \
                      bias_p, C.data_ptr(), M, N, K)
-    if (act == 1) { if (out_f32) PA_GEMM2(1, true); else PA_GEMM2(1, false); }
-    else          { if (out_f32) PA_GEMM2(0, true); else PA_GEMM2(0, false); }
+#define PA_GEMM2_T(ACTV, F32V)                                                \
+  do {                                                                        \
+    if (v2_128) PA_GEMM2(ACTV, F32V, 128);                                    \
+    else PA_GEMM2(ACTV, F32V, 64);                                            \
+  } while (0)
+    if (act == 1) { if (out_f32) PA_GEMM2_T(1, true); else PA_GEMM2_T(1, false); }
+    else          { if (out_f32) PA_GEMM2_T(0, true); else PA_GEMM2_T(0, false); }
+#undef PA_GEMM2_T
 #undef PA_GEMM2
     return C;
   }
