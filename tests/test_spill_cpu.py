@@ -102,3 +102,83 @@ def test_spill_capacity_bounded():
     for s in range(0, 2000, 8):
         store.lookup(_keys(list(range(s, s + 8))), train=True)
     assert len(store.spill) <= 16
+
+
+def _mk_spill_engine(ctx):
+    from persia_amd.core.engine import EmbeddingEngine
+    from persia_amd.core.schema import EmbeddingSchema, GlobalConfig, SlotConfig
+
+    # tiny capacity forces evictions into the spill tier
+    return EmbeddingEngine(
+        schema=EmbeddingSchema(slots={"a": SlotConfig(name="a", dim=8)}),
+        hyper=EmbeddingConfig(emb_initialization=(-0.5, 0.5)),
+        optimizer=Adagrad(lr=0.1),
+        gconf=GlobalConfig(capacity=64, spill_capacity=4096),
+        device=torch.device("cpu"),
+        dist_ctx=ctx,
+        wire_dtype=torch.float32,
+    )
+
+
+def _spill_batches():
+    from persia_amd.embedding.data import IDTypeFeature, Label, PersiaBatch
+
+    rng = np.random.default_rng(7)
+    out = []
+    for _step in range(4):
+        feats = [IDTypeFeature("a", [
+            rng.integers(0, 2000, size=4, dtype=np.uint64) for _ in range(8)
+        ])]
+        out.append(PersiaBatch(
+            feats, labels=[Label(np.ones((8, 1), np.float32))],
+            requires_grad=True,
+        ))
+    return out
+
+
+def _spill_worker(rank, port, result_dir):
+    import os
+
+    import torch.distributed as dist
+
+    from persia_amd.core.comm import DistContext
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=2)
+    eng = _mk_spill_engine(DistContext.new_sparse_group())
+    outs = [
+        [p.sum_tensor.clone() for p in eng.process_batch(b).payloads]
+        for b in _spill_batches()
+    ]
+    if rank == 0:
+        import os.path
+
+        torch.save(outs, os.path.join(result_dir, "spill_w2.pt"))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_spill_with_world2_generic_path(tmp_path):
+    """Spill + distributed falls back to the exact two-phase exchange (the
+    padded fast path requires spill=None): a 2-rank run with a spill tier
+    must still produce the world-1 lookup results bitwise (f32 wire)."""
+    import torch.multiprocessing as mp
+
+    from persia_amd.core.comm import DistContext
+    from persia_amd.utils import find_free_port
+
+    ref_eng = _mk_spill_engine(DistContext(1, 0))
+    ref = [
+        [p.sum_tensor.clone() for p in ref_eng.process_batch(b).payloads]
+        for b in _spill_batches()
+    ]
+    assert any(len(s.spill) > 0 for s in ref_eng.stores.values()), (
+        "test setup must actually spill"
+    )
+    port = find_free_port()
+    mp.spawn(_spill_worker, args=(port, str(tmp_path)), nprocs=2, join=True)
+    got = torch.load(tmp_path / "spill_w2.pt")
+    for step, (r, g) in enumerate(zip(ref, got)):
+        for pr, pg in zip(r, g):
+            assert torch.equal(pr, pg), f"step {step}"
