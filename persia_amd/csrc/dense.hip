@@ -428,6 +428,109 @@ __global__ __launch_bounds__(256) void gemm_nt_glds_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// NT GEMM v3: 3-buffer glds ring with COUNTED vmcnt — one K-tile stays in
+// flight ACROSS each barrier (guide §2 "3-buf span": +83% over serial vs
+// +40% for the 2-buf drain; the v2 kernel above waits vmcnt(0) per tile).
+// 96 KB LDS -> 1 block/CU, the regime where glds pipelining pays.
+// 8 glds per wave per tile => s_waitcnt vmcnt(8) keeps exactly the newest
+// tile's loads outstanding.
+template <int ACT, bool STORE_F32>
+__global__ __launch_bounds__(256) void gemm_nt_glds3_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ B,
+    const float* __restrict__ bias, void* __restrict__ C, int M, int N,
+    int K) {
+  constexpr int TBK = 64;
+  __shared__ short lds_a[3][128 * TBK];
+  __shared__ short lds_b[3][128 * TBK];
+  const int n_tiles_n = N / 128;
+  const int m0 = (blockIdx.x / n_tiles_n) * 128;
+  const int n0 = (blockIdx.x % n_tiles_n) * 128;
+  const int tid = threadIdx.x;
+  const int lane = tid % 64;
+  const int wave = tid / 64;
+  const int wr = wave / 2, wc = wave % 2;
+  const int fi = lane & 15;
+  const int g16 = lane >> 4;
+  const int s_rl = lane >> 3;
+  const int s_gr = (lane & 7) ^ s_rl;
+  auto stage = [&](int bufi, int k0) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int row = wave * 32 + i * 8 + s_rl;
+      const int a_row = min(m0 + row, M - 1);
+      __builtin_amdgcn_global_load_lds(
+          (const uint32_t*)(A + (int64_t)a_row * K + k0 + s_gr * 8),
+          (uint32_t*)&lds_a[bufi][(wave * 32 + i * 8) * TBK], 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const uint32_t*)(B + (int64_t)(n0 + row) * K + k0 + s_gr * 8),
+          (uint32_t*)&lds_b[bufi][(wave * 32 + i * 8) * TBK], 16, 0, 0);
+    }
+  };
+  auto frag = [&](const short* base, int row, int gr) {
+    return *(const bf16x8*)&base[row * TBK + ((gr ^ (row & 7)) << 3)];
+  };
+
+  f32x4 acc[4][4] = {};
+  const int nt = K / TBK;
+  stage(0, 0);
+  if (nt > 1) {
+    stage(1, TBK);
+    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");  // tile 0 landed
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+  __builtin_amdgcn_s_barrier();
+  for (int t = 0; t < nt; ++t) {
+    if (t + 2 < nt) stage((t + 2) % 3, (t + 2) * TBK);
+    const short* la = lds_a[t % 3];
+    const short* lb = lds_b[t % 3];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 bfr[4];
+#pragma unroll
+      for (int bn = 0; bn < 4; ++bn)
+        bfr[bn] = frag(lb, wc * 64 + bn * 16 + fi, kk * 4 + g16);
+#pragma unroll
+      for (int am = 0; am < 4; ++am) {
+        const bf16x8 afr = frag(la, wr * 64 + am * 16 + fi, kk * 4 + g16);
+#pragma unroll
+        for (int bn = 0; bn < 4; ++bn)
+          acc[am][bn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr, bfr[bn], acc[am][bn], 0, 0, 0);
+      }
+    }
+    // keep the NEWEST tile's 8 glds in flight across the barrier; the next
+    // iteration's buffer (t+1) is guaranteed landed.  Near the tail nothing
+    // newer is in flight, so drain fully before the final tiles.
+    if (t + 2 < nt)
+      asm volatile("s_waitcnt vmcnt(8) lgkmcnt(0)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+#pragma unroll
+  for (int am = 0; am < 4; ++am) {
+#pragma unroll
+    for (int bn = 0; bn < 4; ++bn) {
+      const int col = n0 + wc * 64 + bn * 16 + fi;
+      const float bval = bias ? bias[col] : 0.0f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = m0 + wr * 64 + am * 16 + (lane >> 4) * 4 + r;
+        if (row >= M) continue;
+        float v = acc[am][bn][r] + bval;
+        if (ACT == 1) v = fmaxf(v, 0.0f);
+        if (STORE_F32)
+          ((float*)C)[(int64_t)row * N + col] = v;
+        else
+          ((short*)C)[(int64_t)row * N + col] = f2bf(v);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // wgrad: dW[N,K] += sum_m dC[m][n] * A[m][k]   (dC [M,N], A [M,K] bf16;
 // dW f32, torch Linear weight layout).  MFMA with the reduction on the
 // memory-row axis: both tiles are staged TRANSPOSED ([n][m] / [k][m],
@@ -701,6 +804,23 @@ torch::Tensor gemm_nt_bias_act(torch::Tensor A, torch::Tensor B,
   // TBN=64 variant: mid-width layers whose 128-wide grid underfills
   const bool v2_64 = v2_shape && !v2_128 && N % 64 == 0 &&
                      (int64_t)m_tiles * (N / 64) >= 400;
+  static const bool v3_on = [] {
+    const char* e = getenv("PA_GEMM_V3");
+    return e && atoi(e) != 0;
+  }();
+  if (v3_on && v2_128) {
+    const int grid = m_tiles * (N / 128);
+    const float* bias_p = bias.numel() ? bias.data_ptr<float>() : nullptr;
+#define PA_GEMM3(ACTV, F32V)                                                  \
+  hipLaunchKernelGGL((gemm_nt_glds3_kernel<ACTV, F32V>), dim3(grid),          \
+                     dim3(256), 0, dcur_stream(),                             \
+                     (const bf16*)A.data_ptr(), (const bf16*)B.data_ptr(),    \
+                     bias_p, C.data_ptr(), M, N, K)
+    if (act == 1) { if (out_f32) PA_GEMM3(1, true); else PA_GEMM3(1, false); }
+    else          { if (out_f32) PA_GEMM3(0, true); else PA_GEMM3(0, false); }
+#undef PA_GEMM3
+    return C;
+  }
   if (v2_128 || v2_64) {
     const int tbn = v2_128 ? 128 : 64;
     const int grid = m_tiles * (N / tbn);
