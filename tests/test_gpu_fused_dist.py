@@ -123,7 +123,7 @@ def test_a2a_route_kernel_first_principles():
     rng = np.random.default_rng(3)
     world, cap = 8, 700
     n_valid, n_pad = 5000, 6000
-    keys = np.sort(rng.integers(1, 2**63, size=n_valid, dtype=np.uint64))
+    keys = np.sort(rng.integers(1, 2**64, size=n_valid, dtype=np.uint64))
     uniq = np.zeros(n_pad, dtype=np.uint64)
     uniq[:n_valid] = keys
     uniq_t = torch.from_numpy(uniq.view(np.int64)).to(dev)
