@@ -200,7 +200,10 @@ def main():
     if world > 1 and not defer_ddp:
         model = _wrap_ddp(model)
     opt = torch.optim.SGD(model.parameters(), lr=0.01)
-    loss_fn = torch.nn.functional.binary_cross_entropy_with_logits
+    if use_gpu:
+        from persia_amd.ops.dense import fused_bce_with_logits as loss_fn
+    else:
+        loss_fn = torch.nn.functional.binary_cross_entropy_with_logits
 
     # ---- synthetic data: pre-generate host batches (excluded from timing)
     n_batches = args.steps + args.warmup

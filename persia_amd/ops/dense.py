@@ -171,3 +171,35 @@ class FusedMLP(nn.Module):
         for l in self.layers:
             x = l(x)
         return x
+
+
+class FusedBCEFn(torch.autograd.Function):
+    """BCE-with-logits as 2 kernels (fwd: loss + sigmoid cache, bwd: one
+    elementwise) — torch's chain is ~10 launch-bound kernels inside the
+    captured step."""
+
+    @staticmethod
+    def forward(ctx, logits, label):
+        from persia_amd.ops import native
+
+        C = native()
+        loss, sig = C.bce_fwd(logits.float().contiguous(),
+                              label.float().contiguous())
+        ctx.save_for_backward(sig, label)
+        ctx.in_dtype = logits.dtype
+        return loss
+
+    @staticmethod
+    def backward(ctx, g):
+        from persia_amd.ops import native
+
+        C = native()
+        sig, label = ctx.saved_tensors
+        dz = C.bce_bwd(sig, label.float(), g.reshape(1).float().contiguous())
+        if dz.dtype != ctx.in_dtype:
+            dz = dz.to(ctx.in_dtype)
+        return dz, None
+
+
+def fused_bce_with_logits(logits: torch.Tensor, label: torch.Tensor) -> torch.Tensor:
+    return FusedBCEFn.apply(logits, label)

@@ -316,3 +316,19 @@ def test_sideband_grad_accumulation_matches_standard():
     out3.backward(g)
     assert torch.allclose(wslot, 2 * dw_std, atol=1.0, rtol=0.05)
     del layer.weight._sideband_grad, layer.bias._sideband_grad
+
+
+def test_fused_bce_matches_torch():
+    from persia_amd.ops.dense import fused_bce_with_logits
+
+    torch.manual_seed(13)
+    z = torch.randn(8192, device=_dev(), dtype=torch.bfloat16, requires_grad=True)
+    y = (torch.rand(8192, device=_dev()) > 0.5).float()
+    loss = fused_bce_with_logits(z, y)
+    loss.backward()
+    z2 = z.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.binary_cross_entropy_with_logits(z2, y)
+    ref.backward()
+    assert torch.allclose(loss, ref, atol=1e-4, rtol=1e-4)
+    assert torch.allclose(z.grad.float(), z2.grad.to(torch.bfloat16).float(),
+                          atol=1e-6)
