@@ -1056,7 +1056,9 @@ std::vector<torch::Tensor> bce_fwd(torch::Tensor z, torch::Tensor y) {
                      y.data_ptr<float>(), sig.data_ptr<float>(),
                      loss.data_ptr<float>(), n);
   loss.div_((double)n);
-  return {loss.view({}), sig};
+  // NOT loss.view({}): the empty brace list resolves to the
+  // view(ScalarType) overload (uint8 reinterpret), not a 0-dim view
+  return {loss.squeeze(0), sig};
 }
 
 torch::Tensor bce_bwd(torch::Tensor sig, torch::Tensor y, torch::Tensor g) {
