@@ -368,6 +368,12 @@ class EmbeddingEngine:
 
             self.monitor = DistinctIdMonitor()
         self.incremental = None
+        if gconf.enable_incremental_update:
+            # config-driven (reference inc-update lib.rs:79-120: the train
+            # side starts the manager from the global config)
+            self.enable_incremental_update(
+                gconf.incremental_dir, gconf.incremental_buffer_size
+            )
 
     def enable_incremental_update(self, dst_dir: str,
                                   buffer_size: int = 1_000_000):

@@ -163,6 +163,11 @@ class GlobalConfig:
     checkpointing_workers: int = 4
     staleness: int = 8
     reproducible: bool = False
+    # online-inference freshness stream (reference config lib.rs:417-447:
+    # enable_incremental_update / incremental_buffer_size / incremental_dir)
+    enable_incremental_update: bool = False
+    incremental_buffer_size: int = 1_000_000
+    incremental_dir: str = "/workspace/inc_dir" 
 
     @staticmethod
     def from_yaml(path: str) -> "GlobalConfig":
@@ -178,5 +183,18 @@ class GlobalConfig:
             ),
             checkpointing_workers=int(
                 common.get("checkpointing_config", {}).get("num_workers", 4)
+            ),
+            spill_capacity=int(server.get("spill_capacity", 0)),
+            staleness=int(common.get("embedding_staleness",
+                                     common.get("staleness", 8))),
+            reproducible=bool(common.get("reproducible", False)),
+            enable_incremental_update=bool(
+                server.get("enable_incremental_update", False)
+            ),
+            incremental_buffer_size=int(
+                server.get("incremental_buffer_size", 1_000_000)
+            ),
+            incremental_dir=str(
+                server.get("incremental_dir", "/workspace/inc_dir")
             ),
         )

@@ -138,3 +138,38 @@ def test_incremental_packets_appear_automatically(metrics_engine, tmp_path):
     n = loader.scan_once()
     assert n >= 100
     assert eng2.num_resident_rows() >= 100
+
+
+def test_incremental_enabled_from_global_config(tmp_path):
+    """yaml -> GlobalConfig -> engine auto-attaches the incremental dumper
+    (reference: enable_incremental_update in the global config)."""
+    import yaml
+
+    from persia_amd.core.comm import DistContext
+    from persia_amd.core.engine import EmbeddingEngine
+    from persia_amd.core.schema import EmbeddingSchema, GlobalConfig, SlotConfig
+    from persia_amd.embedding import EmbeddingConfig
+    from persia_amd.embedding.optim import SGD
+
+    cfg = {
+        "common_config": {"job_type": "train", "embedding_staleness": 4},
+        "embedding_parameter_server_config": {
+            "capacity": 4096,
+            "enable_incremental_update": True,
+            "incremental_buffer_size": 77,
+            "incremental_dir": str(tmp_path / "inc"),
+        },
+    }
+    path = tmp_path / "global.yml"
+    path.write_text(yaml.safe_dump(cfg), encoding="utf-8")
+    g = GlobalConfig.from_yaml(str(path))
+    assert g.capacity == 4096 and g.staleness == 4
+    assert g.enable_incremental_update and g.incremental_buffer_size == 77
+    eng = EmbeddingEngine(
+        schema=EmbeddingSchema(slots={"a": SlotConfig(name="a", dim=8)}),
+        hyper=EmbeddingConfig(), optimizer=SGD(lr=0.1), gconf=g,
+        device=torch.device("cpu"), dist_ctx=DistContext(1, 0),
+    )
+    assert eng.incremental is not None
+    assert eng.incremental.buffer_size == 77
+    assert eng.incremental.dst_dir == str(tmp_path / "inc")
