@@ -174,6 +174,11 @@ class IncrementalUpdateLoader:
             try:
                 ts = int(name.split("_", 1)[1]) / 1000.0
                 self.last_delay_sec = max(0.0, time.time() - ts)
+                if self.engine.metrics_enabled:
+                    # reference inc-update lib.rs:48-65 gauge
+                    self.engine.metrics.inc_update_delay_sec.set(
+                        self.last_delay_sec
+                    )
             except ValueError:
                 pass
         return loaded

@@ -170,6 +170,10 @@ class EngineMetrics:
             "distinct_id_estimate", "HLL distinct-sign estimate per feature"
         )
         self.inc_packets_dumped = m.counter("inc_packets_dumped")
+        self.inc_update_delay_sec = m.gauge(
+            "inc_update_delay_sec",
+            "age of the newest applied incremental packet (infer side)",
+        )
 
     def sample_values(self) -> Dict[str, float]:
         """Current values of the unlabeled series (tests/debug)."""
