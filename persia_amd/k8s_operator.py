@@ -151,6 +151,28 @@ class K8sApi:
         if r.status_code != 404:
             r.raise_for_status()
 
+    def list_workloads(self, kind: str) -> List[dict]:
+        r = self.s.get(self._wl_base(kind))
+        r.raise_for_status()
+        return r.json().get("items", [])
+
+    # ---- pods (read-only; used by the schedule server) ----
+    def _pod_base(self) -> str:
+        return f"{self.base}/api/v1/namespaces/{self.ns}/pods"
+
+    def list_pods(self, label_selector: Optional[str] = None) -> List[dict]:
+        params = {"labelSelector": label_selector} if label_selector else None
+        r = self.s.get(self._pod_base(), params=params)
+        r.raise_for_status()
+        return r.json().get("items", [])
+
+    def get_pod(self, name: str) -> Optional[dict]:
+        r = self.s.get(f"{self._pod_base()}/{name}")
+        if r.status_code == 404:
+            return None
+        r.raise_for_status()
+        return r.json()
+
 
 def _owned_manifests(job: dict) -> List[dict]:
     """Render the child workloads for a PersiaJob CR and stamp ownership
@@ -251,11 +273,32 @@ def main(argv: Optional[List[str]] = None):
     op.add_argument("--namespace", default=None)
     op.add_argument("--token", default=None)
     op.add_argument("--period-sec", type=float, default=2.0)
+    sv = sub.add_parser(
+        "server", help="HTTP schedule server (apply/delete/list/podstatus)"
+    )
+    sv.add_argument("--port", type=int, required=True)
+    sv.add_argument("--api-server", default=None,
+                    help="k8s API base URL (default: in-cluster config)")
+    sv.add_argument("--token", default=None)
     args = p.parse_args(argv)
     if args.cmd == "gencrd":
         import yaml
 
         print(yaml.safe_dump(CRD))
+        return
+    if args.cmd == "server":
+        from persia_amd.k8s_server import serve
+
+        if args.api_server:
+            serve(args.port, args.api_server, args.token)
+        else:
+            sa = "/var/run/secrets/kubernetes.io/serviceaccount"
+            with open(f"{sa}/token", encoding="utf-8") as f:
+                token = f.read()
+            host = os.environ["KUBERNETES_SERVICE_HOST"]
+            kport = os.environ.get("KUBERNETES_SERVICE_PORT", "443")
+            serve(args.port, f"https://{host}:{kport}", token,
+                  verify=f"{sa}/ca.crt")
         return
     if args.api_server:
         api = K8sApi(args.api_server, args.namespace or "default", args.token)

@@ -144,5 +144,21 @@ def operator(api_server, namespace, token, period_sec):
     op_main(argv)
 
 
+@cli.command("server")
+@click.option("--port", required=True, type=int)
+@click.option("--api-server", default=None, help="k8s API base URL (default: in-cluster)")
+@click.option("--token", default=None)
+def server(port, api_server, token):
+    """Run the HTTP schedule server (reference k8s/src/bin/server.rs)."""
+    from persia_amd.k8s_operator import main as op_main
+
+    argv = ["server", f"--port={port}"]
+    if api_server:
+        argv.append(f"--api-server={api_server}")
+    if token:
+        argv.append(f"--token={token}")
+    op_main(argv)
+
+
 if __name__ == "__main__":
     cli()

@@ -23,6 +23,7 @@ class FakeK8s:
     def __init__(self):
         self.crs = {}        # name -> PersiaJob dict
         self.workloads = {}  # (kind, name) -> manifest dict
+        self.pods = {}       # name -> pod dict (populated by tests)
         self.lock = threading.Lock()
 
     def handler(self):
@@ -45,12 +46,34 @@ class FakeK8s:
                 return json.loads(self.rfile.read(n)) if n else {}
 
             def do_GET(self):
-                path = self.path.split("?")[0]
+                from urllib.parse import parse_qs, unquote, urlparse
+
+                u = urlparse(self.path)
+                path, qs = u.path, parse_qs(u.query)
                 with store.lock:
                     if path.endswith("/persiajobs"):
                         return self._send(200, {"items": list(store.crs.values())})
-                    for kind, seg in (("Job", "/jobs/"), ("Deployment", "/deployments/")):
-                        if seg in path:
+                    if path.endswith("/pods"):
+                        items = list(store.pods.values())
+                        sel = unquote(qs.get("labelSelector", [""])[0])
+                        if sel:
+                            k, v = sel.split("=", 1)
+                            items = [
+                                p for p in items
+                                if (p["metadata"].get("labels") or {}).get(k) == v
+                            ]
+                        return self._send(200, {"items": items})
+                    if "/pods/" in path:
+                        pod = store.pods.get(path.rsplit("/", 1)[1])
+                        return self._send(200, pod) if pod else self._send(404)
+                    for kind, seg in (("Job", "/jobs"), ("Deployment", "/deployments")):
+                        if path.endswith(seg):
+                            items = [
+                                m for (k, _), m in store.workloads.items()
+                                if k == kind
+                            ]
+                            return self._send(200, {"items": items})
+                        if seg + "/" in path:
                             name = path.rsplit("/", 1)[1]
                             wl = store.workloads.get((kind, name))
                             return self._send(200, wl) if wl else self._send(404)
