@@ -483,6 +483,23 @@ def main():
     timing = os.environ.get("PA_BENCH_TIMING", "0") == "1"
     tstats = {"get": 0.0, "copy": 0.0, "replay": 0.0, "apply": 0.0, "n": 0}
 
+    # Overlapped sparse update (PA_OVERLAP_UPDATE=1 enables; default OFF —
+    # measured NEGATIVE): the fused table update runs on a dedicated stream
+    # from a STAGED copy of the base gradient so the main stream serializes
+    # only on the stage copy.  A/B on a fresh box: criteo 6.71M overlapped
+    # vs 6.80M serial (the step is bound by the LOOKUP stream, so hiding
+    # main-stream work buys nothing) and dcn-spill 1.73M vs ~3.4M (the
+    # side-stream update steals HBM bandwidth from the sparse-critical
+    # producer).  Kept as a gated experiment; ordering stays safe (updates
+    # serialized on one stream; eviction-vs-update races are the same
+    # bounded-staleness noise as the base design, csrc/kernels.hip notes).
+    upd_stream = None
+    if graph is not None and os.environ.get("PA_OVERLAP_UPDATE", "0") == "1":
+        upd_stream = torch.cuda.Stream()
+        upd_gstage = torch.zeros_like(static["base"].grad)
+        ev_bwd = torch.cuda.Event()
+        ev_staged = torch.cuda.Event()
+
     def train_step(tb):
         if graph is not None:
             if timing:
@@ -494,12 +511,27 @@ def main():
             if timing:
                 t1 = time.perf_counter()
             graph.replay()
+            if upd_stream is not None:
+                ev_bwd.record()
+                upd_stream.wait_event(ev_bwd)
+                with torch.cuda.stream(upd_stream):
+                    upd_gstage.copy_(static["base"].grad, non_blocking=True)
+                    ev_staged.record(upd_stream)
+                    engine.apply_gradients_base(tb, sum_base_grads=[upd_gstage])
+                    tb.record_stream(upd_stream)
             if graph_upd is not None:
                 allreduce_grads()
                 graph_upd.replay()
             if timing:
                 t2 = time.perf_counter()
-            engine.apply_gradients_base(tb, sum_base_grads=[static["base"].grad])
+            if upd_stream is not None:
+                # gate the next replay (which zeroes base.grad) on the stage
+                # copy only — the table update continues on upd_stream
+                torch.cuda.current_stream().wait_event(ev_staged)
+            else:
+                engine.apply_gradients_base(
+                    tb, sum_base_grads=[static["base"].grad]
+                )
             pipeline.release_permit()
             if timing:
                 t3 = time.perf_counter()
