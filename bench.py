@@ -271,8 +271,17 @@ def main():
     static = {}
     if args.graph and use_gpu and (world == 1 or defer_ddp):
         try:
+            # pre-pad + pre-cast the static dense input for the fused DLRM:
+            # the fused layers zero-pad K to %128 per step anyway (inert zero
+            # columns hit zero-padded weight columns), so a statically padded
+            # bf16 buffer kills that pad pass and the per-step f32->bf16 cast
+            nd = args.num_dense
+            d_dtype = torch.float32
+            if args.model == "dlrm" and bool(args.fused_dense) and bf16_weights:
+                nd = (args.num_dense + 127) // 128 * 128
+                d_dtype = torch.bfloat16
             static = {
-                "dense": torch.zeros(B, args.num_dense, device=device),
+                "dense": torch.zeros(B, nd, dtype=d_dtype, device=device),
                 "base": torch.zeros(
                     n_slots * B, dim, dtype=torch.float16, device=device,
                     requires_grad=True,
@@ -505,7 +514,9 @@ def main():
             if timing:
                 t0 = time.perf_counter()
             with torch.no_grad():
-                static["dense"].copy_(tb.non_id_type_tensors[0], non_blocking=True)
+                static["dense"][:, : args.num_dense].copy_(
+                    tb.non_id_type_tensors[0], non_blocking=True
+                )
                 static["base"].copy_(tb._groups[0].sum_base, non_blocking=True)
                 static["label"].copy_(tb.label_tensors[0], non_blocking=True)
             if timing:

@@ -174,7 +174,17 @@ class DLRM(nn.Module):
                     self.num_sparse + 1, self.dim
                 ):
                     inter = _InteractPackedFn.apply(x, embedding_tensors)
-                    return self.top(torch.cat([x, inter], dim=1)).squeeze(1)
+                    # build the top input PRE-padded to the fused GEMM's
+                    # K%128 by catting the zero tail here — one cat pass
+                    # instead of cat + a full-width _pad32 copy inside the
+                    # first top layer (zero K-columns hit zero-padded weight
+                    # columns: bit-identical output)
+                    k = x.shape[1] + inter.shape[1]
+                    pad = (-k) % 128
+                    parts = [x, inter]
+                    if pad:
+                        parts.append(x.new_zeros(B, pad))
+                    return self.top(torch.cat(parts, dim=1)).squeeze(1)
             # packed slot-major [S*B, D] (the engine's fused sum output) —
             # ONE reshape instead of a 26-way stack
             emb = (

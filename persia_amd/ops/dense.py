@@ -48,7 +48,8 @@ class FusedLinearFn(torch.autograd.Function):
         out = C.gemm_nt_bias_act(x_bf, w_bf, bias.float(), act, 0, 0)
         ctx.save_for_backward(x_bf, w_bf, out)
         ctx.act = act
-        ctx.k = x.shape[1]
+        ctx.k = x.shape[1]       # dx width (input may arrive pre-padded)
+        ctx.kw = weight.shape[1]  # dw width (the stored, unpadded weight)
         ctx.w_dtype = weight.dtype
         ctx.b_dtype = bias.dtype
         # side-band gradient targets (set by the bench's flat-buffer scheme):
@@ -110,8 +111,8 @@ class FusedLinearFn(torch.autograd.Function):
             dw = C.wgrad(g, x_bf)
             if dw.dtype != ctx.w_dtype:
                 dw = dw.to(ctx.w_dtype)
-        if dw is not None and dw.shape[1] != ctx.k:
-            dw = dw[:, : ctx.k].contiguous()
+        if dw is not None and dw.shape[1] != ctx.kw:
+            dw = dw[:, : ctx.kw].contiguous()
         return dx, dw, db, None
 
 
