@@ -377,6 +377,10 @@ class TrainCtx(EmbeddingCtx):
         else:
             self.dense_optimizer.step()
         self.dense_optimizer.zero_grad()
+        if self.distributed_option is not None:
+            # periodic model averaging for the Bagua-async mapping (no-op
+            # for synchronous options)
+            self.distributed_option.post_optimizer_step(self.model)
         return loss
 
     def _on_backward(self, loss_scale: float, check_frequency: int) -> bool:

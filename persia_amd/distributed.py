@@ -30,6 +30,10 @@ class DistributedBaseOption(ABC):
     def wrap_model(self, model: torch.nn.Module, device_id: Optional[int]) -> torch.nn.Module:
         raise NotImplementedError
 
+    def post_optimizer_step(self, model: torch.nn.Module) -> None:
+        """Called by TrainCtx.backward after each dense optimizer step
+        (no-op for synchronous algorithms)."""
+
 
 class DDPOption(DistributedBaseOption):
     """torch DDP over RCCL (reference persia/distributed.py:74-193).
@@ -104,13 +108,18 @@ class BaguaDistributedOption(DistributedBaseOption):
       topologies are pointless on a fully-connected xGMI node)
     * ``decentralized``           -> DDP with a wider bucket (peer averaging
       on an 8-GPU all-to-all mesh degenerates to allreduce)
-    * ``async``                   -> not supported (raises): the sparse side
-      is already asynchronous; async dense averaging trades determinism for
-      nothing on one node
+    * ``async``                   -> periodic (local-SGD style) model
+      averaging: NO per-step gradient sync — each rank trains locally and
+      every ``sync_every_steps`` optimizer steps the weights are averaged
+      with one allreduce (the collective-safe mapping of Bagua's async
+      model-average algorithm — its lock-free peer pulls cannot be expressed
+      with collectives, but the statistical behaviour, bounded weight drift
+      between periodic merges, is the same).  ``sync_interval_ms`` is
+      accepted for reference API parity and treated as a step interval.
     """
 
     _COMPRESSED = {"bytegrad", "qadam", "low_precision_decentralized"}
-    _SUPPORTED = _COMPRESSED | {"gradient_allreduce", "decentralized"}
+    _SUPPORTED = _COMPRESSED | {"gradient_allreduce", "decentralized", "async"}
 
     def __init__(self, algorithm: str = "gradient_allreduce", **options):
         super().__init__()
@@ -120,6 +129,11 @@ class BaguaDistributedOption(DistributedBaseOption):
                 f"Bagua algorithm {algorithm!r} has no MI355X mapping "
                 f"(supported: {sorted(self._SUPPORTED)})"
             )
+        self.sync_every_steps = int(
+            options.pop("sync_every_steps",
+                        options.pop("sync_interval_ms", 8) or 8)
+        )
+        self._step = 0
         bucket = 100 if algorithm == "decentralized" else 50
         self._ddp = DDPOption(bucket_cap_mb=bucket, **options)
 
@@ -127,6 +141,12 @@ class BaguaDistributedOption(DistributedBaseOption):
         self._ddp.init_process_group(device_id)
 
     def wrap_model(self, model, device_id):
+        if self.algorithm == "async":
+            # local training: no DDP wrap, but start from identical weights
+            with torch.no_grad():
+                for t in list(model.parameters()) + list(model.buffers()):
+                    dist.broadcast(t.data, src=0)
+            return model
         wrapped = self._ddp.wrap_model(model, device_id)
         if self.algorithm in self._COMPRESSED:
             from torch.distributed.algorithms.ddp_comm_hooks import (
@@ -137,6 +157,18 @@ class BaguaDistributedOption(DistributedBaseOption):
                 state=None, hook=default_hooks.bf16_compress_hook
             )
         return wrapped
+
+    def post_optimizer_step(self, model):
+        if self.algorithm != "async" or not dist.is_initialized():
+            return
+        self._step += 1
+        if self._step % self.sync_every_steps:
+            return
+        world = dist.get_world_size()
+        with torch.no_grad():
+            for p in model.parameters():
+                dist.all_reduce(p.data)
+                p.data.div_(world)
 
 
 def get_default_distributed_option(device_id: Optional[int] = None) -> DDPOption:

@@ -55,11 +55,12 @@ def test_persia_path_disk(tmp_path):
 
 def test_bagua_option_mappings():
     for algo in ("gradient_allreduce", "bytegrad", "qadam",
-                 "low_precision_decentralized", "decentralized"):
+                 "low_precision_decentralized", "decentralized", "async"):
         opt = BaguaDistributedOption(algorithm=algo)
         assert isinstance(opt._ddp, DDPOption)
-    with pytest.raises(NotImplementedError):
-        BaguaDistributedOption(algorithm="async")
+    assert BaguaDistributedOption(
+        algorithm="async", sync_interval_ms=4
+    ).sync_every_steps == 4
     with pytest.raises(NotImplementedError):
         BaguaDistributedOption(algorithm="nope")
 
