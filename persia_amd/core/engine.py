@@ -48,6 +48,25 @@ _logger = get_default_logger("persia_amd.engine")
 _FLIP = -(2 ** 63)  # XOR with sign bit: signed sort order == unsigned order
 
 
+def _roctx_range(name):
+    """Method decorator: named roctx range (rocprofv3 --marker-trace) when
+    the engine runs with PA_ROCTX=1; a bare call otherwise."""
+    import functools
+
+    def deco(fn):
+        @functools.wraps(fn)
+        def wrap(self, *a, **k):
+            if not getattr(self, "_roctx", False):
+                return fn(self, *a, **k)
+            torch.cuda.nvtx.range_push(name)
+            try:
+                return fn(self, *a, **k)
+            finally:
+                torch.cuda.nvtx.range_pop()
+        return wrap
+    return deco
+
+
 def _owner_of_keys(keys: torch.Tensor, world_size: int) -> torch.Tensor:
     """Monotone range-partition of u64 bit-pattern keys (int64 tensors) to
     ranks — same math as hashing.owner_of."""
@@ -347,6 +366,10 @@ class EmbeddingEngine:
         # (copy in place of the collectives) for single-GPU tests/profiling
         self._fused_dist = _os.environ.get("PA_FUSED_DIST", "1") == "1"
         self._force_dist = _os.environ.get("PA_FORCE_DIST", "0") == "1"
+        # PA_ROCTX=1: named roctx ranges around the pipeline stages for
+        # rocprofv3 --marker-trace (torch.cuda.nvtx IS roctx on ROCm);
+        # off by default so the hot loop never pays the marker calls
+        self._roctx = _os.environ.get("PA_ROCTX", "0") == "1"
         self._a2a_overflow = torch.zeros(1, dtype=torch.int64, device=device)
         from persia_amd.core.metrics import EngineMetrics
 
@@ -538,6 +561,7 @@ class EmbeddingEngine:
         self._a2a_overflow.add_((counts[:world] > cap).sum())
         return send, idx
 
+    @_roctx_range("persia:a2a_exchange")
     def _a2a_exchange_fwd(self, plan: "_GroupPlan", group: _GroupCtx,
                           train: bool):
         """Padded forward exchange: route keys -> even a2a -> owner lookup in
@@ -593,6 +617,7 @@ class EmbeddingEngine:
 
     # ---------------------------------------------------------- forward path
 
+    @_roctx_range("persia:process_batch")
     def process_batch(self, batch: PersiaBatch, train: Optional[bool] = None) -> PersiaTrainingBatch:
         if self._prod_timing:
             if self._pt["skip"] > 0:
@@ -1180,6 +1205,7 @@ class EmbeddingEngine:
 
     # --------------------------------------------------------- backward path
 
+    @_roctx_range("persia:update")
     def apply_gradients(
         self,
         training_batch: PersiaTrainingBatch,
@@ -1392,6 +1418,7 @@ class EmbeddingEngine:
         if self.incremental is not None:
             self.incremental.record_keys(uk)
 
+    @_roctx_range("persia:update")
     def apply_gradients_base(
         self,
         training_batch: PersiaTrainingBatch,
