@@ -8,6 +8,7 @@
   persia_amd.core.engine.ForwardPipeline)
 """
 import threading
+import time
 from abc import ABC, abstractmethod
 from typing import Iterator, Optional
 
@@ -80,6 +81,7 @@ class DataLoader:
         self.reproducible = reproducible
         self.embedding_staleness = embedding_staleness or 8
         self._pipeline: Optional[ForwardPipeline] = None
+        self._last_slow_warn = 0.0
 
     def _ensure_pipeline(self) -> ForwardPipeline:
         if self._pipeline is None:
@@ -113,7 +115,18 @@ class DataLoader:
         feeder.start()
         try:
             while True:
+                _t0 = time.perf_counter()
                 tb = pipeline.get(timeout=self.timeout)
+                _wait = time.perf_counter() - _t0
+                if _wait > 1e-3 and _t0 - self._last_slow_warn > 10.0:
+                    # reference forward.rs:860-897 warns when get_batch
+                    # blocks >1 ms (trainer starving on the lookup
+                    # pipeline); rate-limited to one warning per 10 s
+                    self._last_slow_warn = _t0
+                    _logger.warning(
+                        f"get_batch waited {_wait*1e3:.1f} ms; consider a "
+                        f"larger embedding_staleness/forward_buffer_size"
+                    )
                 if tb is None:
                     break
                 if not tb.requires_grad:
