@@ -170,6 +170,13 @@ class FusedMLP(nn.Module):
 
     def forward(self, x):
         for l in self.layers:
+            if not isinstance(l, FusedLinear):
+                # plain final projection (e.g. the 1-logit head): follow its
+                # weight dtype — FusedLinear outputs bf16 even when the
+                # module holds f32 weights (mixed-precision user models)
+                w = next(l.parameters(), None)
+                if w is not None and x.dtype != w.dtype:
+                    x = x.to(w.dtype)
             x = l(x)
         return x
 
