@@ -198,3 +198,28 @@ def test_gencrd_roundtrip(capsys):
     assert out == CRD
     assert out["spec"]["names"]["kind"] == "PersiaJob"
     assert out["spec"]["versions"][0]["subresources"] == {"status": {}}
+
+
+def test_operator_broken_cr_does_not_stall_healthy_ones(fake_api):
+    """One CR whose spec explodes during manifest rendering must not stop
+    the reconcile loop from driving the healthy CRs (k8s_operator
+    reconcile_all catches per-CR failures)."""
+    from persia_amd.k8s_operator import K8sApi, Operator
+
+    store, url = fake_api
+    store.crs["bad"] = {
+        "apiVersion": "persia.ai/v1",
+        "kind": "PersiaJob",
+        "metadata": {"name": "bad"},
+        "spec": {"gpus_per_node": "not-a-number"},
+    }
+    store.crs["good"] = {
+        "apiVersion": "persia.ai/v1",
+        "kind": "PersiaJob",
+        "metadata": {"name": "good"},
+        "spec": {"gpus_per_node": 8},
+    }
+    phases = Operator(K8sApi(url, "default")).reconcile_all()
+    assert phases["bad"].startswith("Error:")
+    assert phases["good"] == "Running"
+    assert ("Job", "good-trainer") in store.workloads
