@@ -72,3 +72,23 @@ def test_engine_metrics_noop_when_disabled():
     m.staleness.set(3)
     m.nan_count.inc()
     m.index_miss_count.labels("feat_a").inc(2)
+
+
+def test_watchdog_dumps_stacks(monkeypatch):
+    """PERSIA_DEADLOCK_DETECTION=1 starts the stack-dump thread and a dump
+    names live threads (reference parking_lot detector, utils.rs:22-48)."""
+    import time
+
+    from persia_amd.core import watchdog
+
+    dumps = []
+    monkeypatch.setattr(watchdog, "_started", False)
+    monkeypatch.setattr(watchdog._logger, "warning", dumps.append)
+    monkeypatch.setenv("PERSIA_DEADLOCK_DETECTION", "1")
+    watchdog.maybe_start_deadlock_detection(interval_sec=0.1)
+    for _ in range(100):
+        if dumps:
+            break
+        time.sleep(0.05)
+    assert dumps, "no stack dump produced"
+    assert "watchdog:" in dumps[0] and "MainThread" in dumps[0]
