@@ -13,6 +13,14 @@ from persia_amd.logger import get_default_logger
 _logger = get_default_logger("persia_amd.watchdog")
 
 _started = False
+_stop = threading.Event()
+
+
+def stop_deadlock_detection() -> None:
+    """Stop the watchdog thread (tests / shutdown)."""
+    global _started
+    _stop.set()
+    _started = False
 
 
 def maybe_start_deadlock_detection(interval_sec: float = 60.0) -> None:
@@ -20,10 +28,10 @@ def maybe_start_deadlock_detection(interval_sec: float = 60.0) -> None:
     if _started or os.environ.get("PERSIA_DEADLOCK_DETECTION", "0") not in ("1", "true"):
         return
     _started = True
+    _stop.clear()
 
     def loop():
-        while True:
-            threading.Event().wait(interval_sec)
+        while not _stop.wait(interval_sec):
             frames = sys._current_frames()
             lines = [f"--- watchdog: {len(frames)} threads ---"]
             for tid, frame in frames.items():

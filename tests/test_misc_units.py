@@ -86,9 +86,12 @@ def test_watchdog_dumps_stacks(monkeypatch):
     monkeypatch.setattr(watchdog._logger, "warning", dumps.append)
     monkeypatch.setenv("PERSIA_DEADLOCK_DETECTION", "1")
     watchdog.maybe_start_deadlock_detection(interval_sec=0.1)
-    for _ in range(100):
-        if dumps:
-            break
-        time.sleep(0.05)
+    try:
+        for _ in range(100):
+            if dumps:
+                break
+            time.sleep(0.05)
+    finally:
+        watchdog.stop_deadlock_detection()
     assert dumps, "no stack dump produced"
     assert "watchdog:" in dumps[0] and "MainThread" in dumps[0]
