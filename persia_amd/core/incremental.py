@@ -166,7 +166,13 @@ class IncrementalUpdateLoader:
             for fn in sorted(os.listdir(full)):
                 if not fn.endswith(".inc"):
                     continue
-                signs, inner, dim = read_emb_file(os.path.join(full, fn))
+                try:
+                    signs, inner, dim = read_emb_file(os.path.join(full, fn))
+                except Exception as e:
+                    # a torn/corrupt packet file must not wedge freshness:
+                    # skip it (the packet dir is marked seen either way)
+                    _logger.warning(f"skipping corrupt packet {full}/{fn}: {e}")
+                    continue
                 if dim in self.engine.stores and len(signs):
                     self.engine.stores[dim].import_rows(signs, inner)
                     loaded += len(signs)

@@ -173,3 +173,37 @@ def test_incremental_enabled_from_global_config(tmp_path):
     assert eng.incremental is not None
     assert eng.incremental.buffer_size == 77
     assert eng.incremental.dst_dir == str(tmp_path / "inc")
+
+
+def test_incremental_loader_skips_corrupt_packet(metrics_engine, tmp_path):
+    """A torn/corrupt .inc file is skipped with a warning; valid files in
+    the same packet still apply and the loader does not wedge."""
+    import os
+
+    import torch
+
+    eng = metrics_engine
+    eng.enable_incremental_update(str(tmp_path), buffer_size=10)
+    tb = eng.process_batch(_batch(seed=0, B=32))
+    eng.apply_gradients(tb, {
+        "user": torch.full((32, 8), 0.1), "item": torch.full((32, 8), -0.1),
+    })
+    eng.incremental.flush()
+    pkts = sorted(tmp_path.glob("inc_*"))
+    assert pkts
+    # corrupt one packet file in place (truncated header)
+    victim = sorted(pkts[0].glob("*.inc"))[0]
+    victim.write_bytes(b"PAEM")
+    from persia_amd.core.comm import DistContext
+    from persia_amd.core.engine import EmbeddingEngine
+    from persia_amd.core.incremental import IncrementalUpdateLoader
+    from persia_amd.core.schema import GlobalConfig
+
+    eng2 = EmbeddingEngine(
+        schema=eng.schema, hyper=eng.hyper, optimizer=eng.optimizer,
+        gconf=GlobalConfig(capacity=1 << 12), device=torch.device("cpu"),
+        dist_ctx=DistContext(1, 0),
+    )
+    loader = IncrementalUpdateLoader(eng2, str(tmp_path))
+    loader.scan_once()  # must not raise
+    assert pkts[0].name in loader._seen
