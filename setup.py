@@ -4,7 +4,7 @@
 """
 import os
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -31,7 +31,7 @@ ext = CUDAExtension(
 setup(
     name="persia_amd",
     version="0.1.0",
-    packages=["persia_amd"],
+    packages=find_packages(include=["persia_amd", "persia_amd.*"]),
     ext_modules=[ext],
     cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},
 )
